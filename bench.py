@@ -1,0 +1,147 @@
+"""Flagship benchmark: DLRM training on Criteo-TB-shaped synthetic data.
+
+Metric (BASELINE.json): samples/sec (whole node) DLRM at 1/2/4/8 MI355X.
+Weak scaling: per-GPU batch is fixed (default 8192, the reference's GPU
+benchmark batch size, modelzoo/benchmark/gpu/config.yaml), so value is the
+aggregate samples/sec over all ranks.
+
+Launch (single GPU):   python bench.py --steps 50 --warmup 10
+Launch (N GPUs): python -m torch.distributed.run --nnodes=1 \
+    --nproc-per-node N --master-addr 127.0.0.1 bench.py --gpus N ...
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+
+DLRM_BASELINE_SAMPLES_SEC = 141266.06  # DeepRec DLRM FP32+BF16
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=50)
+    p.add_argument("--warmup", type=int, default=10)
+    p.add_argument("--batch", type=int, default=8192)
+    p.add_argument("--device", type=str, default=None)
+    p.add_argument("--optimizer", type=str, default="adamasync")
+    p.add_argument("--no-bf16", action="store_true")
+    args = p.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    distributed = world_size > 1
+
+    if args.device:
+        device = torch.device(args.device)
+    elif torch.cuda.is_available():
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
+
+    if distributed:
+        import torch.distributed as dist
+        backend = "nccl" if device.type == "cuda" else "gloo"
+        dist.init_process_group(backend=backend, rank=rank,
+                                world_size=world_size)
+
+    from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.optimizers import make_optimizer
+
+    torch.manual_seed(42 + rank)
+    bf16 = (not args.no_bf16) and device.type == "cuda"
+    model = DLRM(device=device, bf16=bf16, sharded=distributed)
+    ds = CriteoSyntheticDataset(batch_size=args.batch, device=device,
+                                seed=1234, rank=rank)
+    opt = make_optimizer(args.optimizer, params=model.parameters(),
+                         embedding_variables=model.embedding_variables(),
+                         learning_rate=0.001)
+
+    if distributed:
+        from deeprec_amd.parallel import DenseGradAllreducer
+        reducer = DenseGradAllreducer(model.parameters())
+    else:
+        reducer = None
+
+    # pre-generate batches outside the timed region (CPU RNG is not the
+    # system under test); embedding ids differ per step so hash-table and
+    # optimizer work is real every timed step
+    n_total = args.warmup + args.steps
+    batches = [ds.next_batch() for _ in range(min(n_total, 20))]
+
+    def one_step(i):
+        dense, sparse, labels = batches[i % len(batches)]
+        logits = model(dense, sparse)
+        loss = model.loss_fn(logits, labels)
+        opt.zero_grad()
+        loss.backward()
+        if reducer is not None:
+            reducer.allreduce()
+        opt.step()
+        return loss
+
+    for i in range(args.warmup):
+        one_step(i)
+
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step(args.warmup + i)
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+    if distributed:
+        import torch.distributed as dist
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    if distributed:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if device.type == "cuda" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    n_gpus = world_size if distributed else 1
+    global_batch = args.batch * n_gpus
+    samples_per_sec = global_batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "samples/sec (whole node) DLRM Criteo-TB-shaped synthetic",
+            "value": samples_per_sec,
+            "unit": "samples/sec",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": samples_per_sec / DLRM_BASELINE_SAMPLES_SEC,
+            "dtype": "bf16" if bf16 else "fp32",
+            "data": "synthetic (Criteo-TB-shaped, zipf ids, random-init weights)",
+            "config": {"model": "dlrm", "global_batch": global_batch,
+                       "seq_len": 1,
+                       "parallelism": f"dp{n_gpus}+ep{n_gpus}"
+                       if distributed else "single",
+                       "optimizer": args.optimizer,
+                       "embedding_dim": 16, "num_tables": 26},
+        }))
+
+    if distributed:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

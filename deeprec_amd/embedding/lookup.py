@@ -1,0 +1,136 @@
+"""Embedding lookup front-ends with autograd integration.
+
+Capability parity with the reference's lookup surface
+(reference: python/ops/embedding_ops.py — embedding_lookup,
+embedding_lookup_sparse, safe_embedding_lookup_sparse,
+group_embedding_lookup_sparse). Data flow per lookup:
+
+  unique(ids) -> one hash probe (lookup_or_create -> slots)
+    -> fused gather+segment-pool (forward)
+    -> fused grad scatter to unique keys (backward)
+    -> fused sparse optimizer apply on the same slots (optimizer.step)
+
+The slots returned by the single probe are reused end-to-end — the
+reference's `_OPT_KvResourceLookupID/CollectEmbedding` pointer-passing
+fusion (ops/kv_variable_ops.cc:636).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+import torch
+
+from deeprec_amd.embedding.ragged import RaggedIds
+from deeprec_amd.embedding.variable import EmbeddingVariable
+from deeprec_amd.ops import functional as F
+
+
+class _PooledLookup(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, anchor, ev, uniq, slots, inverse, offsets, row_ids,
+                combiner, weights, out_dtype):
+        out = ev.storage.pooled_lookup(
+            uniq, slots, inverse, offsets, row_ids, combiner, weights,
+            out_dtype) if hasattr(ev.storage, "pooled_lookup") else None
+        if out is None:
+            emb = ev.storage.gather(uniq, slots)
+            out = F.pooled_forward(emb, inverse, offsets, row_ids, combiner,
+                                   weights, out_dtype)
+        ctx.ev = ev
+        ctx.combiner = combiner
+        ctx.save_for_backward(uniq, slots, inverse, offsets, row_ids)
+        ctx.weights = weights
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        uniq, slots, inverse, offsets, row_ids = ctx.saved_tensors
+        ev = ctx.ev
+        if hasattr(ev.storage, "pooled_grad"):
+            grad_unique = ev.storage.pooled_grad(
+                grad_out, inverse, offsets, row_ids, uniq.numel(),
+                ctx.combiner, ctx.weights)
+        else:
+            grad_unique = F.pooled_backward(
+                grad_out, inverse, offsets, row_ids, uniq.numel(),
+                ctx.combiner, ctx.weights)
+        ev.accumulate_grad(slots, uniq, grad_unique)
+        return (torch.zeros_like(ctx.ev._anchor),) + (None,) * 9
+
+
+def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,
+                            combiner: str = "mean",
+                            out_dtype=None,
+                            train: bool = True) -> torch.Tensor:
+    """Pooled lookup: [batch, dim]."""
+    from deeprec_amd.parallel.sharded_embedding import (
+        ShardedEmbeddingVariable, sharded_embedding_lookup_sparse)
+    if isinstance(ev, ShardedEmbeddingVariable):
+        return sharded_embedding_lookup_sparse(ev, sp_ids, combiner,
+                                               out_dtype, train)
+    uniq, inverse, counts = torch.unique(
+        sp_ids.values, return_inverse=True, return_counts=True)
+    slots = ev.lookup_or_create(uniq, counts, train=train)
+    row_ids = sp_ids.row_ids()
+    if not (train and ev.trainable):
+        emb = ev.storage.gather(uniq, slots)
+        return F.pooled_forward(emb, inverse, offsets=sp_ids.offsets,
+                                row_ids=row_ids, combiner=combiner,
+                                weights=sp_ids.weights, out_dtype=out_dtype)
+    return _PooledLookup.apply(ev._anchor, ev, uniq, slots,
+                               inverse.to(torch.int32), sp_ids.offsets,
+                               row_ids, combiner, sp_ids.weights, out_dtype)
+
+
+def safe_embedding_lookup_sparse(ev, sp_ids: RaggedIds, combiner="mean",
+                                 out_dtype=None, train=True):
+    """Like embedding_lookup_sparse; empty rows yield zeros (already the
+    pooled kernels' behavior) and negative ids are dropped.
+    (reference: fused_safe_embedding_lookup_sparse, embedding_ops.py:1306)"""
+    if bool((sp_ids.values < 0).any()):
+        keep = sp_ids.values >= 0
+        row_ids = sp_ids.row_ids()[keep]
+        values = sp_ids.values[keep]
+        weights = sp_ids.weights[keep] if sp_ids.weights is not None else None
+        lengths = torch.zeros(sp_ids.batch_size, dtype=torch.int64,
+                              device=values.device)
+        lengths.index_add_(0, row_ids.long(),
+                           torch.ones_like(row_ids, dtype=torch.int64))
+        offsets = torch.zeros(sp_ids.batch_size + 1, dtype=sp_ids.offsets.dtype,
+                              device=values.device)
+        offsets[1:] = lengths.cumsum(0)
+        sp_ids = RaggedIds(values, offsets, weights)
+    return embedding_lookup_sparse(ev, sp_ids, combiner, out_dtype, train)
+
+
+def embedding_lookup(ev: EmbeddingVariable, ids: torch.Tensor,
+                     out_dtype=None, train: bool = True) -> torch.Tensor:
+    """Unpooled lookup: ids [..., ] -> [..., dim] (sequence features)."""
+    flat = ids.reshape(-1)
+    n = flat.numel()
+    offsets = torch.arange(n + 1, dtype=torch.int32, device=flat.device)
+    out = embedding_lookup_sparse(
+        ev, RaggedIds(flat, offsets), combiner="sum",
+        out_dtype=out_dtype, train=train)
+    return out.reshape(*ids.shape, ev.dim)
+
+
+def group_embedding_lookup_sparse(evs: Sequence[EmbeddingVariable],
+                                  sp_ids_list: Sequence[RaggedIds],
+                                  combiners=None, out_dtype=None,
+                                  train: bool = True) -> List[torch.Tensor]:
+    """N lookups in one call (reference: GroupEmbeddingVarLookup,
+    ops/kv_variable_ops.cc:404). The HIP backend batches the probe +
+    gather/pool of all tables into grouped kernel launches; the generic
+    path simply loops."""
+    if combiners is None:
+        combiners = ["mean"] * len(evs)
+    if len(evs) != len(sp_ids_list) or len(evs) != len(combiners):
+        raise ValueError("group_embedding_lookup_sparse: length mismatch")
+    dev = evs[0].device if evs else torch.device("cpu")
+    all_plain = all(isinstance(e, EmbeddingVariable) for e in evs)
+    if dev.type == "cuda" and train and all_plain:
+        from deeprec_amd.ops.hip_backend import group_pooled_lookup
+        return group_pooled_lookup(evs, sp_ids_list, combiners, out_dtype)
+    return [embedding_lookup_sparse(ev, sp, c, out_dtype, train)
+            for ev, sp, c in zip(evs, sp_ids_list, combiners)]

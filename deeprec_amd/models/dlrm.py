@@ -1,0 +1,107 @@
+"""DLRM — deep learning recommendation model.
+
+Capability parity with the reference model (modelzoo/dlrm/train.py:70-283):
+bottom MLP over 13 dense features -> [*, 16]; 26 EmbeddingVariable lookups
+(dim 16); pairwise-dot (or concat) feature interaction; top MLP
+[512, 256, 1]; BCE loss. bf16 compute with fp32 master weights mirrors the
+reference's keep_weights bf16 scope.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+from deeprec_amd.data.synthetic import NUM_DENSE, NUM_SPARSE
+from deeprec_amd.embedding import (
+    EmbeddingVariable, EmbeddingVariableOption,
+    group_embedding_lookup_sparse,
+)
+
+
+def _mlp(sizes, in_dim, final_activation=True):
+    layers: List[nn.Module] = []
+    d = in_dim
+    for i, h in enumerate(sizes):
+        layers.append(nn.Linear(d, h))
+        if final_activation or i + 1 < len(sizes):
+            layers.append(nn.ReLU(inplace=True))
+        d = h
+    return nn.Sequential(*layers)
+
+
+class DLRM(nn.Module):
+    def __init__(self, embedding_dim: int = 16,
+                 mlp_bot=(512, 256, 64, 16), mlp_top=(512, 256),
+                 interaction_op: str = "dot", bf16: bool = True,
+                 device="cpu", ev_option: Optional[EmbeddingVariableOption] = None,
+                 num_sparse: int = NUM_SPARSE, name_prefix: str = "dlrm",
+                 sharded: bool = False):
+        super().__init__()
+        assert mlp_bot[-1] == embedding_dim, \
+            "bottom MLP output must match embedding dim (dot interaction)"
+        self.embedding_dim = embedding_dim
+        self.interaction_op = interaction_op
+        self.bf16 = bf16
+        self.device_ = torch.device(device)
+        self.num_sparse = num_sparse
+
+        self.mlp_bot = _mlp(mlp_bot, NUM_DENSE)
+        n_f = num_sparse + 1
+        inter_dim = (n_f * (n_f - 1)) // 2 if interaction_op == "dot" \
+            else n_f * embedding_dim
+        self.mlp_top = _mlp(list(mlp_top) + [1], embedding_dim + inter_dim
+                            if interaction_op == "dot" else inter_dim,
+                            final_activation=False)
+        self.to(self.device_)
+
+        if sharded:
+            from deeprec_amd.parallel.sharded_embedding import (
+                ShardedEmbeddingVariable)
+            self.evs = [
+                ShardedEmbeddingVariable(f"{name_prefix}/C{i+1}",
+                                         embedding_dim,
+                                         ev_option=ev_option,
+                                         device=self.device_)
+                for i in range(num_sparse)]
+        else:
+            self.evs = [
+                EmbeddingVariable(f"{name_prefix}/C{i+1}", embedding_dim,
+                                  ev_option=ev_option, device=self.device_)
+                for i in range(num_sparse)]
+
+    def embedding_variables(self):
+        return self.evs
+
+    def _interact(self, feats: torch.Tensor) -> torch.Tensor:
+        """feats: [B, F, D] -> pairwise dots, upper triangle (i<j)
+        (reference: _dot_op, modelzoo/dlrm/train.py:121-132)."""
+        z = torch.bmm(feats, feats.transpose(1, 2))  # [B, F, F]
+        f = feats.shape[1]
+        iu = torch.triu_indices(f, f, offset=1, device=feats.device)
+        return z[:, iu[0], iu[1]]
+
+    def forward(self, dense: torch.Tensor, sparse_ids,
+                train: bool = True) -> torch.Tensor:
+        compute_dtype = torch.bfloat16 if (
+            self.bf16 and dense.device.type == "cuda") else torch.float32
+        emb_list = group_embedding_lookup_sparse(
+            self.evs, sparse_ids, combiners=["mean"] * len(self.evs),
+            out_dtype=compute_dtype, train=train)
+        import contextlib
+        amp = (torch.autocast(device_type="cuda", dtype=torch.bfloat16)
+               if compute_dtype == torch.bfloat16 else contextlib.nullcontext())
+        with amp:
+            bot = self.mlp_bot(dense)
+            feats = torch.stack([bot] + list(emb_list), dim=1)  # [B, F+1, D]
+            if self.interaction_op == "dot":
+                inter = self._interact(feats)
+                top_in = torch.cat([bot, inter], dim=1)
+            else:
+                top_in = feats.flatten(1)
+            logits = self.mlp_top(top_in)
+        return logits.squeeze(1).float()
+
+    def loss_fn(self, logits: torch.Tensor, labels: torch.Tensor):
+        return nn.functional.binary_cross_entropy_with_logits(logits, labels)

@@ -1,0 +1,84 @@
+"""Collective communication helpers.
+
+Single communication stack by design: torch.distributed with the "nccl"
+backend, which on ROCm IS RCCL over xGMI (the reference's four coexisting
+stacks — gRPC/seastar/NCCL/MPI, SURVEY.md §5 — collapse to this one for
+single-node collective training). The gloo backend backs CPU-only tests;
+gloo lacks all_to_all, so an equivalent point-to-point emulation keeps one
+code path testable without GPUs.
+"""
+from __future__ import annotations
+
+import torch
+import torch.distributed as dist
+
+_ALLTOALL_SUPPORTED = None
+
+
+def is_initialized() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def world_size() -> int:
+    return dist.get_world_size() if is_initialized() else 1
+
+
+def rank() -> int:
+    return dist.get_rank() if is_initialized() else 0
+
+
+def _supports_all_to_all() -> bool:
+    global _ALLTOALL_SUPPORTED
+    if _ALLTOALL_SUPPORTED is None:
+        _ALLTOALL_SUPPORTED = dist.get_backend() != "gloo"
+    return _ALLTOALL_SUPPORTED
+
+
+def all_to_all_single(inp: torch.Tensor, in_splits, out_splits) -> torch.Tensor:
+    """Variable-size all-to-all along dim 0. in_splits/out_splits: python
+    lists of per-peer element counts (dim-0 rows)."""
+    w = world_size()
+    if w == 1:
+        return inp.clone()
+    out_shape = (sum(out_splits),) + tuple(inp.shape[1:])
+    out = torch.empty(out_shape, dtype=inp.dtype, device=inp.device)
+    if _supports_all_to_all():
+        dist.all_to_all_single(out, inp.contiguous(),
+                               output_split_sizes=list(out_splits),
+                               input_split_sizes=list(in_splits))
+        return out
+    # gloo emulation: pairwise send/recv
+    r = rank()
+    in_off = [0]
+    for s in in_splits:
+        in_off.append(in_off[-1] + s)
+    out_off = [0]
+    for s in out_splits:
+        out_off.append(out_off[-1] + s)
+    reqs = []
+    for peer in range(w):
+        if peer == r:
+            out[out_off[r]:out_off[r + 1]] = inp[in_off[r]:in_off[r + 1]]
+            continue
+        chunk = inp[in_off[peer]:in_off[peer + 1]].contiguous()
+        reqs.append(dist.isend(chunk, dst=peer, tag=0))
+    for peer in range(w):
+        if peer == r:
+            continue
+        buf = torch.empty((out_splits[peer],) + tuple(inp.shape[1:]),
+                          dtype=inp.dtype, device=inp.device)
+        dist.recv(buf, src=peer, tag=0)
+        out[out_off[peer]:out_off[peer + 1]] = buf
+    for q in reqs:
+        q.wait()
+    return out
+
+
+def exchange_counts(counts: torch.Tensor) -> torch.Tensor:
+    """counts int64[w] (rows destined to each peer) -> int64[w] rows
+    arriving from each peer. The two-phase count-then-payload exchange of
+    the reference's all-to-all dispatcher."""
+    w = world_size()
+    if w == 1:
+        return counts.clone()
+    return all_to_all_single(counts.contiguous(), [1] * w, [1] * w)
