@@ -1,0 +1,770 @@
+// ev_kernels.hip — MI355X (gfx950, CDNA4) embedding-engine kernels.
+//
+// The HBM embedding engine: an open-addressing (linear probe) hash table
+// keyed by int64 feature id mapping to a monotonically-allocated int32
+// value-slot, with per-entry frequency/version metadata for feature
+// admission (counter filter) and eviction. Value rows live in a dense
+// [max_slots, dim] fp32 slab; optimizer states are parallel slabs indexed
+// by the same slots.
+//
+// Capability parity (not a port) with the reference engine:
+//   - hash lookup/insert  ≙ GPUHashTable/cuco dynamic_map probing
+//     (reference: gpu_hash_table.cu.cc:260-541) — re-designed as a single
+//     flat power-of-two table with 64-wide-wavefront-friendly per-thread
+//     probes; keys are pre-uniqued on the host side of the step so probes
+//     are contention-free except first-insert CAS.
+//   - fused gather+segment-pool ≙ EmbeddingLookUp + ApplyCombiner
+//     (reference: fused_embedding_local_ops_gpu.cu.cc:42, SumUpEmbeddingShard)
+//   - grad scatter ≙ DoEmbeddingGrad/DistributeGradToShard
+//   - sparse optimizer applies ≙ training_ali_ops_gpu.cu.cc kernels,
+//     operating directly on slot indices from the step's single probe
+//     (the reference's `_OPT_` indices-as-pointers fusion).
+//
+// CDNA4 notes: wavefront = 64; all elementwise kernels use a (row, dim)
+// thread mapping with grid-stride so consecutive lanes read consecutive
+// value-row elements (coalesced); blocks are multiples of 64; atomics are
+// device-scope by default (per-XCD L2 non-coherence is safe).
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+// current-stream accessor (ROCm ATen name)
+static inline hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+#include <cstdint>
+
+#define DEV_INLINE __device__ __forceinline__
+
+static constexpr int64_t EMPTY_KEY = INT64_MIN;  // reserved sentinel
+static constexpr int kBlock = 256;
+
+namespace {
+
+DEV_INLINE uint64_t mix_hash(uint64_t k) {
+  // splitmix64 finalizer — good avalanche for sequential ids
+  k += 0x9E3779B97F4A7C15ull;
+  k = (k ^ (k >> 30)) * 0xBF58476D1CE4E5B9ull;
+  k = (k ^ (k >> 27)) * 0x94D049BB133111EBull;
+  return k ^ (k >> 31);
+}
+
+DEV_INLINE float bf2f(__hip_bfloat16 v) { return __bfloat162float(v); }
+DEV_INLINE __hip_bfloat16 f2bf(float v) { return __float2bfloat16(v); }
+
+// ---------------------------------------------------------------------
+// hash probe / insert
+// ---------------------------------------------------------------------
+
+// Lookup-or-insert for a batch of UNIQUE keys. Per key:
+//   - find or claim (CAS) a hash entry
+//   - freq += count; version = step (no atomics needed: keys unique)
+//   - admit a value slot once freq >= filter_freq; initialize the value
+//     row from default_values[key % default_value_dim]
+// out_slots[i] = slot or -1 (not admitted).
+__global__ void k_lookup_insert(
+    const int64_t* __restrict__ keys, const int32_t* __restrict__ counts,
+    int n, int64_t* __restrict__ ht_keys, int32_t* __restrict__ ht_slot,
+    int32_t* __restrict__ ht_freq, int64_t* __restrict__ ht_version,
+    int64_t cap_mask, int32_t* __restrict__ slot_counter,
+    int32_t* __restrict__ entry_counter, int max_slots,
+    float* __restrict__ values, const float* __restrict__ default_values,
+    int dim, int default_value_dim, int filter_freq, int64_t step,
+    int train, int32_t* __restrict__ out_slots,
+    int32_t* __restrict__ error_flag) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int64_t key = keys[i];
+  uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
+  int32_t slot = -1;
+  for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+    int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
+    int64_t cur = ht_keys[idx];
+    if (cur != key) {
+      if (cur != EMPTY_KEY) continue;
+      if (!train) { out_slots[i] = -1; return; }
+      int64_t prev = (int64_t)atomicCAS(
+          (unsigned long long*)&ht_keys[idx],
+          (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+      if (prev != EMPTY_KEY && prev != key) continue;  // lost race
+      if (prev == EMPTY_KEY) atomicAdd(entry_counter, 1);
+    }
+    // found or claimed entry idx
+    int32_t freq = ht_freq[idx];
+    if (train) {
+      freq += (counts ? counts[i] : 1);
+      ht_freq[idx] = freq;
+      ht_version[idx] = step;
+    }
+    slot = ht_slot[idx];
+    if (slot < 0 && train && freq >= filter_freq) {
+      slot = atomicAdd(slot_counter, 1);
+      if (slot >= max_slots) {  // out of slab space: host must grow
+        atomicExch(error_flag, 1);
+        out_slots[i] = -1;
+        return;
+      }
+      ht_slot[idx] = slot;
+      const float* src =
+          default_values +
+          (int64_t)((uint64_t)key % (uint64_t)default_value_dim) * dim;
+      float* dst = values + (int64_t)slot * dim;
+      for (int d = 0; d < dim; ++d) dst[d] = src[d];
+    }
+    out_slots[i] = slot;
+    return;
+  }
+  atomicExch(error_flag, 2);  // table full (host sizing bug)
+  out_slots[i] = -1;
+}
+
+// Bulk import used by restore: entries are created admitted with the given
+// slot ids (slots pre-assigned densely by the host).
+__global__ void k_insert_bulk(
+    const int64_t* __restrict__ keys, const int32_t* __restrict__ slots,
+    const int32_t* __restrict__ freqs, const int64_t* __restrict__ versions,
+    int n, int64_t* __restrict__ ht_keys, int32_t* __restrict__ ht_slot,
+    int32_t* __restrict__ ht_freq, int64_t* __restrict__ ht_version,
+    int64_t cap_mask, int32_t* __restrict__ entry_counter,
+    int32_t* __restrict__ error_flag) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int64_t key = keys[i];
+  uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
+  for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+    int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
+    int64_t cur = ht_keys[idx];
+    if (cur != key) {
+      if (cur != EMPTY_KEY) continue;
+      int64_t prev = (int64_t)atomicCAS(
+          (unsigned long long*)&ht_keys[idx],
+          (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+      if (prev != EMPTY_KEY && prev != key) continue;
+      if (prev == EMPTY_KEY) atomicAdd(entry_counter, 1);
+    }
+    ht_slot[idx] = slots[i];
+    if (freqs) ht_freq[idx] = freqs[i];
+    if (versions) ht_version[idx] = versions[i];
+    return;
+  }
+  atomicExch(error_flag, 2);
+}
+
+// Read-only probe (serving / frequency / version queries).
+// out_slots: slot or -1; out_entry: hash index or -1 (metadata access).
+__global__ void k_lookup(
+    const int64_t* __restrict__ keys, int n,
+    const int64_t* __restrict__ ht_keys, const int32_t* __restrict__ ht_slot,
+    int64_t cap_mask, int32_t* __restrict__ out_slots,
+    int64_t* __restrict__ out_entry) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n) return;
+  const int64_t key = keys[i];
+  uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
+  for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+    int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
+    int64_t cur = ht_keys[idx];
+    if (cur == key) {
+      out_slots[i] = ht_slot[idx];
+      if (out_entry) out_entry[i] = idx;
+      return;
+    }
+    if (cur == EMPTY_KEY) break;
+  }
+  out_slots[i] = -1;
+  if (out_entry) out_entry[i] = -1;
+}
+
+// Export scan: compact live hash entries into dense output arrays.
+__global__ void k_export_scan(
+    const int64_t* __restrict__ ht_keys, const int32_t* __restrict__ ht_slot,
+    const int32_t* __restrict__ ht_freq,
+    const int64_t* __restrict__ ht_version, int64_t capacity,
+    int32_t* __restrict__ cursor, int64_t* __restrict__ out_keys,
+    int32_t* __restrict__ out_slots, int32_t* __restrict__ out_freqs,
+    int64_t* __restrict__ out_versions, int out_cap) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < capacity; i += stride) {
+    int64_t k = ht_keys[i];
+    if (k == EMPTY_KEY) continue;
+    int32_t pos = atomicAdd(cursor, 1);
+    if (pos >= out_cap) continue;
+    out_keys[pos] = k;
+    out_slots[pos] = ht_slot[i];
+    out_freqs[pos] = ht_freq[i];
+    out_versions[pos] = ht_version[i];
+  }
+}
+
+// ---------------------------------------------------------------------
+// gather / pooled forward / pooled backward
+// ---------------------------------------------------------------------
+
+// Plain gather of unique rows: out[i] = values[slots[i]] or default row.
+template <typename OutT>
+__global__ void k_gather(
+    const float* __restrict__ values, const float* __restrict__ default_values,
+    const int64_t* __restrict__ keys, const int32_t* __restrict__ slots,
+    int m, int dim, int default_value_dim, float no_permission_value,
+    int use_no_permission, OutT* __restrict__ out) {
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)m * dim;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; t < total; t += stride) {
+    int i = (int)(t / dim);
+    int d = (int)(t % dim);
+    int32_t s = slots[i];
+    float v;
+    if (s >= 0) {
+      v = values[(int64_t)s * dim + d];
+    } else if (use_no_permission) {
+      v = no_permission_value;
+    } else {
+      int64_t row = (int64_t)((uint64_t)keys[i] % (uint64_t)default_value_dim);
+      v = default_values[row * dim + d];
+    }
+    if constexpr (std::is_same_v<OutT, __hip_bfloat16>) out[t] = f2bf(v);
+    else out[t] = v;
+  }
+}
+
+DEV_INLINE float row_coeff(int combiner, int len, const float* wsum) {
+  // combiner: 0=sum 1=mean 2=sqrtn; wsum = sum w (mean) / sum w^2 (sqrtn)
+  if (combiner == 0 || len == 0) return 1.0f;
+  float denom = wsum ? *wsum : (float)len;
+  if (combiner == 2) denom = sqrtf(wsum ? *wsum : (float)len);
+  return denom > 1e-12f ? 1.0f / denom : 0.0f;
+}
+
+// Fused gather + segment pooling. Thread = (batch row b, dim d); loops the
+// row's ids accumulating values[slots[inverse[j]]][d]. No [nnz, dim] or
+// [m, dim] intermediate is materialized.
+template <typename OutT>
+__global__ void k_pooled_fwd(
+    const float* __restrict__ values, const float* __restrict__ default_values,
+    const int64_t* __restrict__ keys, const int32_t* __restrict__ slots,
+    const int32_t* __restrict__ inverse, const int32_t* __restrict__ offsets,
+    const float* __restrict__ weights, int batch, int dim,
+    int default_value_dim, float no_permission_value, int use_no_permission,
+    int combiner, OutT* __restrict__ out) {
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)batch * dim;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; t < total; t += stride) {
+    int b = (int)(t / dim);
+    int d = (int)(t % dim);
+    int beg = offsets[b], end = offsets[b + 1];
+    float acc = 0.0f, wacc = 0.0f;
+    for (int j = beg; j < end; ++j) {
+      int u = inverse[j];
+      int32_t s = slots[u];
+      float v;
+      if (s >= 0) {
+        v = values[(int64_t)s * dim + d];
+      } else if (use_no_permission) {
+        v = no_permission_value;
+      } else {
+        int64_t row =
+            (int64_t)((uint64_t)keys[u] % (uint64_t)default_value_dim);
+        v = default_values[row * dim + d];
+      }
+      float w = weights ? weights[j] : 1.0f;
+      acc += w * v;
+      wacc += (combiner == 2) ? w * w : w;
+    }
+    float coeff = 1.0f;
+    int len = end - beg;
+    if (combiner != 0 && len > 0) {
+      float denom = (combiner == 2) ? sqrtf(wacc) : wacc;
+      coeff = denom > 1e-12f ? 1.0f / denom : 0.0f;
+    }
+    float r = acc * coeff;
+    if constexpr (std::is_same_v<OutT, __hip_bfloat16>) out[t] = f2bf(r);
+    else out[t] = r;
+  }
+}
+
+// Pooled backward: grad_unique[u][d] += coeff(b) * w_j * grad_out[b][d]
+// for every occurrence j of unique key u. Thread = (b, d); atomicAdd into
+// grad_unique (unique keys repeat across rows).
+template <typename GradT>
+__global__ void k_pooled_bwd(
+    const GradT* __restrict__ grad_out, const int32_t* __restrict__ inverse,
+    const int32_t* __restrict__ offsets, const float* __restrict__ weights,
+    int batch, int dim, int combiner, float* __restrict__ grad_unique) {
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)batch * dim;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; t < total; t += stride) {
+    int b = (int)(t / dim);
+    int d = (int)(t % dim);
+    int beg = offsets[b], end = offsets[b + 1];
+    int len = end - beg;
+    if (len == 0) continue;
+    float g;
+    if constexpr (std::is_same_v<GradT, __hip_bfloat16>) g = bf2f(grad_out[t]);
+    else g = grad_out[t];
+    float coeff = 1.0f;
+    if (combiner != 0) {
+      float wacc = 0.0f;
+      if (weights) {
+        for (int j = beg; j < end; ++j)
+          wacc += (combiner == 2) ? weights[j] * weights[j] : weights[j];
+      } else {
+        wacc = (float)len;
+      }
+      float denom = (combiner == 2) ? sqrtf(wacc) : wacc;
+      coeff = denom > 1e-12f ? 1.0f / denom : 0.0f;
+    }
+    float gc = g * coeff;
+    for (int j = beg; j < end; ++j) {
+      float w = weights ? weights[j] : 1.0f;
+      atomicAdd(&grad_unique[(int64_t)inverse[j] * dim + d], w * gc);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
+// fused sparse optimizer applies (thread = (unique key i, dim d))
+// ---------------------------------------------------------------------
+
+#define APPLY_PROLOG                                              \
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;     \
+  int64_t total = (int64_t)m * dim;                               \
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;               \
+  for (; t < total; t += stride) {                                \
+    int i = (int)(t / dim);                                       \
+    int d = (int)(t % dim);                                       \
+    int32_t s = slots[i];                                         \
+    if (s < 0) continue;                                          \
+    int64_t o = (int64_t)s * dim + d;                             \
+    float g = grad[t];
+
+#define APPLY_EPILOG }
+
+__global__ void k_apply_sgd(float* __restrict__ w,
+                            const int32_t* __restrict__ slots,
+                            const float* __restrict__ grad, int m, int dim,
+                            float lr) {
+  APPLY_PROLOG
+  w[o] -= lr * g;
+  APPLY_EPILOG
+}
+
+__global__ void k_apply_adagrad(float* __restrict__ w,
+                                float* __restrict__ accum,
+                                const int32_t* __restrict__ slots,
+                                const float* __restrict__ grad, int m,
+                                int dim, float lr, float epsilon) {
+  APPLY_PROLOG
+  float a = accum[o] + g * g;
+  accum[o] = a;
+  w[o] -= lr * g / (sqrtf(a) + epsilon);
+  APPLY_EPILOG
+}
+
+// Periodic accumulator decay; period bookkeeping is per-slot (1-wide slab).
+// Lane d==0 updates the period after all lanes read it — benign ordering:
+// every thread computes the same decay from the pre-update period value
+// because the period slab is read before any write (separate arrays).
+__global__ void k_apply_adagrad_decay(
+    float* __restrict__ w, float* __restrict__ accum,
+    float* __restrict__ period_slab, const int32_t* __restrict__ slots,
+    const float* __restrict__ grad, int m, int dim, float lr, float epsilon,
+    float cur_period, float decay_rate, float baseline) {
+  APPLY_PROLOG
+  float prev_period = period_slab[s];
+  float dp = cur_period - prev_period;
+  if (dp < 0.0f) dp = 0.0f;
+  float decay = powf(decay_rate, dp);
+  float a = accum[o] * decay;
+  if (a < baseline) a = baseline;
+  a += g * g;
+  accum[o] = a;
+  w[o] -= lr * g / (sqrtf(a) + epsilon);
+  APPLY_EPILOG
+  // period_slab is committed by a separate k_commit_period launch so every
+  // (key, d) thread here reads the pre-update period value.
+}
+
+__global__ void k_commit_period(float* __restrict__ period_slab,
+                                const int32_t* __restrict__ slots, int m,
+                                float cur_period) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= m) return;
+  int32_t s = slots[i];
+  if (s >= 0) period_slab[s] = cur_period;
+}
+
+__global__ void k_apply_adam(float* __restrict__ w, float* __restrict__ mom,
+                             float* __restrict__ vel,
+                             const int32_t* __restrict__ slots,
+                             const float* __restrict__ grad, int m, int dim,
+                             float lr_t, float beta1, float beta2,
+                             float epsilon) {
+  APPLY_PROLOG
+  float mn = beta1 * mom[o] + (1.0f - beta1) * g;
+  float vn = beta2 * vel[o] + (1.0f - beta2) * g * g;
+  mom[o] = mn;
+  vel[o] = vn;
+  w[o] -= lr_t * mn / (sqrtf(vn) + epsilon);
+  APPLY_EPILOG
+}
+
+__global__ void k_apply_adamw(float* __restrict__ w, float* __restrict__ mom,
+                              float* __restrict__ vel,
+                              const int32_t* __restrict__ slots,
+                              const float* __restrict__ grad, int m, int dim,
+                              float lr_t, float lr, float beta1, float beta2,
+                              float epsilon, float weight_decay) {
+  APPLY_PROLOG
+  float mn = beta1 * mom[o] + (1.0f - beta1) * g;
+  float vn = beta2 * vel[o] + (1.0f - beta2) * g * g;
+  mom[o] = mn;
+  vel[o] = vn;
+  float wv = w[o];
+  w[o] = wv - lr_t * mn / (sqrtf(vn) + epsilon) - lr * weight_decay * wv;
+  APPLY_EPILOG
+}
+
+__global__ void k_apply_rmsprop(float* __restrict__ w, float* __restrict__ vel,
+                                const int32_t* __restrict__ slots,
+                                const float* __restrict__ grad, int m,
+                                int dim, float lr, float beta2,
+                                float epsilon) {
+  APPLY_PROLOG
+  float vn = beta2 * vel[o] + (1.0f - beta2) * g * g;
+  vel[o] = vn;
+  w[o] -= lr * g / (sqrtf(vn) + epsilon);
+  APPLY_EPILOG
+}
+
+__global__ void k_apply_ftrl(float* __restrict__ w, float* __restrict__ n,
+                             float* __restrict__ z,
+                             const int32_t* __restrict__ slots,
+                             const float* __restrict__ grad, int m, int dim,
+                             float lr, float l1, float l2, float lr_power) {
+  APPLY_PROLOG
+  float wv = w[o];
+  float n_old = n[o];
+  float n_new = n_old + g * g;
+  float sigma = (powf(n_new, -lr_power) - powf(n_old, -lr_power)) / lr;
+  float z_new = z[o] + g - sigma * wv;
+  n[o] = n_new;
+  z[o] = z_new;
+  float quad = powf(n_new, -lr_power) / lr + 2.0f * l2;
+  float w_new = 0.0f;
+  if (fabsf(z_new) > l1)
+    w_new = ((z_new > 0.0f ? l1 : -l1) - z_new) / quad;
+  w[o] = w_new;
+  APPLY_EPILOG
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------
+
+#define CHECK_DEV(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
+static inline int n_blocks(int64_t total, int block = kBlock) {
+  int64_t b = (total + block - 1) / block;
+  // keep the chip filled but bounded; grid-stride covers the tail
+  return (int)std::min<int64_t>(b, 65535);
+}
+
+torch::Tensor ht_lookup_insert(
+    torch::Tensor keys, torch::Tensor counts, torch::Tensor ht_keys,
+    torch::Tensor ht_slot, torch::Tensor ht_freq, torch::Tensor ht_version,
+    torch::Tensor slot_counter, torch::Tensor entry_counter,
+    torch::Tensor values, torch::Tensor default_values, int64_t filter_freq,
+    int64_t step, bool train, torch::Tensor error_flag) {
+  CHECK_DEV(keys);
+  int n = keys.numel();
+  auto out = torch::empty({n}, keys.options().dtype(torch::kInt32));
+  if (n == 0) return out;
+  auto stream = current_stream();
+  int dim = values.size(1);
+  k_lookup_insert<<<n_blocks(n), kBlock, 0, stream>>>(
+      keys.data_ptr<int64_t>(),
+      counts.defined() && counts.numel() ? counts.data_ptr<int32_t>()
+                                         : nullptr,
+      n, ht_keys.data_ptr<int64_t>(), ht_slot.data_ptr<int32_t>(),
+      ht_freq.data_ptr<int32_t>(), ht_version.data_ptr<int64_t>(),
+      ht_keys.numel() - 1, slot_counter.data_ptr<int32_t>(),
+      entry_counter.data_ptr<int32_t>(), values.size(0),
+      values.data_ptr<float>(), default_values.data_ptr<float>(), dim,
+      default_values.size(0), (int)filter_freq, step, train ? 1 : 0,
+      out.data_ptr<int32_t>(), error_flag.data_ptr<int32_t>());
+  return out;
+}
+
+void ht_insert_bulk(torch::Tensor keys, torch::Tensor slots,
+                    torch::Tensor freqs, torch::Tensor versions,
+                    torch::Tensor ht_keys, torch::Tensor ht_slot,
+                    torch::Tensor ht_freq, torch::Tensor ht_version,
+                    torch::Tensor entry_counter, torch::Tensor error_flag) {
+  int n = keys.numel();
+  if (n == 0) return;
+  auto stream = current_stream();
+  k_insert_bulk<<<n_blocks(n), kBlock, 0, stream>>>(
+      keys.data_ptr<int64_t>(), slots.data_ptr<int32_t>(),
+      freqs.defined() && freqs.numel() ? freqs.data_ptr<int32_t>() : nullptr,
+      versions.defined() && versions.numel() ? versions.data_ptr<int64_t>()
+                                             : nullptr,
+      n, ht_keys.data_ptr<int64_t>(), ht_slot.data_ptr<int32_t>(),
+      ht_freq.data_ptr<int32_t>(), ht_version.data_ptr<int64_t>(),
+      ht_keys.numel() - 1, entry_counter.data_ptr<int32_t>(),
+      error_flag.data_ptr<int32_t>());
+}
+
+std::tuple<torch::Tensor, torch::Tensor> ht_lookup(torch::Tensor keys,
+                                                   torch::Tensor ht_keys,
+                                                   torch::Tensor ht_slot,
+                                                   bool want_entry) {
+  int n = keys.numel();
+  auto slots = torch::empty({n}, keys.options().dtype(torch::kInt32));
+  auto entry = want_entry
+                   ? torch::empty({n}, keys.options().dtype(torch::kInt64))
+                   : torch::Tensor();
+  if (n == 0) return {slots, entry};
+  auto stream = current_stream();
+  k_lookup<<<n_blocks(n), kBlock, 0, stream>>>(
+      keys.data_ptr<int64_t>(), n, ht_keys.data_ptr<int64_t>(),
+      ht_slot.data_ptr<int32_t>(), ht_keys.numel() - 1,
+      slots.data_ptr<int32_t>(),
+      want_entry ? entry.data_ptr<int64_t>() : nullptr);
+  return {slots, entry};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor, torch::Tensor>
+ht_export(torch::Tensor ht_keys, torch::Tensor ht_slot, torch::Tensor ht_freq,
+          torch::Tensor ht_version, int64_t n_entries) {
+  auto opts_i64 = ht_keys.options();
+  auto opts_i32 = ht_slot.options();
+  auto out_keys = torch::empty({n_entries}, opts_i64);
+  auto out_slots = torch::empty({n_entries}, opts_i32);
+  auto out_freqs = torch::empty({n_entries}, opts_i32);
+  auto out_versions = torch::empty({n_entries}, opts_i64);
+  auto cursor = torch::zeros({1}, opts_i32);
+  if (n_entries > 0) {
+    auto stream = current_stream();
+    k_export_scan<<<n_blocks(ht_keys.numel()), kBlock, 0, stream>>>(
+        ht_keys.data_ptr<int64_t>(), ht_slot.data_ptr<int32_t>(),
+        ht_freq.data_ptr<int32_t>(), ht_version.data_ptr<int64_t>(),
+        ht_keys.numel(), cursor.data_ptr<int32_t>(),
+        out_keys.data_ptr<int64_t>(), out_slots.data_ptr<int32_t>(),
+        out_freqs.data_ptr<int32_t>(), out_versions.data_ptr<int64_t>(),
+        (int)n_entries);
+  }
+  return {out_keys, out_slots, out_freqs, out_versions};
+}
+
+torch::Tensor ev_gather(torch::Tensor values, torch::Tensor default_values,
+                        torch::Tensor keys, torch::Tensor slots,
+                        double no_permission_value, bool use_no_permission,
+                        torch::ScalarType out_dtype) {
+  int m = keys.numel();
+  int dim = values.size(1);
+  auto out = torch::empty({m, dim}, values.options().dtype(out_dtype));
+  if (m == 0) return out;
+  auto stream = current_stream();
+  int64_t total = (int64_t)m * dim;
+  if (out_dtype == torch::kBFloat16) {
+    k_gather<__hip_bfloat16><<<n_blocks(total), kBlock, 0, stream>>>(
+        values.data_ptr<float>(), default_values.data_ptr<float>(),
+        keys.data_ptr<int64_t>(), slots.data_ptr<int32_t>(), m, dim,
+        default_values.size(0), (float)no_permission_value,
+        use_no_permission ? 1 : 0,
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr<at::BFloat16>()));
+  } else {
+    k_gather<float><<<n_blocks(total), kBlock, 0, stream>>>(
+        values.data_ptr<float>(), default_values.data_ptr<float>(),
+        keys.data_ptr<int64_t>(), slots.data_ptr<int32_t>(), m, dim,
+        default_values.size(0), (float)no_permission_value,
+        use_no_permission ? 1 : 0, out.data_ptr<float>());
+  }
+  return out;
+}
+
+torch::Tensor pooled_fwd(torch::Tensor values, torch::Tensor default_values,
+                         torch::Tensor keys, torch::Tensor slots,
+                         torch::Tensor inverse, torch::Tensor offsets,
+                         torch::Tensor weights, int64_t combiner,
+                         double no_permission_value, bool use_no_permission,
+                         torch::ScalarType out_dtype) {
+  int batch = offsets.numel() - 1;
+  int dim = values.size(1);
+  auto out = torch::empty({batch, dim}, values.options().dtype(out_dtype));
+  if (batch == 0) return out;
+  auto stream = current_stream();
+  int64_t total = (int64_t)batch * dim;
+  const float* wptr =
+      weights.defined() && weights.numel() ? weights.data_ptr<float>()
+                                           : nullptr;
+  if (out_dtype == torch::kBFloat16) {
+    k_pooled_fwd<__hip_bfloat16><<<n_blocks(total), kBlock, 0, stream>>>(
+        values.data_ptr<float>(), default_values.data_ptr<float>(),
+        keys.data_ptr<int64_t>(), slots.data_ptr<int32_t>(),
+        inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr, batch,
+        dim, default_values.size(0), (float)no_permission_value,
+        use_no_permission ? 1 : 0, (int)combiner,
+        reinterpret_cast<__hip_bfloat16*>(out.data_ptr<at::BFloat16>()));
+  } else {
+    k_pooled_fwd<float><<<n_blocks(total), kBlock, 0, stream>>>(
+        values.data_ptr<float>(), default_values.data_ptr<float>(),
+        keys.data_ptr<int64_t>(), slots.data_ptr<int32_t>(),
+        inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr, batch,
+        dim, default_values.size(0), (float)no_permission_value,
+        use_no_permission ? 1 : 0, (int)combiner, out.data_ptr<float>());
+  }
+  return out;
+}
+
+torch::Tensor pooled_bwd(torch::Tensor grad_out, torch::Tensor inverse,
+                         torch::Tensor offsets, torch::Tensor weights,
+                         int64_t m, int64_t combiner) {
+  int batch = offsets.numel() - 1;
+  int dim = grad_out.size(1);
+  auto grad_unique = torch::zeros(
+      {m, dim}, grad_out.options().dtype(torch::kFloat32));
+  if (batch == 0 || m == 0) return grad_unique;
+  auto stream = current_stream();
+  int64_t total = (int64_t)batch * dim;
+  const float* wptr =
+      weights.defined() && weights.numel() ? weights.data_ptr<float>()
+                                           : nullptr;
+  if (grad_out.scalar_type() == torch::kBFloat16) {
+    k_pooled_bwd<__hip_bfloat16><<<n_blocks(total), kBlock, 0, stream>>>(
+        reinterpret_cast<const __hip_bfloat16*>(
+            grad_out.data_ptr<at::BFloat16>()),
+        inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr, batch,
+        dim, (int)combiner, grad_unique.data_ptr<float>());
+  } else {
+    k_pooled_bwd<float><<<n_blocks(total), kBlock, 0, stream>>>(
+        grad_out.data_ptr<float>(), inverse.data_ptr<int32_t>(),
+        offsets.data_ptr<int32_t>(), wptr, batch, dim, (int)combiner,
+        grad_unique.data_ptr<float>());
+  }
+  return grad_unique;
+}
+
+// ---------------- sparse applies ----------------
+
+void apply_sgd(torch::Tensor w, torch::Tensor slots, torch::Tensor grad,
+               double lr) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_sgd<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), slots.data_ptr<int32_t>(), grad.data_ptr<float>(),
+      m, dim, (float)lr);
+}
+
+void apply_adagrad(torch::Tensor w, torch::Tensor accum, torch::Tensor slots,
+                   torch::Tensor grad, double lr, double epsilon) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_adagrad<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), accum.data_ptr<float>(), slots.data_ptr<int32_t>(),
+      grad.data_ptr<float>(), m, dim, (float)lr, (float)epsilon);
+}
+
+void apply_adagrad_decay(torch::Tensor w, torch::Tensor accum,
+                         torch::Tensor period_slab, torch::Tensor slots,
+                         torch::Tensor grad, double lr, double epsilon,
+                         double cur_period, double decay_rate,
+                         double baseline) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_adagrad_decay<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), accum.data_ptr<float>(),
+      period_slab.data_ptr<float>(), slots.data_ptr<int32_t>(),
+      grad.data_ptr<float>(), m, dim, (float)lr, (float)epsilon,
+      (float)cur_period, (float)decay_rate, (float)baseline);
+  k_commit_period<<<n_blocks(m), kBlock, 0, stream>>>(
+      period_slab.data_ptr<float>(), slots.data_ptr<int32_t>(), m,
+      (float)cur_period);
+}
+
+void apply_adam(torch::Tensor w, torch::Tensor mom, torch::Tensor vel,
+                torch::Tensor slots, torch::Tensor grad, double lr_t,
+                double beta1, double beta2, double epsilon) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_adam<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), mom.data_ptr<float>(), vel.data_ptr<float>(),
+      slots.data_ptr<int32_t>(), grad.data_ptr<float>(), m, dim, (float)lr_t,
+      (float)beta1, (float)beta2, (float)epsilon);
+}
+
+void apply_adamw(torch::Tensor w, torch::Tensor mom, torch::Tensor vel,
+                 torch::Tensor slots, torch::Tensor grad, double lr_t,
+                 double lr, double beta1, double beta2, double epsilon,
+                 double weight_decay) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_adamw<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), mom.data_ptr<float>(), vel.data_ptr<float>(),
+      slots.data_ptr<int32_t>(), grad.data_ptr<float>(), m, dim, (float)lr_t,
+      (float)lr, (float)beta1, (float)beta2, (float)epsilon,
+      (float)weight_decay);
+}
+
+void apply_rmsprop(torch::Tensor w, torch::Tensor vel, torch::Tensor slots,
+                   torch::Tensor grad, double lr, double beta2,
+                   double epsilon) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_rmsprop<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), vel.data_ptr<float>(), slots.data_ptr<int32_t>(),
+      grad.data_ptr<float>(), m, dim, (float)lr, (float)beta2,
+      (float)epsilon);
+}
+
+void apply_ftrl(torch::Tensor w, torch::Tensor n, torch::Tensor z,
+                torch::Tensor slots, torch::Tensor grad, double lr, double l1,
+                double l2, double lr_power) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_ftrl<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), n.data_ptr<float>(), z.data_ptr<float>(),
+      slots.data_ptr<int32_t>(), grad.data_ptr<float>(), m, dim, (float)lr,
+      (float)l1, (float)l2, (float)lr_power);
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  mod.def("ht_lookup_insert", &ht_lookup_insert);
+  mod.def("ht_insert_bulk", &ht_insert_bulk);
+  mod.def("ht_lookup", &ht_lookup);
+  mod.def("ht_export", &ht_export);
+  mod.def("ev_gather", &ev_gather);
+  mod.def("pooled_fwd", &pooled_fwd);
+  mod.def("pooled_bwd", &pooled_bwd);
+  mod.def("apply_sgd", &apply_sgd);
+  mod.def("apply_adagrad", &apply_adagrad);
+  mod.def("apply_adagrad_decay", &apply_adagrad_decay);
+  mod.def("apply_adam", &apply_adam);
+  mod.def("apply_adamw", &apply_adamw);
+  mod.def("apply_rmsprop", &apply_rmsprop);
+  mod.def("apply_ftrl", &apply_ftrl);
+  mod.attr("EMPTY_KEY") = EMPTY_KEY;
+}

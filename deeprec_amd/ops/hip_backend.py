@@ -1,0 +1,404 @@
+"""HBM (GPU) storage backend: Python wrapper over the HIP embedding engine.
+
+Same interface/semantics as ops/cpu_backend.CpuStorage; all hot-path work
+runs in the hand-written gfx950 kernels (ops/hip/ev_kernels.hip). Growth
+and shrink/compaction are host-coordinated (rare, off the hot path) —
+mirroring the reference's coarse multi-tier design choice (SURVEY.md §7
+hard parts: host-coordinated rehash).
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+
+from deeprec_amd.embedding.options import (
+    CBFFilter,
+    CounterFilter,
+    EmbeddingVariableOption,
+    GlobalStepEvict,
+    L2WeightEvict,
+)
+from deeprec_amd.ops.build_ext import require_extension
+
+_COMBINER_ID = {"sum": 0, "mean": 1, "sqrtn": 2}
+_LOAD_FACTOR = 0.6
+
+
+def _pow2(n: int) -> int:
+    return 1 << max(10, (n - 1).bit_length())
+
+
+class HbmStorage:
+    """Single-tier HBM KV storage (reference capability: HbmStorage,
+    single_tier_storage.h:387 + GPUHashMapKV, gpu_hash_map_kv.h:29)."""
+
+    def __init__(self, dim: int, ev_option: EmbeddingVariableOption,
+                 value_dtype=torch.float32, device=None, generator=None):
+        assert value_dtype == torch.float32, \
+            "master values are fp32; bf16 is produced by the fused gather"
+        self.ext = require_extension()
+        self.dim = dim
+        self.device = torch.device(device or "cuda")
+        self.value_dtype = value_dtype
+        self.ev_option = ev_option
+
+        io = ev_option.init_option
+        self.default_value_dim = max(1, io.default_value_dim)
+        dv = torch.empty(self.default_value_dim, dim, dtype=torch.float32)
+        if io.initializer is None:
+            dv.normal_(0.0, 1.0 / math.sqrt(dim), generator=generator)
+        elif callable(io.initializer):
+            io.initializer(dv)
+        else:
+            dv.fill_(float(io.initializer))
+        self.default_values = dv.to(self.device)
+
+        fo = ev_option.filter_option
+        self.filter_freq = 0
+        if isinstance(fo, (CounterFilter, CBFFilter)):
+            # CBF semantics on GPU: per-entry counters are cheap in 288 GB
+            # HBM, so the CBF option maps to exact counter admission. The
+            # memory-bounded probabilistic pre-admission store is a CPU-tier
+            # concern (cpu_backend implements a true counting bloom).
+            self.filter_freq = fo.filter_freq
+
+        cap = _pow2(ev_option.init_capacity)
+        self._alloc_table(cap)
+        self._alloc_slabs(max(1024, cap // 2))
+        self.slabs = {}
+        self._slab_init = {}
+        self.slot_counter = torch.zeros(1, dtype=torch.int32,
+                                        device=self.device)
+        self.entry_counter = torch.zeros(1, dtype=torch.int32,
+                                         device=self.device)
+        self.error_flag = torch.zeros(1, dtype=torch.int32,
+                                      device=self.device)
+        # host-side mirrors, refreshed lazily (avoid per-step D2H sync)
+        self._entries_hint = 0
+        self._slots_hint = 0
+
+    # ---------------- allocation ----------------
+    def _alloc_table(self, capacity: int):
+        self.capacity = capacity
+        self.ht_keys = torch.full((capacity,), -(2 ** 63), dtype=torch.int64,
+                                  device=self.device)
+        self.ht_slot = torch.full((capacity,), -1, dtype=torch.int32,
+                                  device=self.device)
+        self.ht_freq = torch.zeros(capacity, dtype=torch.int32,
+                                   device=self.device)
+        self.ht_version = torch.full((capacity,), -1, dtype=torch.int64,
+                                     device=self.device)
+
+    def _alloc_slabs(self, max_slots: int):
+        self.values = torch.empty(max_slots, self.dim, dtype=torch.float32,
+                                  device=self.device)
+
+    @property
+    def max_slots(self) -> int:
+        return self.values.shape[0]
+
+    def _sync_counters(self):
+        c = torch.stack([self.slot_counter, self.entry_counter]).cpu()
+        self._slots_hint = int(c[0])
+        self._entries_hint = int(c[1])
+
+    def _ensure_capacity(self, incoming: int):
+        """Grow hash table / value slab before a batch that could overflow.
+        Uses host-side hints (upper bounds) so the hot path stays sync-free;
+        hints are refreshed only when a growth decision is near."""
+        if (self._entries_hint + incoming > self.capacity * _LOAD_FACTOR or
+                self._slots_hint + incoming > self.max_slots):
+            self._sync_counters()
+            if self._entries_hint + incoming > self.capacity * _LOAD_FACTOR:
+                self._rehash(_pow2(
+                    int((self._entries_hint + incoming) / _LOAD_FACTOR) + 1))
+            if self._slots_hint + incoming > self.max_slots:
+                self._grow_slots(self._slots_hint + incoming)
+        # track the upper bound without a device sync
+        self._entries_hint += incoming
+        self._slots_hint += incoming
+
+    def _grow_slots(self, need: int):
+        new_cap = self.max_slots
+        while new_cap < need:
+            new_cap *= 2
+        nv = torch.empty(new_cap, self.dim, dtype=torch.float32,
+                         device=self.device)
+        nv[: self.max_slots] = self.values
+        for name, t in list(self.slabs.items()):
+            nt = torch.full((new_cap, t.shape[1]), self._slab_init[name],
+                            dtype=t.dtype, device=self.device)
+            nt[: t.shape[0]] = t
+            self.slabs[name] = nt
+        self.values = nv
+
+    def _rehash(self, new_capacity: int):
+        keys, slots, freqs, versions = self._export_entries()
+        self._alloc_table(new_capacity)
+        self.entry_counter.zero_()
+        self.ext.ht_insert_bulk(keys, slots, freqs, versions, self.ht_keys,
+                                self.ht_slot, self.ht_freq, self.ht_version,
+                                self.entry_counter, self.error_flag)
+        self._check_error()
+
+    def _export_entries(self):
+        n = int(self.entry_counter.cpu())
+        return self.ext.ht_export(self.ht_keys, self.ht_slot, self.ht_freq,
+                                  self.ht_version, n)
+
+    def _check_error(self):
+        err = int(self.error_flag.cpu())
+        if err != 0:
+            raise RuntimeError(
+                f"HBM hash table error {err} (1=slab overflow, 2=table full)"
+                " — capacity management bug")
+
+    # ---------------- public interface ----------------
+    def get_slab(self, name, width, init_value, dtype=torch.float32):
+        if name not in self.slabs:
+            self.slabs[name] = torch.full(
+                (self.max_slots, width), init_value, dtype=dtype,
+                device=self.device)
+            self._slab_init[name] = init_value
+        return self.slabs[name]
+
+    def lookup_or_create(self, keys, counts, step, train=True):
+        if train:
+            self._ensure_capacity(keys.numel())
+        counts_i32 = (counts.to(torch.int32)
+                      if counts is not None else torch.Tensor())
+        return self.ext.ht_lookup_insert(
+            keys, counts_i32, self.ht_keys, self.ht_slot, self.ht_freq,
+            self.ht_version, self.slot_counter, self.entry_counter,
+            self.values, self.default_values, self.filter_freq, step, train,
+            self.error_flag)
+
+    def lookup(self, keys):
+        slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)
+        return slots
+
+    def _use_no_permission(self) -> bool:
+        return (self.filter_freq > 0 and
+                self.ev_option.init_option.default_value_no_permission
+                is not None)
+
+    def _no_permission_value(self) -> float:
+        v = self.ev_option.init_option.default_value_no_permission
+        return float(v) if v is not None else 0.0
+
+    def gather(self, keys, slots, out_dtype=None):
+        return self.ext.ev_gather(
+            self.values, self.default_values, keys, slots,
+            self._no_permission_value(), self._use_no_permission(),
+            out_dtype or torch.float32)
+
+    def pooled_lookup(self, keys, slots, inverse, offsets, row_ids, combiner,
+                      weights, out_dtype):
+        return self.ext.pooled_fwd(
+            self.values, self.default_values, keys, slots,
+            inverse.to(torch.int32), offsets.to(torch.int32),
+            weights.float() if weights is not None else torch.Tensor(),
+            _COMBINER_ID[combiner], self._no_permission_value(),
+            self._use_no_permission(), out_dtype or torch.float32)
+
+    def pooled_grad(self, grad_out, inverse, offsets, row_ids, m, combiner,
+                    weights):
+        return self.ext.pooled_bwd(
+            grad_out.contiguous(), inverse.to(torch.int32),
+            offsets.to(torch.int32),
+            weights.float() if weights is not None else torch.Tensor(),
+            m, _COMBINER_ID[combiner])
+
+    # ---------------- shrink / export / import ----------------
+    def shrink(self, step: int) -> int:
+        eo = self.ev_option.evict_option
+        if eo is None:
+            return 0
+        keys, slots, freqs, versions = self._export_entries()
+        if keys.numel() == 0:
+            return 0
+        if isinstance(eo, GlobalStepEvict) and eo.steps_to_live > 0:
+            keep = versions >= (step - eo.steps_to_live)
+        elif isinstance(eo, L2WeightEvict) and eo.l2_weight_threshold > 0:
+            keep = torch.ones(keys.numel(), dtype=torch.bool,
+                              device=self.device)
+            has = slots >= 0
+            norms = self.values[slots[has].long()].norm(dim=1)
+            kk = keep[has].clone()
+            kk &= norms >= eo.l2_weight_threshold
+            keep[has] = kk
+        else:
+            return 0
+        n_evicted = int((~keep).sum())
+        if n_evicted == 0:
+            return 0
+        self._rebuild(keys[keep], slots[keep], freqs[keep], versions[keep])
+        return n_evicted
+
+    def _rebuild(self, keys, slots, freqs, versions):
+        """Compact: renumber admitted slots densely, rewrite slabs/table."""
+        adm = slots >= 0
+        old_slots = slots[adm].long()
+        n_adm = int(adm.sum())
+        new_slots = torch.full_like(slots, -1)
+        new_slots[adm] = torch.arange(n_adm, dtype=torch.int32,
+                                      device=self.device)
+        self.values[:n_adm] = self.values[old_slots]
+        for name, t in self.slabs.items():
+            t[:n_adm] = t[old_slots]
+        self._alloc_table(self.capacity)
+        self.entry_counter.zero_()
+        self.slot_counter.fill_(n_adm)
+        self.ext.ht_insert_bulk(keys, new_slots, freqs, versions,
+                                self.ht_keys, self.ht_slot, self.ht_freq,
+                                self.ht_version, self.entry_counter,
+                                self.error_flag)
+        self._check_error()
+        self._sync_counters()
+
+    def export(self, include_filtered: bool = False):
+        keys, slots, freqs, versions = self._export_entries()
+        adm = slots >= 0
+        out = (keys[adm], self.values[slots[adm].long()].clone(),
+               freqs[adm].to(torch.int64), versions[adm])
+        if include_filtered:
+            out = out + (keys[~adm], freqs[~adm].to(torch.int64))
+        return out
+
+    def export_slabs(self, names):
+        keys, slots, freqs, versions = self._export_entries()
+        adm = slots >= 0
+        s = slots[adm].long()
+        return [self.slabs[nm][s].clone() for nm in names]
+
+    def import_(self, keys, values, freqs=None, versions=None,
+                slab_rows=None):
+        m = keys.numel()
+        if m == 0:
+            return
+        keys = keys.to(self.device)
+        values = values.to(self.device, torch.float32)
+        # existing keys keep their slots; new keys get fresh ones
+        existing = self.lookup(keys)
+        is_new = existing < 0
+        n_new = int(is_new.sum())
+        self._ensure_capacity(m)
+        self._sync_counters()
+        base = self._slots_hint
+        slots = existing.clone()
+        slots[is_new] = base + torch.arange(n_new, dtype=torch.int32,
+                                            device=self.device)
+        self.slot_counter.fill_(base + n_new)
+        self._slots_hint = base + n_new
+        self.ext.ht_insert_bulk(
+            keys, slots,
+            freqs.to(self.device, torch.int32) if freqs is not None
+            else torch.Tensor(),
+            versions.to(self.device, torch.int64) if versions is not None
+            else torch.Tensor(),
+            self.ht_keys, self.ht_slot, self.ht_freq, self.ht_version,
+            self.entry_counter, self.error_flag)
+        self.values[slots.long()] = values
+        if slab_rows is not None:
+            for name, rows in slab_rows.items():
+                self.get_slab(name, rows.shape[1], 0.0)[slots.long()] = \
+                    rows.to(self.device)
+        self._check_error()
+        self._sync_counters()
+
+    def size(self) -> int:
+        _, slots, _, _ = self._export_entries()
+        return int((slots >= 0).sum())
+
+    def total_count(self) -> int:
+        return int(self.entry_counter.cpu())
+
+    def frequencies(self, keys):
+        keys = keys.to(self.device)
+        slots, entry = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot,
+                                          True)
+        out = torch.zeros(keys.numel(), dtype=torch.int64, device=self.device)
+        found = entry >= 0
+        out[found] = self.ht_freq[entry[found]].to(torch.int64)
+        return out
+
+    def versions(self, keys):
+        keys = keys.to(self.device)
+        slots, entry = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot,
+                                          True)
+        out = torch.full((keys.numel(),), -1, dtype=torch.int64,
+                         device=self.device)
+        found = entry >= 0
+        out[found] = self.ht_version[entry[found]]
+        return out
+
+
+# ---------------- optimizer dispatch ----------------
+
+def sparse_apply(name: str, storage: HbmStorage, slots, grad, hyper: dict):
+    ext = storage.ext
+    grad = grad.float().contiguous()
+    slots = slots.to(torch.int32)
+    w = storage.values
+    d = storage.dim
+    if name == "sgd":
+        ext.apply_sgd(w, slots, grad, hyper["lr"])
+    elif name == "adagrad":
+        accum = storage.get_slab("adagrad_accum", d,
+                                 hyper.get("initial_accumulator", 0.1))
+        ext.apply_adagrad(w, accum, slots, grad, hyper["lr"],
+                          hyper.get("epsilon", 0.0))
+    elif name == "adagrad_decay":
+        accum = storage.get_slab("adagrad_accum", d,
+                                 hyper.get("initial_accumulator", 0.1))
+        period = storage.get_slab("adagrad_decay_period", 1, 0.0)
+        cur_period = float(hyper["global_step"]
+                           // max(1, hyper["accumulator_decay_step"]))
+        ext.apply_adagrad_decay(
+            w, accum, period, slots, grad, hyper["lr"],
+            hyper.get("epsilon", 0.0), cur_period,
+            hyper["accumulator_decay_rate"], hyper["accumulator_baseline"])
+    elif name in ("adam", "adamw"):
+        import math as _m
+        mom = storage.get_slab("adam_m", d, 0.0)
+        vel = storage.get_slab("adam_v", d, 0.0)
+        b1, b2 = hyper["beta1"], hyper["beta2"]
+        t = hyper["step_t"]
+        lr_t = hyper["lr"] * _m.sqrt(1 - b2 ** t) / (1 - b1 ** t)
+        if name == "adam":
+            ext.apply_adam(w, mom, vel, slots, grad, lr_t, b1, b2,
+                           hyper["epsilon"])
+        else:
+            ext.apply_adamw(w, mom, vel, slots, grad, lr_t, hyper["lr"],
+                            b1, b2, hyper["epsilon"], hyper["weight_decay"])
+    elif name == "adam_async":
+        import math as _m
+        b1, b2 = hyper["beta1"], hyper["beta2"]
+        if hyper.get("sparse_rmsprop"):
+            vel = storage.get_slab("adam_v", d, 0.0)
+            ext.apply_rmsprop(w, vel, slots, grad, hyper["lr"], b2,
+                              hyper["epsilon"])
+        else:
+            mom = storage.get_slab("adam_m", d, 0.0)
+            vel = storage.get_slab("adam_v", d, 0.0)
+            lr_t = hyper["lr"] * _m.sqrt(1 - hyper["beta2_power"]) / \
+                (1 - hyper["beta1_power"])
+            ext.apply_adam(w, mom, vel, slots, grad, lr_t, b1, b2,
+                           hyper["epsilon"])
+    elif name == "ftrl":
+        n = storage.get_slab("ftrl_accum", d, 0.1)
+        z = storage.get_slab("ftrl_linear", d, 0.0)
+        ext.apply_ftrl(w, n, z, slots, grad, hyper["lr"], hyper["l1"],
+                       hyper["l2"], hyper["lr_power"])
+    else:
+        raise ValueError(f"unknown sparse optimizer {name!r}")
+
+
+def group_pooled_lookup(evs, sp_ids_list, combiners, out_dtype):
+    """Grouped lookup fast path — per-table loop for now; the fused
+    multi-table kernel lands with the EmbeddingCollection."""
+    from deeprec_amd.embedding.lookup import embedding_lookup_sparse
+    return [embedding_lookup_sparse(ev, sp, c, out_dtype)
+            for ev, sp, c in zip(evs, sp_ids_list, combiners)]
