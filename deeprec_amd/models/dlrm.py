@@ -37,7 +37,7 @@ class DLRM(nn.Module):
                  interaction_op: str = "dot", bf16: bool = True,
                  device="cpu", ev_option: Optional[EmbeddingVariableOption] = None,
                  num_sparse: int = NUM_SPARSE, name_prefix: str = "dlrm",
-                 sharded: bool = False):
+                 sharded: bool = False, use_collection: bool = True):
         super().__init__()
         assert mlp_bot[-1] == embedding_dim, \
             "bottom MLP output must match embedding dim (dot interaction)"
@@ -56,6 +56,7 @@ class DLRM(nn.Module):
                             final_activation=False)
         self.to(self.device_)
 
+        self.collection = None
         if sharded:
             from deeprec_amd.parallel.sharded_embedding import (
                 ShardedEmbeddingVariable)
@@ -65,6 +66,14 @@ class DLRM(nn.Module):
                                          ev_option=ev_option,
                                          device=self.device_)
                 for i in range(num_sparse)]
+        elif use_collection:
+            from deeprec_amd.embedding.collection import EmbeddingCollection
+            self.collection = EmbeddingCollection(
+                f"{name_prefix}/sparse",
+                [f"C{i+1}" for i in range(num_sparse)], embedding_dim,
+                ev_option=ev_option, combiners=["mean"] * num_sparse,
+                device=self.device_)
+            self.evs = []
         else:
             self.evs = [
                 EmbeddingVariable(f"{name_prefix}/C{i+1}", embedding_dim,
@@ -72,7 +81,7 @@ class DLRM(nn.Module):
                 for i in range(num_sparse)]
 
     def embedding_variables(self):
-        return self.evs
+        return [self.collection] if self.collection is not None else self.evs
 
     def _interact(self, feats: torch.Tensor) -> torch.Tensor:
         """feats: [B, F, D] -> pairwise dots, upper triangle (i<j)
@@ -86,15 +95,23 @@ class DLRM(nn.Module):
                 train: bool = True) -> torch.Tensor:
         compute_dtype = torch.bfloat16 if (
             self.bf16 and dense.device.type == "cuda") else torch.float32
-        emb_list = group_embedding_lookup_sparse(
-            self.evs, sparse_ids, combiners=["mean"] * len(self.evs),
-            out_dtype=compute_dtype, train=train)
+        if self.collection is not None:
+            emb_cat = self.collection.lookup(sparse_ids,
+                                             out_dtype=compute_dtype,
+                                             train=train)
+            emb_feats = emb_cat.view(-1, self.num_sparse, self.embedding_dim)
+        else:
+            emb_list = group_embedding_lookup_sparse(
+                self.evs, sparse_ids, combiners=["mean"] * len(self.evs),
+                out_dtype=compute_dtype, train=train)
+            emb_feats = torch.stack(list(emb_list), dim=1)
         import contextlib
         amp = (torch.autocast(device_type="cuda", dtype=torch.bfloat16)
                if compute_dtype == torch.bfloat16 else contextlib.nullcontext())
         with amp:
             bot = self.mlp_bot(dense)
-            feats = torch.stack([bot] + list(emb_list), dim=1)  # [B, F+1, D]
+            feats = torch.cat([bot.unsqueeze(1), emb_feats.to(bot.dtype)],
+                              dim=1)  # [B, F+1, D]
             if self.interaction_op == "dot":
                 inter = self._interact(feats)
                 top_in = torch.cat([bot, inter], dim=1)

@@ -70,6 +70,9 @@ class CpuStorage:
 
         io = ev_option.init_option
         self.default_value_dim = max(1, io.default_value_dim)
+        # composite-key support (EmbeddingCollection): 0 = plain keys
+        self.key_bits = 0
+        self.dvd_per_table = self.default_value_dim
         self.default_values = torch.empty(
             self.default_value_dim, dim, dtype=value_dtype)
         if io.initializer is None:
@@ -154,6 +157,21 @@ class CpuStorage:
             self.slot_count += 1
         return s
 
+    def _default_row(self, k: int) -> int:
+        if self.key_bits > 0:
+            mask = (1 << self.key_bits) - 1
+            return ((k >> self.key_bits) * self.dvd_per_table
+                    + (k & mask) % self.dvd_per_table)
+        return k % self.default_value_dim
+
+    def _default_rows(self, keys):
+        import torch as _t
+        if self.key_bits > 0:
+            mask = (1 << self.key_bits) - 1
+            return ((keys >> self.key_bits) * self.dvd_per_table
+                    + (keys & mask) % self.dvd_per_table)
+        return (keys % self.default_value_dim).clamp(min=0)
+
     # ---------------- public interface (mirrors HbmStorage) ----------------
     def get_slab(self, name: str, width: int, init_value: float,
                  dtype=torch.float32) -> torch.Tensor:
@@ -212,7 +230,7 @@ class CpuStorage:
             if s < 0 and train and int(self.entry_freq[e]) >= self.filter_freq:
                 s = self._alloc_slot()
                 self.entry_slot[e] = s
-                self.values[s] = self.default_values[k % self.default_value_dim]
+                self.values[s] = self.default_values[self._default_row(k)]
                 for name, t in self.slabs.items():
                     t[s].fill_(self._slab_init[name])
             slots[i] = s
@@ -227,8 +245,7 @@ class CpuStorage:
         """values[slots] with default-value fill for slot<0."""
         out_dtype = out_dtype or self.value_dtype
         admitted = slots >= 0
-        out = self.default_values[
-            (keys % self.default_value_dim).clamp(min=0)].clone()
+        out = self.default_values[self._default_rows(keys)].clone()
         if admitted.any():
             out[admitted] = self.values[slots[admitted]]
         if self.ev_option.init_option.default_value_no_permission is not None \

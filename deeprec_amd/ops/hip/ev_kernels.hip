@@ -72,8 +72,8 @@ __global__ void k_lookup_insert(
     int64_t cap_mask, int32_t* __restrict__ slot_counter,
     int32_t* __restrict__ entry_counter, int max_slots,
     float* __restrict__ values, const float* __restrict__ default_values,
-    int dim, int default_value_dim, int filter_freq, int64_t step,
-    int train, int32_t* __restrict__ out_slots,
+    int dim, int default_value_dim, int key_bits, int filter_freq,
+    int64_t step, int train, int32_t* __restrict__ out_slots,
     int32_t* __restrict__ error_flag) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
@@ -108,9 +108,18 @@ __global__ void k_lookup_insert(
         return;
       }
       ht_slot[idx] = slot;
-      const float* src =
-          default_values +
-          (int64_t)((uint64_t)key % (uint64_t)default_value_dim) * dim;
+      // composite keys (EmbeddingCollection): default row is
+      // table * dvd + (raw_key % dvd); key_bits == 0 means plain keys
+      int64_t dvrow;
+      if (key_bits > 0) {
+        int64_t mask = ((int64_t)1 << key_bits) - 1;
+        dvrow = (key >> key_bits) * default_value_dim +
+                (int64_t)((uint64_t)(key & mask) %
+                          (uint64_t)default_value_dim);
+      } else {
+        dvrow = (int64_t)((uint64_t)key % (uint64_t)default_value_dim);
+      }
+      const float* src = default_values + dvrow * dim;
       float* dst = values + (int64_t)slot * dim;
       for (int d = 0; d < dim; ++d) dst[d] = src[d];
     }
@@ -329,6 +338,110 @@ __global__ void k_pooled_bwd(
 }
 
 // ---------------------------------------------------------------------
+// grouped (multi-table) fused pooling — the EmbeddingCollection hot path
+// ---------------------------------------------------------------------
+//
+// N tables of equal dim share one storage via composite keys
+// (table_id << key_bits) | id. One unique + one probe + one fused kernel
+// per step regardless of table count (reference capability:
+// GroupEmbeddingVarLookup, ops/kv_variable_ops.cc:404, re-designed so the
+// whole group is ONE launch instead of one block-set per table).
+//
+// Ragged layout: the N tables' ragged batches are concatenated —
+// offsets[(N*B)+1]; pooled row r (< N*B) is table t = r / B, sample
+// b = r % B. Output is written interleaved as out[b, t*D + d] giving the
+// [B, N*D] concat layout models consume with no extra copy.
+
+template <typename OutT>
+__global__ void k_group_pooled_fwd(
+    const float* __restrict__ values, const float* __restrict__ default_values,
+    const int64_t* __restrict__ keys, const int32_t* __restrict__ slots,
+    const int32_t* __restrict__ inverse, const int32_t* __restrict__ offsets,
+    const float* __restrict__ weights,
+    const int32_t* __restrict__ combiner_ids, int batch, int n_tables,
+    int dim, int default_value_dim, int key_bits, float no_permission_value,
+    int use_no_permission, OutT* __restrict__ out) {
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)batch * n_tables * dim;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  const int64_t key_mask = ((int64_t)1 << key_bits) - 1;
+  for (; t < total; t += stride) {
+    int d = (int)(t % dim);
+    int64_t rid = t / dim;           // pooled row in [0, N*B)
+    int table = (int)(rid / batch);
+    int b = (int)(rid % batch);
+    int beg = offsets[rid], end = offsets[rid + 1];
+    int combiner = combiner_ids[table];
+    float acc = 0.0f, wacc = 0.0f;
+    for (int j = beg; j < end; ++j) {
+      int u = inverse[j];
+      int32_t s = slots[u];
+      float v;
+      if (s >= 0) {
+        v = values[(int64_t)s * dim + d];
+      } else if (use_no_permission) {
+        v = no_permission_value;
+      } else {
+        int64_t k = keys[u];
+        int64_t row = (int64_t)(table)*default_value_dim +
+                      (int64_t)((uint64_t)(k & key_mask) %
+                                (uint64_t)default_value_dim);
+        v = default_values[row * dim + d];
+      }
+      float w = weights ? weights[j] : 1.0f;
+      acc += w * v;
+      wacc += (combiner == 2) ? w * w : w;
+    }
+    float coeff = 1.0f;
+    if (combiner != 0 && end > beg) {
+      float denom = (combiner == 2) ? sqrtf(wacc) : wacc;
+      coeff = denom > 1e-12f ? 1.0f / denom : 0.0f;
+    }
+    float r = acc * coeff;
+    int64_t oidx = ((int64_t)b * n_tables + table) * dim + d;
+    if constexpr (std::is_same_v<OutT, __hip_bfloat16>) out[oidx] = f2bf(r);
+    else out[oidx] = r;
+  }
+}
+
+// Atomic-free grouped backward: occurrences are CSR-grouped by unique key
+// (order = argsort(inverse), bounds = cumsum(counts)); thread (u, d) sums
+// its occurrences and writes grad_unique[u][d] once. row_coeff[N*B] is the
+// per-pooled-row combiner coefficient precomputed on device.
+template <typename GradT>
+__global__ void k_group_pooled_bwd(
+    const GradT* __restrict__ grad_out,  // [B, N*D]
+    const int32_t* __restrict__ order, const int32_t* __restrict__ bounds,
+    const int32_t* __restrict__ row_ids,  // [nnz] pooled row (t*B + b)
+    const float* __restrict__ weights, const float* __restrict__ row_coeff,
+    int m, int batch, int n_tables, int dim,
+    float* __restrict__ grad_unique) {
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)m * dim;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; t < total; t += stride) {
+    int u = (int)(t / dim);
+    int d = (int)(t % dim);
+    float acc = 0.0f;
+    for (int k = bounds[u]; k < bounds[u + 1]; ++k) {
+      int j = order[k];
+      int rid = row_ids[j];
+      int table = rid / batch;
+      int b = rid % batch;
+      float g;
+      int64_t gidx = ((int64_t)b * n_tables + table) * dim + d;
+      if constexpr (std::is_same_v<GradT, __hip_bfloat16>)
+        g = bf2f(grad_out[gidx]);
+      else
+        g = grad_out[gidx];
+      float w = weights ? weights[j] : 1.0f;
+      acc += w * row_coeff[rid] * g;
+    }
+    grad_unique[t] = acc;
+  }
+}
+
+// ---------------------------------------------------------------------
 // fused sparse optimizer applies (thread = (unique key i, dim d))
 // ---------------------------------------------------------------------
 
@@ -482,8 +595,9 @@ torch::Tensor ht_lookup_insert(
     torch::Tensor keys, torch::Tensor counts, torch::Tensor ht_keys,
     torch::Tensor ht_slot, torch::Tensor ht_freq, torch::Tensor ht_version,
     torch::Tensor slot_counter, torch::Tensor entry_counter,
-    torch::Tensor values, torch::Tensor default_values, int64_t filter_freq,
-    int64_t step, bool train, torch::Tensor error_flag) {
+    torch::Tensor values, torch::Tensor default_values, int64_t dvd_per_table,
+    int64_t key_bits, int64_t filter_freq, int64_t step, bool train,
+    torch::Tensor error_flag) {
   CHECK_DEV(keys);
   int n = keys.numel();
   auto out = torch::empty({n}, keys.options().dtype(torch::kInt32));
@@ -499,8 +613,9 @@ torch::Tensor ht_lookup_insert(
       ht_keys.numel() - 1, slot_counter.data_ptr<int32_t>(),
       entry_counter.data_ptr<int32_t>(), values.size(0),
       values.data_ptr<float>(), default_values.data_ptr<float>(), dim,
-      default_values.size(0), (int)filter_freq, step, train ? 1 : 0,
-      out.data_ptr<int32_t>(), error_flag.data_ptr<int32_t>());
+      (int)dvd_per_table, (int)key_bits, (int)filter_freq, step,
+      train ? 1 : 0, out.data_ptr<int32_t>(),
+      error_flag.data_ptr<int32_t>());
   return out;
 }
 
@@ -654,6 +769,76 @@ torch::Tensor pooled_bwd(torch::Tensor grad_out, torch::Tensor inverse,
   return grad_unique;
 }
 
+torch::Tensor group_pooled_fwd(
+    torch::Tensor values, torch::Tensor default_values, torch::Tensor keys,
+    torch::Tensor slots, torch::Tensor inverse, torch::Tensor offsets,
+    torch::Tensor weights, torch::Tensor combiner_ids, int64_t batch,
+    int64_t n_tables, int64_t key_bits, double no_permission_value,
+    bool use_no_permission, torch::ScalarType out_dtype) {
+  int dim = values.size(1);
+  auto out = torch::empty({batch, n_tables * dim},
+                          values.options().dtype(out_dtype));
+  int64_t total = batch * n_tables * dim;
+  if (total == 0) return out;
+  auto stream = current_stream();
+  const float* wptr =
+      weights.defined() && weights.numel() ? weights.data_ptr<float>()
+                                           : nullptr;
+  if (out_dtype == torch::kBFloat16) {
+    k_group_pooled_fwd<__hip_bfloat16>
+        <<<n_blocks(total), kBlock, 0, stream>>>(
+            values.data_ptr<float>(), default_values.data_ptr<float>(),
+            keys.data_ptr<int64_t>(), slots.data_ptr<int32_t>(),
+            inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr,
+            combiner_ids.data_ptr<int32_t>(), (int)batch, (int)n_tables, dim,
+            default_values.size(0) / (int)n_tables, (int)key_bits,
+            (float)no_permission_value, use_no_permission ? 1 : 0,
+            reinterpret_cast<__hip_bfloat16*>(out.data_ptr<at::BFloat16>()));
+  } else {
+    k_group_pooled_fwd<float><<<n_blocks(total), kBlock, 0, stream>>>(
+        values.data_ptr<float>(), default_values.data_ptr<float>(),
+        keys.data_ptr<int64_t>(), slots.data_ptr<int32_t>(),
+        inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr,
+        combiner_ids.data_ptr<int32_t>(), (int)batch, (int)n_tables, dim,
+        default_values.size(0) / (int)n_tables, (int)key_bits,
+        (float)no_permission_value, use_no_permission ? 1 : 0,
+        out.data_ptr<float>());
+  }
+  return out;
+}
+
+torch::Tensor group_pooled_bwd(torch::Tensor grad_out, torch::Tensor order,
+                               torch::Tensor bounds, torch::Tensor row_ids,
+                               torch::Tensor weights, torch::Tensor row_coeff,
+                               int64_t m, int64_t batch, int64_t n_tables,
+                               int64_t dim) {
+  auto grad_unique = torch::empty(
+      {m, dim}, grad_out.options().dtype(torch::kFloat32));
+  int64_t total = m * dim;
+  if (total == 0) return grad_unique;
+  auto stream = current_stream();
+  const float* wptr =
+      weights.defined() && weights.numel() ? weights.data_ptr<float>()
+                                           : nullptr;
+  if (grad_out.scalar_type() == torch::kBFloat16) {
+    k_group_pooled_bwd<__hip_bfloat16>
+        <<<n_blocks(total), kBlock, 0, stream>>>(
+            reinterpret_cast<const __hip_bfloat16*>(
+                grad_out.data_ptr<at::BFloat16>()),
+            order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
+            row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
+            (int)m, (int)batch, (int)n_tables, (int)dim,
+            grad_unique.data_ptr<float>());
+  } else {
+    k_group_pooled_bwd<float><<<n_blocks(total), kBlock, 0, stream>>>(
+        grad_out.data_ptr<float>(), order.data_ptr<int32_t>(),
+        bounds.data_ptr<int32_t>(), row_ids.data_ptr<int32_t>(), wptr,
+        row_coeff.data_ptr<float>(), (int)m, (int)batch, (int)n_tables,
+        (int)dim, grad_unique.data_ptr<float>());
+  }
+  return grad_unique;
+}
+
 // ---------------- sparse applies ----------------
 
 void apply_sgd(torch::Tensor w, torch::Tensor slots, torch::Tensor grad,
@@ -759,6 +944,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ev_gather", &ev_gather);
   mod.def("pooled_fwd", &pooled_fwd);
   mod.def("pooled_bwd", &pooled_bwd);
+  mod.def("group_pooled_fwd", &group_pooled_fwd);
+  mod.def("group_pooled_bwd", &group_pooled_bwd);
   mod.def("apply_sgd", &apply_sgd);
   mod.def("apply_adagrad", &apply_adagrad);
   mod.def("apply_adagrad_decay", &apply_adagrad_decay);

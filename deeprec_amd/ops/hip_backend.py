@@ -64,6 +64,10 @@ class HbmStorage:
             # concern (cpu_backend implements a true counting bloom).
             self.filter_freq = fo.filter_freq
 
+        # composite-key support (EmbeddingCollection): 0 = plain keys
+        self.key_bits = 0
+        self.dvd_per_table = self.default_value_dim
+
         cap = _pow2(ev_option.init_capacity)
         self._alloc_table(cap)
         self._alloc_slabs(max(1024, cap // 2))
@@ -172,8 +176,8 @@ class HbmStorage:
         return self.ext.ht_lookup_insert(
             keys, counts_i32, self.ht_keys, self.ht_slot, self.ht_freq,
             self.ht_version, self.slot_counter, self.entry_counter,
-            self.values, self.default_values, self.filter_freq, step, train,
-            self.error_flag)
+            self.values, self.default_values, self.dvd_per_table,
+            self.key_bits, self.filter_freq, step, train, self.error_flag)
 
     def lookup(self, keys):
         slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)
