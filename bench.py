@@ -58,7 +58,8 @@ def main():
     bf16 = (not args.no_bf16) and device.type == "cuda"
     model = DLRM(device=device, bf16=bf16, sharded=distributed)
     ds = CriteoSyntheticDataset(batch_size=args.batch, device=device,
-                                seed=1234, rank=rank)
+                                seed=1234, rank=rank,
+                                matrix_format=not distributed)
     opt = make_optimizer(args.optimizer, params=model.parameters(),
                          embedding_variables=model.embedding_variables(),
                          learning_rate=0.001)

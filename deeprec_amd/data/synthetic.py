@@ -32,7 +32,8 @@ class CriteoSyntheticDataset:
     def __init__(self, batch_size: int, device="cpu", seed: int = 1234,
                  rank: int = 0, zipf_alpha: float = 1.05,
                  multi_hot: Optional[int] = None,
-                 cardinalities=None, num_batches: Optional[int] = None):
+                 cardinalities=None, num_batches: Optional[int] = None,
+                 matrix_format: bool = False):
         self.batch_size = batch_size
         self.device = torch.device(device)
         self.gen = torch.Generator(device="cpu")
@@ -41,6 +42,9 @@ class CriteoSyntheticDataset:
         self.zipf_alpha = zipf_alpha
         self.multi_hot = multi_hot  # ids per sample per feature (None=1)
         self.num_batches = num_batches
+        # matrix_format: sparse ids as one [batch, 26] int64 tensor (the
+        # 1-hot Criteo layout) feeding EmbeddingCollection.lookup_matrix
+        self.matrix_format = matrix_format
         # precompute per-feature Zipf CDF over a capped support for sampling
         self._cdfs = []
         for card in self.cardinalities:
@@ -63,6 +67,15 @@ class CriteoSyntheticDataset:
     def next_batch(self):
         b = self.batch_size
         dense = torch.randn(b, NUM_DENSE, generator=self.gen)
+        if self.matrix_format:
+            ids = torch.stack([self._sample_ids(f, b)
+                               for f in range(NUM_SPARSE)], dim=1)
+            labels = (torch.rand(b, generator=self.gen) < 0.3).float()
+            if self.device.type != "cpu":
+                dense = dense.to(self.device, non_blocking=True)
+                ids = ids.to(self.device, non_blocking=True)
+                labels = labels.to(self.device, non_blocking=True)
+            return dense, ids, labels
         sparse = []
         k = self.multi_hot or 1
         for f in range(NUM_SPARSE):

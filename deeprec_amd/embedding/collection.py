@@ -64,6 +64,7 @@ class EmbeddingCollection:
             device=self.device)
         self._anchor = torch.zeros((), device=self.device,
                                    requires_grad=trainable)
+        self._matrix_cache = {}
         self._pending_grads: List = []
         self._recorded_ids: List[torch.Tensor] = []
         self._record_sparse_ids = False
@@ -152,17 +153,80 @@ class EmbeddingCollection:
         if not train:
             return self._forward(uniq, slots, inverse, offsets_cat,
                                  weights_cat, batch, out_dtype)
-        # CSR over unique keys for the atomic-free backward
-        order = torch.argsort(inverse).to(torch.int32)
-        bounds = torch.zeros(uniq.numel() + 1, dtype=torch.int32,
-                             device=self.device)
-        bounds[1:] = counts.to(torch.int32).cumsum(0)
+        order, bounds, chunk_u, chunk_k0 = self._prep_backward(inverse,
+                                                               counts)
         row_coeff = self._row_coeffs(offsets_cat, row_ids_cat, weights_cat,
                                      batch)
         return _CollectionLookup.apply(
             self._anchor, self, uniq, slots, inverse, offsets_cat,
-            row_ids_cat, order, bounds, row_coeff, weights_cat, batch,
-            out_dtype)
+            row_ids_cat, order, bounds, chunk_u, chunk_k0, row_coeff,
+            weights_cat, batch, out_dtype)
+
+    _CHUNK = 128
+
+    def _prep_backward(self, inverse, counts):
+        """CSR over unique keys + fixed-size occurrence chunks so the
+        backward is balanced under zipf-hot keys (no 10k-iteration
+        threads)."""
+        m = counts.numel()
+        dev = self.device
+        order = torch.argsort(inverse).to(torch.int32)
+        c32 = counts.to(torch.int32)
+        bounds = torch.zeros(m + 1, dtype=torch.int32, device=dev)
+        bounds[1:] = c32.cumsum(0)
+        if self.device.type != "cuda":
+            return order, bounds, None, None
+        nch = (counts + (self._CHUNK - 1)) // self._CHUNK
+        chunk_u = torch.repeat_interleave(
+            torch.arange(m, device=dev, dtype=torch.int64), nch)
+        chunk_base = nch.cumsum(0) - nch
+        pos_in_u = (torch.arange(chunk_u.numel(), device=dev,
+                                 dtype=torch.int64)
+                    - chunk_base[chunk_u])
+        chunk_k0 = (bounds[:-1].to(torch.int64)[chunk_u]
+                    + pos_in_u * self._CHUNK).to(torch.int32)
+        return order, bounds, chunk_u.to(torch.int32), chunk_k0
+
+    def lookup_matrix(self, ids: torch.Tensor, out_dtype=None,
+                      train: bool = True) -> torch.Tensor:
+        """Fixed-shape fast path: ids [batch, n_tables] (one id per table
+        per sample, the Criteo layout). All ragged glue (offsets, row ids,
+        combiner coeffs, table tags) is cached per batch size, so a step
+        costs: 1 transpose+add, 1 unique, the probe, the fused fwd/bwd and
+        the apply."""
+        batch, n = ids.shape
+        assert n == self.n_tables
+        cache = self._matrix_cache.get(batch)
+        if cache is None:
+            dev = self.device
+            nb = n * batch
+            cache = {
+                "offsets": torch.arange(nb + 1, dtype=torch.int32,
+                                        device=dev),
+                "row_ids": torch.arange(nb, dtype=torch.int32, device=dev),
+                "row_coeff": torch.ones(nb, device=dev),
+                "tags": (torch.arange(n, dtype=torch.int64, device=dev)
+                         << KEY_BITS).repeat_interleave(batch),
+            }
+            self._matrix_cache[batch] = cache
+        values_cat = ids.t().reshape(-1) + cache["tags"]
+        uniq, inverse, counts = torch.unique(
+            values_cat, return_inverse=True, return_counts=True)
+        train = train and self.trainable
+        slots = self.storage.lookup_or_create(
+            uniq, counts, get_global_step(), train=train)
+        if train and self._record_sparse_ids:
+            self._recorded_ids.append(uniq.detach())
+        inverse = inverse.to(torch.int32)
+        if not train:
+            return self._forward(uniq, slots, inverse, cache["offsets"],
+                                 None, batch, out_dtype)
+        order, bounds, chunk_u, chunk_k0 = self._prep_backward(inverse,
+                                                               counts)
+        return _CollectionLookup.apply(
+            self._anchor, self, uniq, slots, inverse, cache["offsets"],
+            cache["row_ids"], order, bounds, chunk_u, chunk_k0,
+            cache["row_coeff"], None, batch, out_dtype)
 
     def _forward(self, uniq, slots, inverse, offsets_cat, weights_cat, batch,
                  out_dtype):
@@ -197,13 +261,14 @@ class EmbeddingCollection:
             outs.append(out_t)
         return torch.cat(outs, dim=1)  # [B, N*D], table-major inner
 
-    def _backward(self, grad_out, order, bounds, row_ids_cat, weights_cat,
-                  row_coeff, m, batch):
+    def _backward(self, grad_out, order, bounds, chunk_u, chunk_k0,
+                  row_ids_cat, weights_cat, row_coeff, m, batch):
         if self.device.type == "cuda":
-            return self.storage.ext.group_pooled_bwd(
+            return self.storage.ext.group_pooled_bwd_chunked(
                 grad_out.contiguous(), order, bounds, row_ids_cat,
                 weights_cat if weights_cat is not None else torch.Tensor(),
-                row_coeff, m, batch, self.n_tables, self.dim)
+                row_coeff, chunk_u, chunk_k0, self._CHUNK, m, batch,
+                self.n_tables, self.dim)
         # CPU reference path
         g = grad_out.float().reshape(batch, self.n_tables, self.dim)
         grad_unique = torch.zeros(m, self.dim)
@@ -270,14 +335,15 @@ class EmbeddingCollection:
 class _CollectionLookup(torch.autograd.Function):
     @staticmethod
     def forward(ctx, anchor, coll, uniq, slots, inverse, offsets_cat,
-                row_ids_cat, order, bounds, row_coeff, weights_cat, batch,
-                out_dtype):
+                row_ids_cat, order, bounds, chunk_u, chunk_k0, row_coeff,
+                weights_cat, batch, out_dtype):
         out = coll._forward(uniq, slots, inverse, offsets_cat, weights_cat,
                             batch, out_dtype)
         ctx.coll = coll
         ctx.batch = batch
         ctx.save_for_backward(uniq, slots, order, bounds, row_ids_cat,
                               row_coeff)
+        ctx.chunks = (chunk_u, chunk_k0)
         ctx.weights_cat = weights_cat
         return out
 
@@ -285,8 +351,10 @@ class _CollectionLookup(torch.autograd.Function):
     def backward(ctx, grad_out):
         uniq, slots, order, bounds, row_ids_cat, row_coeff = ctx.saved_tensors
         coll = ctx.coll
-        grad_unique = coll._backward(grad_out, order, bounds, row_ids_cat,
+        chunk_u, chunk_k0 = ctx.chunks
+        grad_unique = coll._backward(grad_out, order, bounds, chunk_u,
+                                     chunk_k0, row_ids_cat,
                                      ctx.weights_cat, row_coeff,
                                      uniq.numel(), ctx.batch)
         coll.accumulate_grad(slots, uniq, grad_unique)
-        return (torch.zeros_like(coll._anchor),) + (None,) * 12
+        return (torch.zeros_like(coll._anchor),) + (None,) * 14

@@ -135,3 +135,27 @@ def test_dlrm_with_collection_cpu():
         opt.step()
         assert torch.isfinite(loss)
     assert m.collection.size() > 0
+
+
+def test_lookup_matrix_matches_ragged():
+    torch.manual_seed(5)
+    coll = EmbeddingCollection("coll_m", ["a", "b", "c"], 8,
+                               ev_option=_const_opt(dvd=8))
+    ids = torch.randint(0, 50, (16, 3))
+    out_m = coll.lookup_matrix(ids)
+    sp_list = [RaggedIds.from_dense(ids[:, t:t + 1]) for t in range(3)]
+    out_r = coll.lookup(sp_list, train=False)
+    torch.testing.assert_close(out_m, out_r)
+
+
+def test_lookup_matrix_training():
+    from deeprec_amd.optimizers import AdagradOptimizer
+    coll = EmbeddingCollection("coll_mt", ["a", "b"], 4,
+                               ev_option=_const_opt())
+    opt = AdagradOptimizer(embedding_variables=[coll], learning_rate=0.1)
+    ids = torch.tensor([[1, 2], [1, 3]])
+    out = coll.lookup_matrix(ids)
+    out.sum().backward()
+    opt.step()
+    out2 = coll.lookup_matrix(ids, train=False)
+    assert not torch.allclose(out, out2)
