@@ -1,0 +1,75 @@
+"""FusedMLP — bf16 MFMA linear layers with fp32 master weights.
+
+GPU path uses the hand-written gfx950 kernels (ops/hip/dense_kernels.hip);
+CPU path falls back to torch (the numerics reference). This mirrors the
+reference's keep_weights bf16 scope (fp32 master weights, bf16 compute —
+reference: python/ops/variable_scope.py:3008) with the cast fused into the
+layer instead of graph-level cast nodes.
+"""
+from __future__ import annotations
+
+import math
+from typing import List
+
+import torch
+import torch.nn as nn
+
+
+class _FusedLinear(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, relu):
+        from deeprec_amd.ops.build_ext import require_extension
+        ext = require_extension()
+        w16 = weight.detach().to(torch.bfloat16)
+        x16 = x.to(torch.bfloat16).contiguous()
+        out = ext.linear_fwd(x16, w16, bias.detach().float(), relu)
+        ctx.ext = ext
+        ctx.relu = relu
+        ctx.save_for_backward(x16, w16, out)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x16, w16, out = ctx.saved_tensors
+        ext = ctx.ext
+        g = grad_out.to(torch.bfloat16).contiguous()
+        if ctx.relu:
+            g = ext.relu_bwd(g, out)
+        dx = ext.linear_dx(g, w16)
+        dw, db = ext.linear_dw(g, x16, True)
+        return dx, dw, db, None
+
+
+class FusedLinear(nn.Module):
+    """Linear(+ReLU) with fp32 master weight, bf16 MFMA compute on GPU."""
+
+    def __init__(self, in_features: int, out_features: int,
+                 relu: bool = True):
+        super().__init__()
+        self.in_features = in_features
+        self.out_features = out_features
+        self.relu = relu
+        self.weight = nn.Parameter(torch.empty(out_features, in_features))
+        self.bias = nn.Parameter(torch.zeros(out_features))
+        # torch Linear default init (kaiming uniform)
+        nn.init.kaiming_uniform_(self.weight, a=math.sqrt(5))
+        bound = 1.0 / math.sqrt(in_features)
+        nn.init.uniform_(self.bias, -bound, bound)
+
+    def forward(self, x):
+        if x.device.type == "cuda" and x.shape[0] % 16 == 0:
+            return _FusedLinear.apply(x, self.weight, self.bias, self.relu)
+        out = nn.functional.linear(x, self.weight.to(x.dtype),
+                                   self.bias.to(x.dtype))
+        return nn.functional.relu(out) if self.relu else out
+
+
+def fused_mlp(sizes: List[int], in_dim: int,
+              final_activation: bool = True) -> nn.Sequential:
+    layers = []
+    d = in_dim
+    for i, h in enumerate(sizes):
+        layers.append(FusedLinear(d, h,
+                                  relu=final_activation or i + 1 < len(sizes)))
+        d = h
+    return nn.Sequential(*layers)

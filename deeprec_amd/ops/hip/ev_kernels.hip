@@ -1016,7 +1016,10 @@ void apply_ftrl(torch::Tensor w, torch::Tensor n, torch::Tensor z,
       (float)l1, (float)l2, (float)lr_power);
 }
 
+void register_dense(pybind11::module_& mod);  // dense_kernels.hip
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
+  register_dense(mod);
   mod.def("ht_lookup_insert", &ht_lookup_insert);
   mod.def("ht_insert_bulk", &ht_insert_bulk);
   mod.def("ht_lookup", &ht_lookup);
