@@ -50,13 +50,19 @@ class DLRM(nn.Module):
         # dense input padded 13 -> 16: K=13 GEMMs pick poor hipBLASLt tiles
         # (measured 148us for 8192x512x13); zero-padding is math-identical
         self.dense_in = 16
-        self.mlp_bot = _mlp(mlp_bot, self.dense_in)
-        n_f = num_sparse + 1
-        inter_dim = (n_f * (n_f - 1)) // 2 if interaction_op == "dot" \
-            else n_f * embedding_dim
-        self.mlp_top = _mlp(list(mlp_top) + [1], embedding_dim + inter_dim
-                            if interaction_op == "dot" else inter_dim,
-                            final_activation=False)
+        use_fused = bf16 and torch.device(device).type == "cuda"
+        top_in = (embedding_dim + (n_f := num_sparse + 1) * (n_f - 1) // 2
+                  if interaction_op == "dot"
+                  else (num_sparse + 1) * embedding_dim)
+        if use_fused:
+            from deeprec_amd.ops.fused_mlp import fused_mlp
+            self.mlp_bot = fused_mlp(mlp_bot, self.dense_in)
+            self.mlp_top = fused_mlp(list(mlp_top) + [1], top_in,
+                                     final_activation=False)
+        else:
+            self.mlp_bot = _mlp(mlp_bot, self.dense_in)
+            self.mlp_top = _mlp(list(mlp_top) + [1], top_in,
+                                final_activation=False)
         self.to(self.device_)
 
         self.collection = None

@@ -25,6 +25,7 @@ class _FusedLinear(torch.autograd.Function):
         out = ext.linear_fwd(x16, w16, bias.detach().float(), relu)
         ctx.ext = ext
         ctx.relu = relu
+        ctx.x_dtype = x.dtype
         ctx.save_for_backward(x16, w16, out)
         return out
 
@@ -37,7 +38,7 @@ class _FusedLinear(torch.autograd.Function):
             g = ext.relu_bwd(g, out)
         dx = ext.linear_dx(g, w16)
         dw, db = ext.linear_dw(g, x16, True)
-        return dx, dw, db, None
+        return dx.to(ctx.x_dtype), dw, db, None
 
 
 class FusedLinear(nn.Module):
