@@ -51,9 +51,13 @@ class DLRM(nn.Module):
         # (measured 148us for 8192x512x13); zero-padding is math-identical
         self.dense_in = 16
         use_fused = bf16 and torch.device(device).type == "cuda"
-        top_in = (embedding_dim + (n_f := num_sparse + 1) * (n_f - 1) // 2
-                  if interaction_op == "dot"
-                  else (num_sparse + 1) * embedding_dim)
+        n_f = num_sparse + 1
+        n_pairs = n_f * (n_f - 1) // 2
+        # pad pair count to a multiple of 16 so the fused top-MLP GEMM gets
+        # aligned K fragments (351 -> 352 for the 26-feature config)
+        self.n_pairs_pad = (n_pairs + 15) & ~15
+        top_in = (embedding_dim + self.n_pairs_pad
+                  if interaction_op == "dot" else n_f * embedding_dim)
         if use_fused:
             from deeprec_amd.ops.fused_mlp import fused_mlp
             self.mlp_bot = fused_mlp(mlp_bot, self.dense_in)
@@ -93,12 +97,10 @@ class DLRM(nn.Module):
         return [self.collection] if self.collection is not None else self.evs
 
     def _interact(self, feats: torch.Tensor) -> torch.Tensor:
-        """feats: [B, F, D] -> pairwise dots, upper triangle (i<j)
+        """feats: [B, F, D] -> pairwise dots, upper triangle (i<j), padded
         (reference: _dot_op, modelzoo/dlrm/train.py:121-132)."""
-        z = torch.bmm(feats, feats.transpose(1, 2))  # [B, F, F]
-        f = feats.shape[1]
-        iu = torch.triu_indices(f, f, offset=1, device=feats.device)
-        return z[:, iu[0], iu[1]]
+        from deeprec_amd.ops.fused_mlp import dot_interaction
+        return dot_interaction(feats, self.n_pairs_pad)
 
     def forward(self, dense: torch.Tensor, sparse_ids,
                 train: bool = True) -> torch.Tensor:

@@ -65,6 +65,41 @@ class FusedLinear(nn.Module):
         return nn.functional.relu(out) if self.relu else out
 
 
+class _DotInteraction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, feats, p_pad):
+        from deeprec_amd.ops.build_ext import require_extension
+        ext = require_extension()
+        f16 = feats.to(torch.bfloat16).contiguous()
+        out = ext.interact_fwd(f16, p_pad)
+        ctx.ext = ext
+        ctx.f_dtype = feats.dtype
+        ctx.save_for_backward(f16)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        (f16,) = ctx.saved_tensors
+        df = ctx.ext.interact_bwd(grad_out.to(torch.bfloat16), f16)
+        return df.to(ctx.f_dtype), None
+
+
+def dot_interaction(feats: torch.Tensor, p_pad: int = None) -> torch.Tensor:
+    """feats [B, F, D] -> pairwise dots [B, p_pad] (i<j upper triangle,
+    zero-padded). Fused gfx950 kernel on GPU; torch fallback elsewhere."""
+    b, f, d = feats.shape
+    p = f * (f - 1) // 2
+    p_pad = p_pad or p
+    if feats.device.type == "cuda" and (f * d) % 8 == 0:
+        return _DotInteraction.apply(feats, p_pad)
+    z = torch.bmm(feats.float(), feats.float().transpose(1, 2))
+    iu = torch.triu_indices(f, f, offset=1, device=feats.device)
+    out = z[:, iu[0], iu[1]]
+    if p_pad > p:
+        out = torch.nn.functional.pad(out, (0, p_pad - p))
+    return out.to(feats.dtype)
+
+
 def fused_mlp(sizes: List[int], in_dim: int,
               final_activation: bool = True) -> nn.Sequential:
     layers = []

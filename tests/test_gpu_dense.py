@@ -89,3 +89,26 @@ def test_fused_mlp_training_step():
                                    rtol=5e-2, atol=2e-1)
         torch.testing.assert_close(fl.bias.grad, rl.bias.grad,
                                    rtol=5e-2, atol=2e-1)
+
+
+@pytest.mark.parametrize("b,f,d", [(64, 27, 16), (100, 8, 8), (256, 5, 16)])
+def test_dot_interaction(b, f, d):
+    from deeprec_amd.ops.fused_mlp import dot_interaction
+    torch.manual_seed(4)
+    p = f * (f - 1) // 2
+    p_pad = (p + 15) & ~15
+    feats = torch.randn(b, f, d, device=DEV, requires_grad=True)
+    out = dot_interaction(feats, p_pad)
+    assert out.shape == (b, p_pad)
+    # reference
+    fr = feats.detach().clone().requires_grad_(True)
+    z = torch.bmm(fr, fr.transpose(1, 2))
+    iu = torch.triu_indices(f, f, offset=1, device=DEV)
+    ref = z[:, iu[0], iu[1]]
+    torch.testing.assert_close(out[:, :p].float(), ref, rtol=2e-2, atol=2e-2)
+    assert (out[:, p:] == 0).all()
+    g = torch.randn(b, p, device=DEV)
+    gp = torch.nn.functional.pad(g, (0, p_pad - p))
+    out.backward(gp)
+    ref.backward(g)
+    torch.testing.assert_close(feats.grad, fr.grad, rtol=2e-2, atol=1e-1)
