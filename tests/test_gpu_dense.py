@@ -97,7 +97,9 @@ def test_dot_interaction(b, f, d):
     torch.manual_seed(4)
     p = f * (f - 1) // 2
     p_pad = (p + 15) & ~15
-    feats = torch.randn(b, f, d, device=DEV, requires_grad=True)
+    feats = torch.randn(b, f, d, device=DEV)
+    # quantize to bf16 grid so the fp32 torch reference sees the same inputs
+    feats = feats.to(torch.bfloat16).float().requires_grad_(True)
     out = dot_interaction(feats, p_pad)
     assert out.shape == (b, p_pad)
     # reference
