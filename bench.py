@@ -59,7 +59,7 @@ def main():
     model = DLRM(device=device, bf16=bf16, sharded=distributed)
     ds = CriteoSyntheticDataset(batch_size=args.batch, device=device,
                                 seed=1234, rank=rank,
-                                matrix_format=not distributed)
+                                matrix_format=True)
     opt = make_optimizer(args.optimizer, params=model.parameters(),
                          embedding_variables=model.embedding_variables(),
                          learning_rate=0.001)

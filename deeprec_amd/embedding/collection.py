@@ -229,8 +229,15 @@ class EmbeddingCollection:
             cache["row_coeff"], None, batch, out_dtype)
 
     def _forward(self, uniq, slots, inverse, offsets_cat, weights_cat, batch,
-                 out_dtype):
+                 out_dtype, emb_override=None):
         if self.device.type == "cuda":
+            if emb_override is not None:
+                return self.storage.ext.group_pooled_fwd_direct(
+                    emb_override.contiguous(), uniq, inverse, offsets_cat,
+                    weights_cat if weights_cat is not None
+                    else torch.Tensor(),
+                    self._combiner_ids, batch, self.n_tables,
+                    out_dtype or torch.float32)
             return self.storage.ext.group_pooled_fwd(
                 self.storage.values, self.storage.default_values, uniq, slots,
                 inverse, offsets_cat,
@@ -240,7 +247,8 @@ class EmbeddingCollection:
                 self.storage._use_no_permission(),
                 out_dtype or torch.float32)
         # CPU reference: gather then pool per-table slices
-        emb = self.storage.gather(uniq, slots)  # [m, D]
+        emb = (emb_override if emb_override is not None
+               else self.storage.gather(uniq, slots))  # [m, D]
         outs = []
         row_ids_all = None
         nb = offsets_cat.numel() - 1  # == n_tables * batch

@@ -71,14 +71,14 @@ class DLRM(nn.Module):
 
         self.collection = None
         if sharded:
-            from deeprec_amd.parallel.sharded_embedding import (
-                ShardedEmbeddingVariable)
-            self.evs = [
-                ShardedEmbeddingVariable(f"{name_prefix}/C{i+1}",
-                                         embedding_dim,
-                                         ev_option=ev_option,
-                                         device=self.device_)
-                for i in range(num_sparse)]
+            from deeprec_amd.parallel.sharded_collection import (
+                ShardedEmbeddingCollection)
+            self.collection = ShardedEmbeddingCollection(
+                f"{name_prefix}/sparse",
+                [f"C{i+1}" for i in range(num_sparse)], embedding_dim,
+                ev_option=ev_option, combiners=["mean"] * num_sparse,
+                device=self.device_)
+            self.evs = []
         elif use_collection:
             from deeprec_amd.embedding.collection import EmbeddingCollection
             self.collection = EmbeddingCollection(

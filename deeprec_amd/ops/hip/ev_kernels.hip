@@ -352,6 +352,9 @@ __global__ void k_pooled_bwd(
 // b = r % B. Output is written interleaved as out[b, t*D + d] giving the
 // [B, N*D] concat layout models consume with no extra copy.
 
+// direct != nullptr (sharded path): embedding rows come pre-gathered per
+// unique key (from peer shards over RCCL all-to-all) instead of the local
+// slot slab.
 template <typename OutT>
 __global__ void k_group_pooled_fwd(
     const float* __restrict__ values, const float* __restrict__ default_values,
@@ -360,7 +363,8 @@ __global__ void k_group_pooled_fwd(
     const float* __restrict__ weights,
     const int32_t* __restrict__ combiner_ids, int batch, int n_tables,
     int dim, int default_value_dim, int key_bits, float no_permission_value,
-    int use_no_permission, OutT* __restrict__ out) {
+    int use_no_permission, const float* __restrict__ direct,
+    OutT* __restrict__ out) {
   int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t total = (int64_t)batch * n_tables * dim;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
@@ -375,18 +379,22 @@ __global__ void k_group_pooled_fwd(
     float acc = 0.0f, wacc = 0.0f;
     for (int j = beg; j < end; ++j) {
       int u = inverse[j];
-      int32_t s = slots[u];
       float v;
-      if (s >= 0) {
-        v = values[(int64_t)s * dim + d];
-      } else if (use_no_permission) {
-        v = no_permission_value;
+      if (direct) {
+        v = direct[(int64_t)u * dim + d];
       } else {
-        int64_t k = keys[u];
-        int64_t row = (int64_t)(table)*default_value_dim +
-                      (int64_t)((uint64_t)(k & key_mask) %
-                                (uint64_t)default_value_dim);
-        v = default_values[row * dim + d];
+        int32_t s = slots[u];
+        if (s >= 0) {
+          v = values[(int64_t)s * dim + d];
+        } else if (use_no_permission) {
+          v = no_permission_value;
+        } else {
+          int64_t k = keys[u];
+          int64_t row = (int64_t)(table)*default_value_dim +
+                        (int64_t)((uint64_t)(k & key_mask) %
+                                  (uint64_t)default_value_dim);
+          v = default_values[row * dim + d];
+        }
       }
       float w = weights ? weights[j] : 1.0f;
       acc += w * v;
@@ -835,7 +843,7 @@ torch::Tensor group_pooled_fwd(
             inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr,
             combiner_ids.data_ptr<int32_t>(), (int)batch, (int)n_tables, dim,
             default_values.size(0) / (int)n_tables, (int)key_bits,
-            (float)no_permission_value, use_no_permission ? 1 : 0,
+            (float)no_permission_value, use_no_permission ? 1 : 0, nullptr,
             reinterpret_cast<__hip_bfloat16*>(out.data_ptr<at::BFloat16>()));
   } else {
     k_group_pooled_fwd<float><<<n_blocks(total), kBlock, 0, stream>>>(
@@ -844,8 +852,39 @@ torch::Tensor group_pooled_fwd(
         inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr,
         combiner_ids.data_ptr<int32_t>(), (int)batch, (int)n_tables, dim,
         default_values.size(0) / (int)n_tables, (int)key_bits,
-        (float)no_permission_value, use_no_permission ? 1 : 0,
+        (float)no_permission_value, use_no_permission ? 1 : 0, nullptr,
         out.data_ptr<float>());
+  }
+  return out;
+}
+
+torch::Tensor group_pooled_fwd_direct(
+    torch::Tensor emb_rows, torch::Tensor keys, torch::Tensor inverse,
+    torch::Tensor offsets, torch::Tensor weights, torch::Tensor combiner_ids,
+    int64_t batch, int64_t n_tables, torch::ScalarType out_dtype) {
+  int dim = emb_rows.size(1);
+  auto out = torch::empty({batch, n_tables * dim},
+                          emb_rows.options().dtype(out_dtype));
+  int64_t total = batch * n_tables * dim;
+  if (total == 0) return out;
+  auto stream = current_stream();
+  const float* wptr =
+      weights.defined() && weights.numel() ? weights.data_ptr<float>()
+                                           : nullptr;
+  if (out_dtype == torch::kBFloat16) {
+    k_group_pooled_fwd<__hip_bfloat16>
+        <<<n_blocks(total), kBlock, 0, stream>>>(
+            nullptr, nullptr, keys.data_ptr<int64_t>(), nullptr,
+            inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr,
+            combiner_ids.data_ptr<int32_t>(), (int)batch, (int)n_tables, dim,
+            1, 0, 0.0f, 0, emb_rows.data_ptr<float>(),
+            reinterpret_cast<__hip_bfloat16*>(out.data_ptr<at::BFloat16>()));
+  } else {
+    k_group_pooled_fwd<float><<<n_blocks(total), kBlock, 0, stream>>>(
+        nullptr, nullptr, keys.data_ptr<int64_t>(), nullptr,
+        inverse.data_ptr<int32_t>(), offsets.data_ptr<int32_t>(), wptr,
+        combiner_ids.data_ptr<int32_t>(), (int)batch, (int)n_tables, dim, 1,
+        0, 0.0f, 0, emb_rows.data_ptr<float>(), out.data_ptr<float>());
   }
   return out;
 }
@@ -1028,6 +1067,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("pooled_fwd", &pooled_fwd);
   mod.def("pooled_bwd", &pooled_bwd);
   mod.def("group_pooled_fwd", &group_pooled_fwd);
+  mod.def("group_pooled_fwd_direct", &group_pooled_fwd_direct);
   mod.def("group_pooled_bwd", &group_pooled_bwd);
   mod.def("group_pooled_bwd_chunked", &group_pooled_bwd_chunked);
   mod.def("apply_sgd", &apply_sgd);

@@ -1,0 +1,208 @@
+"""ShardedEmbeddingCollection — model-parallel multi-table embeddings.
+
+The multi-GPU fast path: all N tables share one composite-key space, so a
+training step does ONE two-phase all-to-all for keys (counts then payload,
+reference SOK protocol SURVEY.md §3.3), one local probe+gather on the
+owner, ONE all-to-all back for embedding rows, then the same fused group
+pooling as the single-GPU collection. Backward reverses the exchange; the
+owner applies the fused sparse update.
+
+xGMI note: one big all-to-all per step (vs 26 per-table exchanges) is the
+shape that drives all 7 point-to-point links concurrently.
+"""
+from __future__ import annotations
+
+from typing import Optional, Sequence
+
+import torch
+
+from deeprec_amd.embedding.collection import KEY_BITS, EmbeddingCollection
+from deeprec_amd.embedding.options import EmbeddingVariableOption
+from deeprec_amd.embedding.ragged import RaggedIds
+from deeprec_amd.embedding.variable import get_global_step
+from deeprec_amd.parallel import comm
+
+
+class ShardedEmbeddingCollection:
+    """Per-rank shard of an EmbeddingCollection; keys routed by
+    raw_id % world_size (composite tags are multiples of 2^48, so routing
+    is table-independent for power-of-two worlds and deterministic for
+    all)."""
+
+    def __init__(self, name: str, table_names: Sequence[str],
+                 embedding_dim: int,
+                 ev_option: Optional[EmbeddingVariableOption] = None,
+                 combiners=None, device=None, value_dtype=torch.float32,
+                 generator=None, trainable: bool = True):
+        self.world = comm.world_size()
+        self.rank = comm.rank()
+        self.local = EmbeddingCollection(
+            f"{name}/part_{self.rank}", table_names, embedding_dim,
+            ev_option, combiners, device, value_dtype, generator, trainable)
+        self.name = name
+        self.dim = embedding_dim
+        self.n_tables = self.local.n_tables
+
+    # ---- optimizer-facing delegation ----
+    @property
+    def device(self):
+        return self.local.device
+
+    @property
+    def storage(self):
+        return self.local.storage
+
+    @property
+    def trainable(self):
+        return self.local.trainable
+
+    @property
+    def _anchor(self):
+        return self.local._anchor
+
+    @property
+    def _pending_grads(self):
+        return self.local._pending_grads
+
+    def consume_grads(self):
+        return self.local.consume_grads()
+
+    def get_slab(self, *a, **kw):
+        return self.local.get_slab(*a, **kw)
+
+    def size(self):
+        return self.local.size()
+
+    def shrink(self, step=None):
+        return self.local.shrink(step)
+
+    def export_tables(self, include_filtered=False):
+        return self.local.export_tables(include_filtered)
+
+    def restore_table(self, table, keys, values, freqs=None, versions=None):
+        mask = (keys % self.world) == self.rank
+        self.local.restore_table(
+            table, keys[mask], values[mask],
+            None if freqs is None else freqs[mask],
+            None if versions is None else versions[mask])
+
+    # ---- lookups ----
+    def lookup_matrix(self, ids: torch.Tensor, out_dtype=None,
+                      train: bool = True) -> torch.Tensor:
+        batch, n = ids.shape
+        assert n == self.n_tables
+        coll = self.local
+        cache = coll._matrix_cache.get(batch)
+        if cache is None:
+            dev = coll.device
+            nb = n * batch
+            cache = {
+                "offsets": torch.arange(nb + 1, dtype=torch.int32,
+                                        device=dev),
+                "row_ids": torch.arange(nb, dtype=torch.int32, device=dev),
+                "row_coeff": torch.ones(nb, device=dev),
+                "tags": (torch.arange(n, dtype=torch.int64, device=dev)
+                         << KEY_BITS).repeat_interleave(batch),
+            }
+            coll._matrix_cache[batch] = cache
+        values_cat = ids.t().reshape(-1) + cache["tags"]
+        return self._lookup_cat(values_cat, cache["offsets"],
+                                cache["row_ids"], cache["row_coeff"], None,
+                                batch, out_dtype, train)
+
+    def lookup(self, sp_list: Sequence[RaggedIds], out_dtype=None,
+               train: bool = True) -> torch.Tensor:
+        coll = self.local
+        batch, values_cat, offsets_cat, row_ids_cat, weights_cat = \
+            coll._concat_inputs(sp_list)
+        row_coeff = coll._row_coeffs(offsets_cat, row_ids_cat, weights_cat,
+                                     batch)
+        return self._lookup_cat(values_cat, offsets_cat, row_ids_cat,
+                                row_coeff, weights_cat, batch, out_dtype,
+                                train)
+
+    def _lookup_cat(self, values_cat, offsets_cat, row_ids_cat, row_coeff,
+                    weights_cat, batch, out_dtype, train):
+        coll = self.local
+        uniq, inverse, counts = torch.unique(
+            values_cat, return_inverse=True, return_counts=True)
+        inverse = inverse.to(torch.int32)
+        train = train and coll.trainable
+        if not train:
+            emb = _exchange_lookup(self, uniq, counts, train=False)[0]
+            return coll._forward(uniq, None, inverse, offsets_cat,
+                                 weights_cat, batch, out_dtype,
+                                 emb_override=emb)
+        order, bounds, chunk_u, chunk_k0 = coll._prep_backward(inverse,
+                                                               counts)
+        return _ShardedCollectionLookup.apply(
+            coll._anchor, self, uniq, counts, inverse, offsets_cat,
+            row_ids_cat, order, bounds, chunk_u, chunk_k0, row_coeff,
+            weights_cat, batch, out_dtype)
+
+
+def _exchange_lookup(sev: ShardedEmbeddingCollection, uniq, counts, train):
+    """Route unique composite keys to owners, lookup there, return
+    ([m, D] fp32 embedding rows in uniq order, owner-side context)."""
+    coll = sev.local
+    w = sev.world
+    raw = uniq & ((1 << KEY_BITS) - 1)
+    owner = (raw % w).to(torch.int64)
+    order_o = torch.argsort(owner, stable=True)
+    send_keys = uniq[order_o]
+    send_counts = counts[order_o]
+    send_splits = torch.bincount(owner, minlength=w)
+    recv_splits = comm.exchange_counts(send_splits)
+    in_sp, out_sp = send_splits.tolist(), recv_splits.tolist()
+    recv_keys = comm.all_to_all_single(send_keys, in_sp, out_sp)
+    recv_counts = comm.all_to_all_single(send_counts, in_sp, out_sp)
+    uniq2, inv2 = torch.unique(recv_keys, return_inverse=True)
+    counts2 = torch.zeros(uniq2.numel(), dtype=recv_counts.dtype,
+                          device=uniq.device)
+    counts2.index_add_(0, inv2, recv_counts)
+    slots2 = coll.storage.lookup_or_create(uniq2, counts2, get_global_step(),
+                                           train=train)
+    emb2 = coll.storage.gather(uniq2, slots2)        # [m2, D] fp32
+    emb_out = emb2[inv2]                              # [n_recv, D]
+    emb_back = comm.all_to_all_single(emb_out.contiguous(), out_sp, in_sp)
+    emb = torch.empty_like(emb_back)
+    emb[order_o] = emb_back                           # uniq order
+    ctx = (order_o, inv2, slots2, uniq2, in_sp, out_sp)
+    return emb, ctx
+
+
+class _ShardedCollectionLookup(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, anchor, sev, uniq, counts, inverse, offsets_cat,
+                row_ids_cat, order, bounds, chunk_u, chunk_k0, row_coeff,
+                weights_cat, batch, out_dtype):
+        emb, ex_ctx = _exchange_lookup(sev, uniq, counts, train=True)
+        out = sev.local._forward(uniq, None, inverse, offsets_cat,
+                                 weights_cat, batch, out_dtype,
+                                 emb_override=emb)
+        ctx.sev = sev
+        ctx.batch = batch
+        ctx.weights_cat = weights_cat
+        ctx.chunks = (chunk_u, chunk_k0)
+        ctx.ex = ex_ctx
+        ctx.save_for_backward(uniq, order, bounds, row_ids_cat, row_coeff)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        sev = ctx.sev
+        coll = sev.local
+        uniq, order, bounds, row_ids_cat, row_coeff = ctx.saved_tensors
+        chunk_u, chunk_k0 = ctx.chunks
+        order_o, inv2, slots2, uniq2, in_sp, out_sp = ctx.ex
+        grad_unique = coll._backward(grad_out, order, bounds, chunk_u,
+                                     chunk_k0, row_ids_cat, ctx.weights_cat,
+                                     row_coeff, uniq.numel(), ctx.batch)
+        grad_send = grad_unique[order_o]
+        grad_recv = comm.all_to_all_single(grad_send.contiguous(),
+                                           in_sp, out_sp)
+        grad2 = torch.zeros(uniq2.numel(), grad_recv.shape[1],
+                            device=grad_recv.device, dtype=grad_recv.dtype)
+        grad2.index_add_(0, inv2.long(), grad_recv)
+        coll.accumulate_grad(slots2, uniq2, grad2)
+        return (torch.zeros_like(coll._anchor),) + (None,) * 14

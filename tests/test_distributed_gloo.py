@@ -128,3 +128,83 @@ def _body_dlrm_sharded_step(rank, world):
 ])
 def test_distributed(body, port):
     _run_dist(body, world_size=2, port=port)
+
+
+def _body_sharded_collection(rank, world):
+    from deeprec_amd import EmbeddingVariableOption
+    from deeprec_amd.embedding.collection import EmbeddingCollection
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.parallel.sharded_collection import (
+        ShardedEmbeddingCollection)
+
+    def init(t):
+        gen = torch.Generator().manual_seed(77)
+        t.normal_(0, 1, generator=gen)
+
+    opt_ev = EmbeddingVariableOption(
+        init_option=InitializerOption(initializer=init, default_value_dim=4))
+    sc = ShardedEmbeddingCollection("sc", ["a", "b"], 8, ev_option=opt_ev)
+    # single-process reference collection with identical init
+    ref = EmbeddingCollection("ref", ["a", "b"], 8, ev_option=opt_ev)
+    torch.testing.assert_close(sc.local.storage.default_values,
+                               ref.storage.default_values)
+    o_s = AdagradOptimizer(embedding_variables=[sc], learning_rate=0.1)
+    o_r = AdagradOptimizer(embedding_variables=[ref], learning_rate=0.1)
+    for step in range(3):
+        g = torch.Generator().manual_seed(500 + step)
+        ids_all = torch.randint(0, 40, (8 * world, 2), generator=g)
+        ids_mine = ids_all[rank * 8:(rank + 1) * 8]
+        out = sc.lookup_matrix(ids_mine)
+        (out ** 2).sum().backward()
+        o_s.step()
+        # reference trains on the full global batch in one process
+        out_r = ref.lookup_matrix(ids_all)
+        (out_r ** 2).sum().backward()
+        o_r.step()
+    # shards together must equal the reference table
+    tabs_s = sc.export_tables()
+    tabs_r = ref.export_tables()
+    for name in ("a", "b"):
+        ks, vs, _, _ = tabs_s[name]
+        kr, vr, _, _ = tabs_r[name]
+        gathered_k = [None] * world
+        gathered_v = [None] * world
+        dist.all_gather_object(gathered_k, ks)
+        dist.all_gather_object(gathered_v, vs)
+        ks_all = torch.cat(gathered_k)
+        vs_all = torch.cat(gathered_v)
+        oi, ri = torch.argsort(ks_all), torch.argsort(kr)
+        torch.testing.assert_close(ks_all[oi], kr[ri])
+        torch.testing.assert_close(vs_all[oi], vr[ri], rtol=1e-4, atol=1e-5)
+
+
+def test_sharded_collection():
+    _run_dist(_body_sharded_collection, world_size=2, port=29531)
+
+
+def _body_dlrm_sharded_collection_step(rank, world):
+    from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.optimizers import AdamAsyncOptimizer
+    from deeprec_amd.parallel import DenseGradAllreducer, broadcast_parameters
+
+    torch.manual_seed(100 + rank)
+    m = DLRM(device="cpu", bf16=False, sharded=True, num_sparse=4)
+    broadcast_parameters(m.parameters())
+    opt = AdamAsyncOptimizer(params=m.parameters(),
+                             embedding_variables=m.embedding_variables())
+    red = DenseGradAllreducer(m.parameters())
+    ds = CriteoSyntheticDataset(batch_size=32, seed=5, rank=rank,
+                                matrix_format=True)
+    for step in range(2):
+        dense, ids, labels = ds.next_batch()
+        loss = m.loss_fn(m(dense, ids[:, :4]), labels)
+        opt.zero_grad()
+        loss.backward()
+        red.allreduce()
+        opt.step()
+        assert torch.isfinite(loss)
+
+
+def test_dlrm_sharded_collection():
+    _run_dist(_body_dlrm_sharded_collection_step, world_size=2, port=29532)
