@@ -296,6 +296,14 @@ class EmbeddingCollection:
         return grad_unique
 
     # ---------------- checkpoint / maintenance ----------------
+    def export(self, include_filtered: bool = False):
+        """Whole-collection export with composite keys (checkpoint format);
+        per-table views come from export_tables()."""
+        return self.storage.export(include_filtered)
+
+    def restore(self, keys, values, freqs=None, versions=None):
+        self.storage.import_(keys.to(self.device), values, freqs, versions)
+
     def export_tables(self, include_filtered: bool = False):
         """-> dict table_name -> (keys, values, freqs, versions)."""
         keys, values, freqs, versions = self.storage.export()

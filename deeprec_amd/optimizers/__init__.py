@@ -76,6 +76,23 @@ class Optimizer:
     def _post_step(self):
         pass
 
+    # ---- checkpoint integration ----
+    def state_dict(self) -> dict:
+        return {
+            "dense": self._dense.state_dict() if self._dense else None,
+            "step_count": self._step_count,
+            "beta_powers": getattr(self, "_beta_powers", None),
+            "global_step": GLOBAL_STEP.value,
+        }
+
+    def load_state_dict(self, sd: dict):
+        if sd.get("dense") is not None and self._dense is not None:
+            self._dense.load_state_dict(sd["dense"])
+        self._step_count = sd.get("step_count", 0)
+        if sd.get("beta_powers") is not None:
+            self._beta_powers = sd["beta_powers"]
+        GLOBAL_STEP.value = sd.get("global_step", GLOBAL_STEP.value)
+
 
 class GradientDescentOptimizer(Optimizer):
     sparse_name = "sgd"
