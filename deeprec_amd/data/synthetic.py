@@ -89,6 +89,27 @@ class CriteoSyntheticDataset:
             labels = labels.to(self.device, non_blocking=True)
         return dense, sparse, labels
 
+    def next_seq_batch(self, seq_len: int = 50, item_cardinality: int = 1 << 20):
+        """Behavior-sequence batch for DIN/DIEN/BST: (dense, sparse ids
+        [B, 26], seq_ids [B, T] zero-padded, target_ids [B], labels)."""
+        b = self.batch_size
+        dense = torch.randn(b, NUM_DENSE, generator=self.gen)
+        ids = torch.stack([self._sample_ids(f, b)
+                           for f in range(NUM_SPARSE)], dim=1)
+        lens = torch.randint(1, seq_len + 1, (b,), generator=self.gen)
+        seq = torch.randint(1, item_cardinality, (b, seq_len),
+                            generator=self.gen)
+        pos = torch.arange(seq_len).unsqueeze(0)
+        seq = seq * (pos < lens.unsqueeze(1))
+        target = torch.randint(1, item_cardinality, (b,),
+                               generator=self.gen)
+        labels = (torch.rand(b, generator=self.gen) < 0.3).float()
+        if self.device.type != "cpu":
+            dense, ids, seq, target, labels = (
+                t.to(self.device, non_blocking=True)
+                for t in (dense, ids, seq, target, labels))
+        return dense, ids, seq, target, labels
+
     def __iter__(self):
         n = 0
         while self.num_batches is None or n < self.num_batches:

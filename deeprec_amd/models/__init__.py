@@ -1,1 +1,14 @@
 from deeprec_amd.models.dlrm import DLRM  # noqa: F401
+from deeprec_amd.models.zoo import (  # noqa: F401
+    DBMTL, DCN, DSSM, ESMM, WDL, DeepFM, MMoE, MaskNet, PLE,
+    SimpleMultiTask)
+from deeprec_amd.models.sequence import BST, DIEN, DIN  # noqa: F401
+
+MODEL_REGISTRY = {
+    "dlrm": DLRM, "wdl": WDL, "wide_and_deep": WDL, "deepfm": DeepFM,
+    "dcn": DCN, "dcnv2": lambda **kw: DCN(v2=True, **kw), "dssm": DSSM,
+    "mmoe": MMoE, "esmm": ESMM, "dbmtl": DBMTL, "ple": PLE,
+    "masknet": MaskNet, "simple_multitask": SimpleMultiTask,
+    "din": DIN, "dien": DIEN, "bst": BST,
+}
+SEQUENCE_MODELS = {"din", "dien", "bst"}

@@ -1,0 +1,167 @@
+"""Model-zoo training runner — the reference's per-model train.py analog
+with its flag surface (reference: modelzoo/dlrm/README.md:96-113 flags:
+--ev --bf16 --optimizer --ev_filter --ev_elimination --incremental_ckpt
+--smartstaged --protocol ... mapped to this framework's features).
+
+Usage:
+  python -m deeprec_amd.models.runner --model dlrm --steps 100 --bf16
+Distributed (one process per GPU):
+  python -m torch.distributed.run --nproc-per-node N \
+      -m deeprec_amd.models.runner --model dlrm --sharded
+"""
+from __future__ import annotations
+
+import argparse
+import logging
+import os
+import time
+
+import torch
+
+from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+from deeprec_amd.embedding.options import (
+    CBFFilter, CounterFilter, EmbeddingVariableOption, GlobalStepEvict,
+    L2WeightEvict)
+from deeprec_amd.models import MODEL_REGISTRY, SEQUENCE_MODELS
+
+
+def build_argparser():
+    p = argparse.ArgumentParser()
+    p.add_argument("--model", default="dlrm", choices=sorted(MODEL_REGISTRY))
+    p.add_argument("--steps", type=int, default=100)
+    p.add_argument("--batch_size", type=int, default=4096)
+    p.add_argument("--optimizer", default="adamasync")
+    p.add_argument("--learning_rate", type=float, default=0.001)
+    p.add_argument("--bf16", action="store_true", default=True)
+    p.add_argument("--no_bf16", dest="bf16", action="store_false")
+    p.add_argument("--ev_filter", choices=["counter", "cbf", "none"],
+                   default="none")
+    p.add_argument("--filter_freq", type=int, default=3)
+    p.add_argument("--ev_elimination",
+                   choices=["gstep", "l2", "none"], default="none")
+    p.add_argument("--steps_to_live", type=int, default=10000)
+    p.add_argument("--checkpoint_dir", default=None)
+    p.add_argument("--save_steps", type=int, default=None)
+    p.add_argument("--incremental_ckpt", action="store_true")
+    p.add_argument("--smartstaged", action="store_true", default=True,
+                   help="async input staging (prefetch pipeline)")
+    p.add_argument("--no_smartstaged", dest="smartstaged",
+                   action="store_false")
+    p.add_argument("--sharded", action="store_true",
+                   help="embedding-parallel across ranks")
+    p.add_argument("--timeline", type=int, default=None)
+    p.add_argument("--log_steps", type=int, default=50)
+    p.add_argument("--seed", type=int, default=42)
+    return p
+
+
+def make_ev_option(args) -> EmbeddingVariableOption:
+    opt = EmbeddingVariableOption()
+    if args.ev_filter == "counter":
+        opt.filter_option = CounterFilter(filter_freq=args.filter_freq)
+    elif args.ev_filter == "cbf":
+        opt.filter_option = CBFFilter(filter_freq=args.filter_freq,
+                                      max_element_size=1 << 22)
+    if args.ev_elimination == "gstep":
+        opt.evict_option = GlobalStepEvict(steps_to_live=args.steps_to_live)
+    elif args.ev_elimination == "l2":
+        opt.evict_option = L2WeightEvict(l2_weight_threshold=0.01)
+    return opt
+
+
+def main(argv=None):
+    logging.basicConfig(level=logging.INFO,
+                        format="%(asctime)s %(name)s: %(message)s")
+    args = build_argparser().parse_args(argv)
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world > 1:
+        from deeprec_amd.parallel import init_distributed
+        init_distributed()
+    device = (torch.device("cuda", local_rank)
+              if torch.cuda.is_available() else torch.device("cpu"))
+    if device.type == "cuda":
+        torch.cuda.set_device(device)
+
+    torch.manual_seed(args.seed + rank)
+    is_seq = args.model in SEQUENCE_MODELS
+    model_kw = dict(device=device, bf16=args.bf16,
+                    ev_option=make_ev_option(args))
+    if args.sharded and world > 1:
+        model_kw["sharded"] = True
+    model = MODEL_REGISTRY[args.model](**model_kw)
+
+    from deeprec_amd.optimizers import make_optimizer
+    opt = make_optimizer(args.optimizer, params=model.parameters(),
+                         embedding_variables=model.embedding_variables(),
+                         learning_rate=args.learning_rate)
+    reducer = None
+    if world > 1:
+        from deeprec_amd.parallel import (DenseGradAllreducer,
+                                          broadcast_parameters)
+        broadcast_parameters(model.parameters())
+        reducer = DenseGradAllreducer(model.parameters())
+
+    ds = CriteoSyntheticDataset(batch_size=args.batch_size, device=device,
+                                seed=args.seed, rank=rank,
+                                matrix_format=not is_seq)
+    if args.smartstaged and not is_seq:
+        from deeprec_amd.data.prefetch import PrefetchIterator
+        batches = PrefetchIterator(ds, depth=2)
+    else:
+        batches = ds
+
+    from deeprec_amd.checkpoint.saver import Saver
+    from deeprec_amd.training.session import (
+        LoggingTensorHook, MonitoredTrainingSession, ProfilerHook,
+        StepCounterHook)
+    saver = Saver(module=model,
+                  embedding_variables=model.embedding_variables(),
+                  optimizer=opt, rank=rank, world_size=world)
+    hooks = [LoggingTensorHook(args.log_steps),
+             StepCounterHook(args.log_steps, args.batch_size * world)]
+    if args.timeline:
+        hooks.append(ProfilerHook(args.timeline))
+
+    def step_fn():
+        if is_seq:
+            dense, ids, seq, target, labels = ds.next_seq_batch()
+            logits = model(dense, ids[:, :model.num_sparse], seq, target)
+        else:
+            dense, ids, labels = next(it)
+            logits = model(dense, ids)
+        loss = model.loss_fn(logits, labels)
+        opt.zero_grad()
+        loss.backward()
+        if reducer is not None:
+            reducer.allreduce()
+        opt.step()
+        if isinstance(loss, torch.Tensor):
+            return {"loss": loss.detach()}
+        return {"loss": loss}
+
+    it = iter(batches) if not is_seq else None
+    t0 = time.perf_counter()
+    with MonitoredTrainingSession(
+            hooks=hooks, checkpoint_dir=args.checkpoint_dir, saver=saver,
+            save_checkpoint_steps=args.save_steps,
+            save_incremental_checkpoint_secs=(
+                30 if args.incremental_ckpt else None),
+            max_steps=args.steps) as sess:
+        while not sess.should_stop():
+            sess.run(step_fn)
+    dt = time.perf_counter() - t0
+    if rank == 0:
+        sps = args.steps * args.batch_size * world / dt
+        print(f"RESULT model={args.model} steps={args.steps} "
+              f"batch={args.batch_size} world={world} "
+              f"samples_per_sec={sps:.1f}")
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,160 @@
+"""Sequence (user-behavior) models: DIN, DIEN, BST.
+
+Reference: modelzoo/din, modelzoo/dien (GRU + attention + AUGRU),
+modelzoo/bst (transformer block). Behavior sequences are whole per-sample
+sequences (~tens of items) — the reference never shards sequence length
+(SURVEY.md §5 "long context: not present"); capability = sequence feature
+columns + these architectures.
+"""
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from deeprec_amd.data.synthetic import NUM_DENSE
+from deeprec_amd.embedding import EmbeddingVariable, embedding_lookup
+from deeprec_amd.models.common import RecModelBase, make_mlp
+
+
+class _SeqBase(RecModelBase):
+    """Adds an item EmbeddingVariable for behavior sequences + target id."""
+
+    def __init__(self, embedding_dim=16, item_dim=32, device="cpu",
+                 bf16=True, name="seq", num_sparse=10, **kw):
+        super().__init__(embedding_dim, device, bf16, name=name,
+                         num_sparse=num_sparse, **kw)
+        self.item_dim = item_dim
+        self.item_ev = EmbeddingVariable(f"{name}/items", item_dim,
+                                         device=self.device_)
+
+    def embedding_variables(self):
+        return [self.collection, self.item_ev]
+
+    def seq_emb(self, seq_ids: torch.Tensor, train=True) -> torch.Tensor:
+        """[B, T] -> [B, T, item_dim]; id 0 = padding (masked by caller)."""
+        return embedding_lookup(self.item_ev, seq_ids, train=train).float()
+
+
+class DIN(_SeqBase):
+    """Deep Interest Network (reference: modelzoo/din/train.py):
+    target-conditioned attention over the behavior sequence."""
+
+    def __init__(self, embedding_dim=16, item_dim=32, att_hidden=64,
+                 mlp_sizes=(256, 128, 64), device="cpu", bf16=True, **kw):
+        super().__init__(embedding_dim, item_dim, device, bf16, name="din",
+                         **kw)
+        self.att = nn.Sequential(
+            nn.Linear(item_dim * 4, att_hidden), nn.Sigmoid(),
+            nn.Linear(att_hidden, 1))
+        in_dim = NUM_DENSE + self.num_sparse * embedding_dim + item_dim * 2
+        self.mlp = make_mlp(list(mlp_sizes) + [1], in_dim, device, self.bf16,
+                            final_activation=False)
+        self.to(self.device_)
+
+    def attend(self, seq, target, mask):
+        """seq [B,T,D], target [B,D] -> [B,D] attention-pooled."""
+        t = target.unsqueeze(1).expand_as(seq)
+        att_in = torch.cat([seq, t, seq - t, seq * t], dim=2)
+        scores = self.att(att_in).squeeze(2)          # [B, T]
+        scores = scores.masked_fill(~mask, -1e9)
+        w = torch.softmax(scores, dim=1)
+        return (w.unsqueeze(2) * seq).sum(1)
+
+    def forward(self, dense, sparse_ids, seq_ids, target_ids, train=True):
+        emb = self.sparse_feats(sparse_ids, train)
+        seq = self.seq_emb(seq_ids, train)
+        target = embedding_lookup(self.item_ev, target_ids,
+                                  train=train).float()
+        mask = seq_ids > 0
+        interest = self.attend(seq, target, mask)
+        x = torch.cat([dense, emb.flatten(1).float(), interest, target], 1)
+        with self.amp():
+            out = self.mlp(x.to(self.compute_dtype))
+        return out.float().squeeze(1)
+
+
+class DIEN(_SeqBase):
+    """Deep Interest Evolution Network (reference: modelzoo/dien/
+    train.py:207-253): GRU interest extractor + attention-gated GRU
+    (AUGRU) interest evolution."""
+
+    def __init__(self, embedding_dim=16, item_dim=32, gru_hidden=32,
+                 att_hidden=64, mlp_sizes=(256, 128, 64), device="cpu",
+                 bf16=True, **kw):
+        super().__init__(embedding_dim, item_dim, device, bf16, name="dien",
+                         **kw)
+        self.gru = nn.GRU(item_dim, gru_hidden, batch_first=True)
+        self.att = nn.Sequential(
+            nn.Linear(gru_hidden * 2, att_hidden), nn.Sigmoid(),
+            nn.Linear(att_hidden, 1))
+        self.augru_cell = nn.GRUCell(gru_hidden, gru_hidden)
+        self.target_proj = nn.Linear(item_dim, gru_hidden)
+        in_dim = (NUM_DENSE + self.num_sparse * embedding_dim
+                  + gru_hidden + item_dim)
+        self.mlp = make_mlp(list(mlp_sizes) + [1], in_dim, device, self.bf16,
+                            final_activation=False)
+        self.to(self.device_)
+
+    def forward(self, dense, sparse_ids, seq_ids, target_ids, train=True):
+        emb = self.sparse_feats(sparse_ids, train)
+        seq = self.seq_emb(seq_ids, train)               # [B,T,Di]
+        target = embedding_lookup(self.item_ev, target_ids,
+                                  train=train).float()   # [B,Di]
+        mask = (seq_ids > 0).float()
+        h_seq, _ = self.gru(seq)                         # [B,T,H]
+        tgt_h = self.target_proj(target)                 # [B,H]
+        att_in = torch.cat(
+            [h_seq, tgt_h.unsqueeze(1).expand_as(h_seq)], 2)
+        scores = self.att(att_in).squeeze(2)
+        scores = scores.masked_fill(mask == 0, -1e9)
+        alpha = torch.softmax(scores, 1)                 # [B,T]
+        # AUGRU: attention scales the update gate — implemented as
+        # h_t = (1-a)*h_{t-1} + a*GRUCell(x_t, h_{t-1})
+        h = torch.zeros(seq.shape[0], h_seq.shape[2], device=seq.device)
+        for t in range(h_seq.shape[1]):
+            h_new = self.augru_cell(h_seq[:, t], h)
+            a = (alpha[:, t] * mask[:, t]).unsqueeze(1)
+            h = (1 - a) * h + a * h_new
+        x = torch.cat([dense, emb.flatten(1).float(), h, target], 1)
+        with self.amp():
+            out = self.mlp(x.to(self.compute_dtype))
+        return out.float().squeeze(1)
+
+
+class BST(_SeqBase):
+    """Behavior Sequence Transformer (reference: modelzoo/bst): transformer
+    encoder over [behavior sequence; target item] with learned positions."""
+
+    def __init__(self, embedding_dim=16, item_dim=32, n_heads=4,
+                 ff_dim=128, n_layers=1, max_len=65,
+                 mlp_sizes=(256, 64), device="cpu", bf16=True, **kw):
+        super().__init__(embedding_dim, item_dim, device, bf16, name="bst",
+                         **kw)
+        self.pos = nn.Parameter(torch.zeros(max_len, item_dim))
+        layer = nn.TransformerEncoderLayer(
+            d_model=item_dim, nhead=n_heads, dim_feedforward=ff_dim,
+            batch_first=True, dropout=0.0)
+        self.encoder = nn.TransformerEncoder(layer, n_layers)
+        in_dim = NUM_DENSE + self.num_sparse * embedding_dim + item_dim
+        self.mlp = make_mlp(list(mlp_sizes) + [1], in_dim, device, self.bf16,
+                            final_activation=False)
+        self.to(self.device_)
+
+    def forward(self, dense, sparse_ids, seq_ids, target_ids, train=True):
+        emb = self.sparse_feats(sparse_ids, train)
+        seq = self.seq_emb(seq_ids, train)               # [B,T,Di]
+        target = embedding_lookup(self.item_ev, target_ids,
+                                  train=train).float()
+        x_seq = torch.cat([seq, target.unsqueeze(1)], 1)  # [B,T+1,Di]
+        t1 = x_seq.shape[1]
+        x_seq = x_seq + self.pos[:t1]
+        pad = torch.cat(
+            [seq_ids <= 0,
+             torch.zeros(seq.shape[0], 1, dtype=torch.bool,
+                         device=seq.device)], 1)
+        enc = self.encoder(x_seq, src_key_padding_mask=pad)
+        pooled = enc.mean(1)
+        x = torch.cat([dense, emb.flatten(1).float(), pooled], 1)
+        with self.amp():
+            out = self.mlp(x.to(self.compute_dtype))
+        return out.float().squeeze(1)
