@@ -1,0 +1,291 @@
+"""HBM_DRAM multi-tier storage: hot tier in 288 GB HBM, cold tier in host
+pinned DRAM.
+
+Capability parity with the reference's HbmDramStorage
+(reference: hbm_dram_storage.h:37,83-105 — GPU hot values + CPU cold tier,
+batched staging CopyEmbeddingsFromDramToHbm, multi_tier_storage.cu.cc:42),
+re-designed for this engine's slot scheme:
+
+- the hash table + per-key metadata live entirely in HBM (24 B/entry —
+  cheap even for 10^10 keys);
+- value slots are monotonic: slot < hot_rows -> HBM slab row, otherwise
+  row (slot - hot_rows) of a pinned host slab. Optimizer slabs mirror the
+  same split;
+- a lookup materializes the step's unique embeddings [m, dim]: hot rows
+  gathered on-GPU, cold rows gathered on host and staged in one batched
+  pinned hipMemcpyAsync (the reference's staging design), then the fused
+  group pooling runs on the materialized rows;
+- sparse updates: hot rows via the fused HIP applies; cold rows via the
+  fp32 host applies on the pinned slabs (the cold tier is the slow path by
+  construction);
+- promotion: at shrink/compaction the hottest keys by frequency are
+  re-packed into the HBM tier (LFU placement, reference CacheStrategy).
+"""
+from __future__ import annotations
+
+import torch
+
+from deeprec_amd.embedding.options import EmbeddingVariableOption
+from deeprec_amd.ops.hip_backend import _COMBINER_ID, HbmStorage
+
+_GROW = 2
+
+
+class _ColdView:
+    """Adapter exposing the cold (host) slabs with the CpuStorage layout the
+    fp32 reference applies expect (values / dim / get_slab)."""
+
+    def __init__(self, parent: "HbmDramStorage"):
+        self._p = parent
+        self.dim = parent.dim
+
+    @property
+    def values(self):
+        return self._p.values_cold
+
+    def get_slab(self, name, width, init_value, dtype=torch.float32):
+        return self._p._cold_slab(name, width, init_value, dtype)
+
+
+class HbmDramStorage(HbmStorage):
+    def __init__(self, dim: int, ev_option: EmbeddingVariableOption,
+                 value_dtype=torch.float32, device=None, generator=None):
+        so = ev_option.storage_option
+        hot_bytes = None
+        if so.storage_size:
+            hot_bytes = so.storage_size[0]
+        super().__init__(dim, ev_option, value_dtype, device, generator)
+        row_bytes = dim * 4
+        self.hot_rows = max(1024, (hot_bytes // row_bytes)
+                            if hot_bytes else self.max_slots)
+        # keep the HBM slab at exactly the hot-tier budget
+        if self.max_slots > self.hot_rows:
+            self.values = self.values[: self.hot_rows].clone()
+        self.values_cold = torch.empty(1024, dim, dtype=torch.float32,
+                                       pin_memory=True)
+        self.cold_slabs = {}
+        self._cold_slab_init = {}
+        self.default_values_cpu = self.default_values.cpu()
+
+    # ---------------- tier plumbing ----------------
+    def _init_limit(self) -> int:
+        return self.hot_rows
+
+    def _grow_slots(self, need: int):
+        """Hot slab is fixed at hot_rows; growth goes to the cold tier."""
+        if need <= self.hot_rows:
+            return
+        cold_need = need - self.hot_rows
+        cap = self.values_cold.shape[0]
+        if cold_need <= cap:
+            return
+        new_cap = cap
+        while new_cap < cold_need:
+            new_cap *= _GROW
+        nv = torch.empty(new_cap, self.dim, dtype=torch.float32,
+                         pin_memory=True)
+        nv[:cap] = self.values_cold
+        self.values_cold = nv
+        for name, t in list(self.cold_slabs.items()):
+            nt = torch.full((new_cap, t.shape[1]),
+                            self._cold_slab_init[name], dtype=t.dtype,
+                            pin_memory=True)
+            nt[: t.shape[0]] = t
+            self.cold_slabs[name] = nt
+
+    @property
+    def max_slots(self) -> int:
+        # logical slot space = hot + cold capacity
+        base = self.values.shape[0]
+        cold = self.values_cold.shape[0] if hasattr(self, "values_cold") \
+            else 0
+        return base + cold
+
+    def _cold_slab(self, name, width, init_value, dtype=torch.float32):
+        if name not in self.cold_slabs:
+            self.cold_slabs[name] = torch.full(
+                (self.values_cold.shape[0], width), init_value, dtype=dtype,
+                pin_memory=True)
+            self._cold_slab_init[name] = init_value
+        return self.cold_slabs[name]
+
+    def get_slab(self, name, width, init_value, dtype=torch.float32):
+        if name not in self.slabs:
+            self.slabs[name] = torch.full(
+                (self.hot_rows, width), init_value, dtype=dtype,
+                device=self.device)
+            self._slab_init[name] = init_value
+        self._cold_slab(name, width, init_value, dtype)
+        return self.slabs[name]
+
+    # ---------------- lookup/create ----------------
+    def lookup_or_create(self, keys, counts, step, train=True):
+        prev = self._slots_hint
+        slots = super().lookup_or_create(keys, counts, step, train)
+        if train:
+            # host-initialize freshly-created cold rows (kernel skips them)
+            cold_new = (slots >= self.hot_rows) & (slots >= prev)
+            if bool(cold_new.any()):
+                ks = keys[cold_new].cpu()
+                ss = (slots[cold_new].cpu().long() - self.hot_rows)
+                rows = self._default_rows_cpu(ks)
+                self.values_cold[ss] = self.default_values_cpu[rows]
+                for name, t in self.cold_slabs.items():
+                    t[ss] = self._cold_slab_init[name]
+        return slots
+
+    def _default_rows_cpu(self, keys):
+        if self.key_bits > 0:
+            mask = (1 << self.key_bits) - 1
+            return ((keys >> self.key_bits) * self.dvd_per_table
+                    + (keys & mask) % self.dvd_per_table)
+        return keys % self.default_value_dim
+
+    # ---------------- materialized gather (staging) ----------------
+    def materialize(self, keys, slots) -> torch.Tensor:
+        """[m, dim] fp32 rows on GPU: hot from HBM, cold staged H2D in one
+        batched copy (≙ CopyEmbeddingsFromDramToHbm)."""
+        m = keys.numel()
+        out = torch.empty(m, self.dim, dtype=torch.float32,
+                          device=self.device)
+        hot = (slots >= 0) & (slots < self.hot_rows)
+        cold = slots >= self.hot_rows
+        none = slots < 0
+        if bool(hot.any()):
+            out[hot] = self.values[slots[hot].long()]
+        if bool(cold.any()):
+            cs = (slots[cold].cpu().long() - self.hot_rows)
+            staged = self.values_cold[cs]  # host gather (pinned source)
+            out[cold] = staged.to(self.device, non_blocking=True)
+        if bool(none.any()):
+            ku = keys[none]
+            out[none] = self.ext.ev_gather(
+                self.values, self.default_values, ku,
+                torch.full((ku.numel(),), -1, dtype=torch.int32,
+                           device=self.device),
+                self._no_permission_value(), self._use_no_permission(),
+                torch.float32)
+        return out
+
+    def gather(self, keys, slots, out_dtype=None):
+        out = self.materialize(keys, slots)
+        return out.to(out_dtype) if out_dtype else out
+
+    def pooled_lookup(self, keys, slots, inverse, offsets, row_ids, combiner,
+                      weights, out_dtype):
+        emb = self.materialize(keys, slots)
+        # reuse the direct-rows fused pooling (single-table = 1 "table")
+        comb = torch.tensor([_COMBINER_ID[combiner]], dtype=torch.int32,
+                            device=self.device)
+        return self.ext.group_pooled_fwd_direct(
+            emb, keys, inverse.to(torch.int32), offsets.to(torch.int32),
+            weights.float() if weights is not None else torch.Tensor(),
+            comb, offsets.numel() - 1, 1, out_dtype or torch.float32)
+
+    # ---------------- sparse apply split ----------------
+    def apply_split(self, name, slots, grad, hyper):
+        from deeprec_amd.ops import hip_backend, sparse_optim_cpu
+        hot = (slots >= 0) & (slots < self.hot_rows)
+        cold = slots >= self.hot_rows
+        if bool(hot.any()):
+            hip_backend.sparse_apply(name, self, slots[hot].to(torch.int32),
+                                     grad[hot], dict(hyper))
+        if bool(cold.any()):
+            cold_slots = (slots[cold].cpu().long() - self.hot_rows)
+            cold_grad = grad[cold].cpu().float()
+            getattr(sparse_optim_cpu, f"apply_{name}")(
+                _ColdView(self), cold_slots, cold_grad, **hyper)
+
+    # ---------------- export / import / shrink ----------------
+    def shrink(self, step: int) -> int:
+        from deeprec_amd.embedding.options import (GlobalStepEvict,
+                                                   L2WeightEvict)
+        eo = self.ev_option.evict_option
+        if eo is None:
+            return 0
+        names = list(self.slabs.keys())
+        keys, values, freqs, versions = self.export()
+        if keys.numel() == 0:
+            return 0
+        slab_rows = dict(zip(names, self.export_slabs(names))) \
+            if names else None
+        if isinstance(eo, GlobalStepEvict) and eo.steps_to_live > 0:
+            keep = versions >= (step - eo.steps_to_live)
+        elif isinstance(eo, L2WeightEvict) and eo.l2_weight_threshold > 0:
+            keep = values.norm(dim=1) >= eo.l2_weight_threshold
+        else:
+            return 0
+        n_evicted = int((~keep).sum())
+        if n_evicted == 0:
+            return 0
+        # rebuild both tiers from the surviving rows
+        self._alloc_table(self.capacity)
+        self.entry_counter.zero_()
+        self.slot_counter.zero_()
+        self._entries_hint = 0
+        self._slots_hint = 0
+        kc = keep.cpu()
+        self.import_(keys[keep], values[keep], freqs[keep].to(torch.int32),
+                     versions[keep],
+                     {n: r[kc] for n, r in slab_rows.items()}
+                     if slab_rows else None)
+        return n_evicted
+
+    def export(self, include_filtered: bool = False):
+        keys, slots, freqs, versions = self._export_entries()
+        adm = slots >= 0
+        k, s = keys[adm], slots[adm]
+        values = self.materialize(k, s)
+        out = (k, values, freqs[adm].to(torch.int64), versions[adm])
+        if include_filtered:
+            out = out + (keys[~adm], freqs[~adm].to(torch.int64))
+        return out
+
+    def export_slabs(self, names):
+        keys, slots, freqs, versions = self._export_entries()
+        adm = slots >= 0
+        s = slots[adm]
+        hot = s < self.hot_rows
+        outs = []
+        for nm in names:
+            t = torch.empty(s.numel(), self.slabs[nm].shape[1])
+            t[hot.cpu()] = self.slabs[nm][s[hot].long()].cpu()
+            cold_idx = (s[~hot].cpu().long() - self.hot_rows)
+            t[(~hot).cpu()] = self.cold_slabs[nm][cold_idx]
+            outs.append(t)
+        return outs
+
+    def import_(self, keys, values, freqs=None, versions=None,
+                slab_rows=None):
+        # route through lookup_or_create so tier placement stays consistent
+        keys = keys.to(self.device)
+        m = keys.numel()
+        if m == 0:
+            return
+        counts = torch.ones(m, dtype=torch.int32, device=self.device)
+        slots = self.lookup_or_create(keys, counts, step=0, train=True)
+        values = values.to(self.device, torch.float32)
+        hot = (slots >= 0) & (slots < self.hot_rows)
+        cold = slots >= self.hot_rows
+        if bool(hot.any()):
+            self.values[slots[hot].long()] = values[hot]
+        if bool(cold.any()):
+            ci = (slots[cold].cpu().long() - self.hot_rows)
+            self.values_cold[ci] = values[cold].cpu()
+        if freqs is not None:
+            self.ext.ht_insert_bulk(
+                keys, slots, freqs.to(self.device, torch.int32),
+                versions.to(self.device, torch.int64)
+                if versions is not None else torch.Tensor(),
+                self.ht_keys, self.ht_slot, self.ht_freq, self.ht_version,
+                self.entry_counter, self.error_flag)
+        if slab_rows:
+            for name, rows in slab_rows.items():
+                self.get_slab(name, rows.shape[1], 0.0)
+                if bool(hot.any()):
+                    self.slabs[name][slots[hot].long()] = \
+                        rows[hot.cpu()].to(self.device)
+                if bool(cold.any()):
+                    ci = (slots[cold].cpu().long() - self.hot_rows)
+                    self.cold_slabs[name][ci] = rows[cold.cpu()]
+        self._sync_counters()

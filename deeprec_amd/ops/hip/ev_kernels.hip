@@ -72,9 +72,9 @@ __global__ void k_lookup_insert(
     int64_t cap_mask, int32_t* __restrict__ slot_counter,
     int32_t* __restrict__ entry_counter, int max_slots,
     float* __restrict__ values, const float* __restrict__ default_values,
-    int dim, int default_value_dim, int key_bits, int filter_freq,
-    int64_t step, int train, int32_t* __restrict__ out_slots,
-    int32_t* __restrict__ error_flag) {
+    int dim, int default_value_dim, int key_bits, int init_limit,
+    int filter_freq, int64_t step, int train,
+    int32_t* __restrict__ out_slots, int32_t* __restrict__ error_flag) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= n) return;
   const int64_t key = keys[i];
@@ -108,6 +108,12 @@ __global__ void k_lookup_insert(
         return;
       }
       ht_slot[idx] = slot;
+      // multi-tier (HBM_DRAM): rows at slot >= init_limit live in the host
+      // cold slab; the host initializes those (kernel must not touch them)
+      if (slot >= init_limit) {
+        out_slots[i] = slot;
+        return;
+      }
       // composite keys (EmbeddingCollection): default row is
       // table * dvd + (raw_key % dvd); key_bits == 0 means plain keys
       int64_t dvrow;
@@ -647,8 +653,8 @@ torch::Tensor ht_lookup_insert(
     torch::Tensor ht_slot, torch::Tensor ht_freq, torch::Tensor ht_version,
     torch::Tensor slot_counter, torch::Tensor entry_counter,
     torch::Tensor values, torch::Tensor default_values, int64_t dvd_per_table,
-    int64_t key_bits, int64_t filter_freq, int64_t step, bool train,
-    torch::Tensor error_flag) {
+    int64_t key_bits, int64_t init_limit, int64_t filter_freq, int64_t step,
+    bool train, torch::Tensor error_flag) {
   CHECK_DEV(keys);
   int n = keys.numel();
   auto out = torch::empty({n}, keys.options().dtype(torch::kInt32));
@@ -664,8 +670,8 @@ torch::Tensor ht_lookup_insert(
       ht_keys.numel() - 1, slot_counter.data_ptr<int32_t>(),
       entry_counter.data_ptr<int32_t>(), values.size(0),
       values.data_ptr<float>(), default_values.data_ptr<float>(), dim,
-      (int)dvd_per_table, (int)key_bits, (int)filter_freq, step,
-      train ? 1 : 0, out.data_ptr<int32_t>(),
+      (int)dvd_per_table, (int)key_bits, (int)init_limit, (int)filter_freq,
+      step, train ? 1 : 0, out.data_ptr<int32_t>(),
       error_flag.data_ptr<int32_t>());
   return out;
 }

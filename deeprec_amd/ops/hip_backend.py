@@ -103,6 +103,11 @@ class HbmStorage:
     def max_slots(self) -> int:
         return self.values.shape[0]
 
+    def _init_limit(self) -> int:
+        """Rows the insert kernel may default-initialize; the HBM_DRAM
+        subclass caps this at the hot-tier size."""
+        return self.max_slots
+
     def _sync_counters(self):
         c = torch.stack([self.slot_counter, self.entry_counter]).cpu()
         self._slots_hint = int(c[0])
@@ -177,7 +182,8 @@ class HbmStorage:
             keys, counts_i32, self.ht_keys, self.ht_slot, self.ht_freq,
             self.ht_version, self.slot_counter, self.entry_counter,
             self.values, self.default_values, self.dvd_per_table,
-            self.key_bits, self.filter_freq, step, train, self.error_flag)
+            self.key_bits, self._init_limit(), self.filter_freq, step,
+            train, self.error_flag)
 
     def lookup(self, keys):
         slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)

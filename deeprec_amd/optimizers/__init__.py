@@ -50,7 +50,9 @@ class Optimizer:
             ev._pending_grads.clear()
 
     def _apply_sparse(self, ev, slots, grad, hyper):
-        if ev.device.type == "cuda":
+        if hasattr(ev.storage, "apply_split"):  # multi-tier HBM_DRAM
+            ev.storage.apply_split(self.sparse_name, slots, grad, hyper)
+        elif ev.device.type == "cuda":
             from deeprec_amd.ops import hip_backend
             hip_backend.sparse_apply(self.sparse_name, ev.storage, slots,
                                      grad, hyper)
