@@ -1,0 +1,202 @@
+"""feature_column API subset.
+
+Capability parity with the reference's feature-column surface used by the
+modelzoo (reference: python/feature_column/feature_column_v2.py —
+numeric_column, categorical_column_with_embedding:2080,
+categorical_column_with_hash_bucket, embedding_column,
+shared_embedding_columns, sequence_categorical_column_with_embedding,
+group_embedding_column_scope:4239, and tf.feature_column.input_layer).
+
+Design: columns are declarative specs; `input_layer(features, columns)`
+builds/looks-up the backing EmbeddingVariables. Embedding columns declared
+inside a `group_embedding_column_scope` share one EmbeddingCollection (the
+fused multi-table path).
+"""
+from __future__ import annotations
+
+import contextlib
+import threading
+from dataclasses import dataclass, field
+from typing import List, Optional, Sequence
+
+import torch
+
+from deeprec_amd.embedding.collection import EmbeddingCollection
+from deeprec_amd.embedding.options import EmbeddingVariableOption
+from deeprec_amd.embedding.ragged import RaggedIds
+from deeprec_amd.embedding.variable import get_embedding_variable
+from deeprec_amd.embedding.lookup import (
+    embedding_lookup, embedding_lookup_sparse)
+
+_SCOPE = threading.local()
+
+
+@dataclass
+class NumericColumn:
+    key: str
+    shape: int = 1
+    normalizer_fn: Optional[object] = None
+
+
+@dataclass
+class CategoricalColumn:
+    key: str
+    hash_bucket_size: Optional[int] = None   # None => EV (unbounded keys)
+    ev_option: Optional[EmbeddingVariableOption] = None
+    sequence: bool = False
+
+
+@dataclass
+class EmbeddingColumnSpec:
+    categorical: CategoricalColumn
+    dimension: int
+    combiner: str = "mean"
+    shared_name: Optional[str] = None
+    group: Optional[str] = None
+
+
+def numeric_column(key: str, shape: int = 1, normalizer_fn=None):
+    return NumericColumn(key, shape, normalizer_fn)
+
+
+def categorical_column_with_embedding(key: str, dtype=torch.int64,
+                                      ev_option=None):
+    """EV-backed categorical column (unbounded id space) — the reference's
+    signature addition over stock TF."""
+    return CategoricalColumn(key, None, ev_option)
+
+
+def categorical_column_with_hash_bucket(key: str, hash_bucket_size: int):
+    return CategoricalColumn(key, hash_bucket_size)
+
+
+def sequence_categorical_column_with_embedding(key: str, dtype=torch.int64,
+                                               ev_option=None):
+    return CategoricalColumn(key, None, ev_option, sequence=True)
+
+
+def embedding_column(categorical: CategoricalColumn, dimension: int,
+                     combiner: str = "mean"):
+    group = getattr(_SCOPE, "group", None)
+    return EmbeddingColumnSpec(categorical, dimension, combiner,
+                               group=group)
+
+
+def shared_embedding_columns(categoricals: Sequence[CategoricalColumn],
+                             dimension: int, combiner: str = "mean",
+                             shared_embedding_collection_name: str = None):
+    name = shared_embedding_collection_name or "_".join(
+        c.key for c in categoricals) + "_shared"
+    return [EmbeddingColumnSpec(c, dimension, combiner, shared_name=name)
+            for c in categoricals]
+
+
+@contextlib.contextmanager
+def group_embedding_column_scope(name: str = "group"):
+    """Embedding columns created inside share one EmbeddingCollection
+    (reference: group_embedding_column_scope, feature_column_v2.py:4239)."""
+    prev = getattr(_SCOPE, "group", None)
+    _SCOPE.group = name
+    try:
+        yield
+    finally:
+        _SCOPE.group = prev
+
+
+class InputLayer(torch.nn.Module):
+    """Materializes columns into a dense [B, total_dim] tensor.
+
+    features: dict key -> tensor ([B] / [B, k] int64 for categorical,
+    float tensor or RaggedIds for numeric/sequence).
+    """
+
+    def __init__(self, columns: Sequence, device="cpu", out_dtype=None):
+        super().__init__()
+        self.columns = list(columns)
+        self.device_ = torch.device(device)
+        self.out_dtype = out_dtype
+        self._collections = {}
+        self._evs = {}
+        # build grouped collections
+        groups = {}
+        for col in self.columns:
+            if isinstance(col, EmbeddingColumnSpec) and col.group:
+                groups.setdefault((col.group, col.dimension), []).append(col)
+        for (gname, dim), cols in groups.items():
+            self._collections[(gname, dim)] = EmbeddingCollection(
+                f"fc/{gname}/{dim}", [c.categorical.key for c in cols], dim,
+                ev_option=cols[0].categorical.ev_option,
+                combiners=[c.combiner for c in cols], device=self.device_)
+
+    def _ev_for(self, col: EmbeddingColumnSpec):
+        name = col.shared_name or f"fc/{col.categorical.key}"
+        if name not in self._evs:
+            self._evs[name] = get_embedding_variable(
+                name, col.dimension, ev_option=col.categorical.ev_option,
+                device=self.device_)
+        return self._evs[name]
+
+    def embedding_variables(self):
+        return list(self._collections.values()) + list(self._evs.values())
+
+    def forward(self, features: dict, train: bool = True) -> torch.Tensor:
+        outs: List[torch.Tensor] = []
+        group_inputs = {}
+        group_cols = {}
+        for col in self.columns:
+            if isinstance(col, NumericColumn):
+                x = features[col.key].float()
+                if col.normalizer_fn:
+                    x = col.normalizer_fn(x)
+                outs.append(x.reshape(x.shape[0], -1))
+            elif isinstance(col, EmbeddingColumnSpec):
+                key = col.categorical.key
+                ids = features[key]
+                if col.group:
+                    group_inputs.setdefault((col.group, col.dimension),
+                                            []).append(self._as_ragged(
+                                                ids, col))
+                    group_cols.setdefault((col.group, col.dimension),
+                                          []).append(col)
+                    outs.append(None)  # placeholder, filled below
+                else:
+                    outs.append(self._single(col, ids, train))
+        # grouped lookups (one fused call per collection)
+        fills = {}
+        for gk, sp_list in group_inputs.items():
+            coll = self._collections[gk]
+            cat = coll.lookup(sp_list, out_dtype=self.out_dtype, train=train)
+            dim = gk[1]
+            for i, col in enumerate(group_cols[gk]):
+                fills[id(col)] = cat[:, i * dim:(i + 1) * dim]
+        gi = 0
+        for i, col in enumerate(self.columns):
+            if outs[i] is None:
+                outs[i] = fills[id(col)]
+                gi += 1
+        ref = next(o for o in outs if o is not None)
+        return torch.cat([o.to(ref.dtype) for o in outs], dim=1)
+
+    def _as_ragged(self, ids, col) -> RaggedIds:
+        if isinstance(ids, RaggedIds):
+            return ids
+        if ids.dim() == 1:
+            ids = ids.unsqueeze(1)
+        if col.categorical.hash_bucket_size:
+            ids = ids % col.categorical.hash_bucket_size
+        return RaggedIds.from_dense(ids)
+
+    def _single(self, col, ids, train):
+        ev = self._ev_for(col)
+        if col.categorical.sequence:
+            return embedding_lookup(ev, ids, train=train).flatten(1)
+        sp = self._as_ragged(ids, col)
+        return embedding_lookup_sparse(ev, sp, combiner=col.combiner,
+                                       train=train)
+
+
+def input_layer(features: dict, columns: Sequence, device="cpu",
+                train: bool = True) -> torch.Tensor:
+    """One-shot functional form (builds EVs on first use via the registry)."""
+    layer = InputLayer(columns, device)
+    return layer(features, train)
