@@ -56,7 +56,7 @@ class HbmDramStorage(HbmStorage):
             hot_bytes = so.storage_size[0]
         super().__init__(dim, ev_option, value_dtype, device, generator)
         row_bytes = dim * 4
-        self.hot_rows = max(1024, (hot_bytes // row_bytes)
+        self.hot_rows = max(16, (hot_bytes // row_bytes)
                             if hot_bytes else self.max_slots)
         # keep the HBM slab at exactly the hot-tier budget
         if self.max_slots > self.hot_rows:
