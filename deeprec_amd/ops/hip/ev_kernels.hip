@@ -652,9 +652,10 @@ torch::Tensor ht_lookup_insert(
     torch::Tensor keys, torch::Tensor counts, torch::Tensor ht_keys,
     torch::Tensor ht_slot, torch::Tensor ht_freq, torch::Tensor ht_version,
     torch::Tensor slot_counter, torch::Tensor entry_counter,
-    torch::Tensor values, torch::Tensor default_values, int64_t dvd_per_table,
-    int64_t key_bits, int64_t init_limit, int64_t filter_freq, int64_t step,
-    bool train, torch::Tensor error_flag) {
+    torch::Tensor values, torch::Tensor default_values, int64_t max_slots,
+    int64_t dvd_per_table, int64_t key_bits, int64_t init_limit,
+    int64_t filter_freq, int64_t step, bool train,
+    torch::Tensor error_flag) {
   CHECK_DEV(keys);
   int n = keys.numel();
   auto out = torch::empty({n}, keys.options().dtype(torch::kInt32));
@@ -668,7 +669,7 @@ torch::Tensor ht_lookup_insert(
       n, ht_keys.data_ptr<int64_t>(), ht_slot.data_ptr<int32_t>(),
       ht_freq.data_ptr<int32_t>(), ht_version.data_ptr<int64_t>(),
       ht_keys.numel() - 1, slot_counter.data_ptr<int32_t>(),
-      entry_counter.data_ptr<int32_t>(), values.size(0),
+      entry_counter.data_ptr<int32_t>(), (int)max_slots,
       values.data_ptr<float>(), default_values.data_ptr<float>(), dim,
       (int)dvd_per_table, (int)key_bits, (int)init_limit, (int)filter_freq,
       step, train ? 1 : 0, out.data_ptr<int32_t>(),

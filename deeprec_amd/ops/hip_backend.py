@@ -181,9 +181,9 @@ class HbmStorage:
         return self.ext.ht_lookup_insert(
             keys, counts_i32, self.ht_keys, self.ht_slot, self.ht_freq,
             self.ht_version, self.slot_counter, self.entry_counter,
-            self.values, self.default_values, self.dvd_per_table,
-            self.key_bits, self._init_limit(), self.filter_freq, step,
-            train, self.error_flag)
+            self.values, self.default_values, self.max_slots,
+            self.dvd_per_table, self.key_bits, self._init_limit(),
+            self.filter_freq, step, train, self.error_flag)
 
     def lookup(self, keys):
         slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)
