@@ -120,11 +120,16 @@ class HbmDramStorage(HbmStorage):
 
     # ---------------- lookup/create ----------------
     def lookup_or_create(self, keys, counts, step, train=True):
-        prev = self._slots_hint
+        # exact pre-call slot watermark: slots >= prev are new this call
+        # (cold tier tolerates the D2H sync; the hot-only path never pays it
+        # because new slots only exceed hot_rows once the hot tier is full)
+        prev = (int(self.slot_counter.cpu())
+                if self._slots_hint >= self.hot_rows else 0)
         slots = super().lookup_or_create(keys, counts, step, train)
-        if train:
+        if train and prev >= 0:
             # host-initialize freshly-created cold rows (kernel skips them)
-            cold_new = (slots >= self.hot_rows) & (slots >= prev)
+            cold_new = (slots >= self.hot_rows) & (slots >= max(
+                self.hot_rows, prev))
             if bool(cold_new.any()):
                 ks = keys[cold_new].cpu()
                 ss = (slots[cold_new].cpu().long() - self.hot_rows)
