@@ -1,0 +1,115 @@
+"""Estimator facade.
+
+Capability parity with the reference's tf.estimator surface used by the
+modelzoo (train/evaluate/predict loops over a model_fn, model_dir
+checkpointing, RunConfig). The eager-first contract:
+
+    def model_fn(params) -> (model, optimizer)
+    def input_fn() -> iterable of (features..., labels)
+
+Estimator wires MonitoredTrainingSession + Saver + hooks around them.
+Micro-batch gradient accumulation (reference: micro_batch_num,
+graph_execution_state.cc:635-727 graph-duplication) maps to N sub-batch
+backward passes with merged sparse grads before one optimizer step.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import Callable, Optional
+
+import torch
+
+from deeprec_amd.checkpoint.saver import Saver
+from deeprec_amd.training.session import (
+    LoggingTensorHook, MonitoredTrainingSession, StepCounterHook)
+
+
+@dataclass
+class RunConfig:
+    save_checkpoints_steps: Optional[int] = None
+    save_checkpoints_secs: Optional[float] = None
+    save_incremental_checkpoint_secs: Optional[float] = None
+    keep_checkpoint_max: int = 5
+    log_step_count_steps: int = 100
+
+
+class Estimator:
+    def __init__(self, model_fn: Callable, model_dir: str = None,
+                 config: Optional[RunConfig] = None, params: dict = None):
+        self.model_dir = model_dir
+        self.config = config or RunConfig()
+        self.params = params or {}
+        self.model, self.optimizer = model_fn(self.params)
+        self.saver = Saver(
+            module=self.model,
+            embedding_variables=self.model.embedding_variables(),
+            optimizer=self.optimizer,
+            keep_checkpoint_max=self.config.keep_checkpoint_max)
+
+    def _step(self, batch, micro_batch: int = 1):
+        *features, labels = batch
+        self.optimizer.zero_grad()
+        if micro_batch <= 1:
+            logits = self.model(*features)
+            loss = self.model.loss_fn(logits, labels)
+            loss.backward()
+        else:
+            total = 0.0
+            bsz = labels.shape[0] // micro_batch
+            for i in range(micro_batch):
+                sl = slice(i * bsz, (i + 1) * bsz)
+                f_i = [x[sl] if torch.is_tensor(x) else x for x in features]
+                lg = self.model(*f_i)
+                sub = self.model.loss_fn(lg, labels[sl]) / micro_batch
+                sub.backward()
+                total += float(sub)
+            loss = torch.tensor(total)
+        self.optimizer.step()
+        return {"loss": loss.detach() if torch.is_tensor(loss) else loss}
+
+    def train(self, input_fn: Callable, steps: Optional[int] = None,
+              max_steps: Optional[int] = None, hooks=None,
+              micro_batch: int = 1):
+        it = iter(input_fn())
+        from deeprec_amd.embedding.variable import get_global_step
+        stop_at = max_steps or (get_global_step() + (steps or 1))
+        all_hooks = list(hooks or []) + [
+            LoggingTensorHook(self.config.log_step_count_steps),
+            StepCounterHook(self.config.log_step_count_steps)]
+        with MonitoredTrainingSession(
+                hooks=all_hooks, checkpoint_dir=self.model_dir,
+                saver=self.saver if self.model_dir else None,
+                save_checkpoint_steps=self.config.save_checkpoints_steps,
+                save_checkpoint_secs=self.config.save_checkpoints_secs,
+                save_incremental_checkpoint_secs=(
+                    self.config.save_incremental_checkpoint_secs),
+                max_steps=stop_at) as sess:
+            while not sess.should_stop():
+                sess.run(lambda: self._step(next(it), micro_batch))
+        return self
+
+    @torch.no_grad()
+    def evaluate(self, input_fn: Callable, steps: int = 10) -> dict:
+        it = iter(input_fn())
+        losses, n_correct, n_total = [], 0, 0
+        for _ in range(steps):
+            *features, labels = next(it)
+            logits = self.model(*features, train=False)
+            if isinstance(logits, (list, tuple)):
+                logits = logits[0]
+            losses.append(float(self.model.loss_fn(logits, labels)))
+            preds = (torch.sigmoid(logits) > 0.5).float()
+            n_correct += int((preds == labels).sum())
+            n_total += labels.numel()
+        return {"loss": sum(losses) / len(losses),
+                "accuracy": n_correct / max(n_total, 1)}
+
+    @torch.no_grad()
+    def predict(self, input_fn: Callable):
+        for batch in input_fn():
+            features = batch[:-1] if isinstance(batch, tuple) else batch
+            logits = self.model(*features, train=False)
+            if isinstance(logits, (list, tuple)):
+                yield [torch.sigmoid(lg) for lg in logits]
+            else:
+                yield torch.sigmoid(logits)

@@ -162,3 +162,33 @@ def test_monitored_session_with_hooks(tmp_path):
         while not sess.should_stop():
             sess.run(step_fn2)
     assert GLOBAL_STEP.value == 10
+
+
+def test_estimator_train_eval(tmp_path):
+    from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.training.estimator import Estimator, RunConfig
+
+    def model_fn(params):
+        m = DLRM(device="cpu", bf16=False)
+        opt = AdagradOptimizer(params=m.parameters(),
+                               embedding_variables=m.embedding_variables(),
+                               learning_rate=0.05)
+        return m, opt
+
+    def input_fn():
+        ds = CriteoSyntheticDataset(batch_size=32, seed=1,
+                                    matrix_format=True)
+        return iter(ds)
+
+    est = Estimator(model_fn, model_dir=str(tmp_path),
+                    config=RunConfig(save_checkpoints_steps=4,
+                                     log_step_count_steps=5))
+    est.train(input_fn, steps=6, micro_batch=2)
+    metrics = est.evaluate(input_fn, steps=2)
+    assert "loss" in metrics and metrics["loss"] > 0
+    from deeprec_amd.checkpoint.saver import latest_checkpoint
+    assert latest_checkpoint(str(tmp_path)) is not None
+    preds = next(est.predict(input_fn))
+    assert preds.shape == (32,)
