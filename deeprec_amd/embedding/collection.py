@@ -142,14 +142,11 @@ class EmbeddingCollection:
         assert len(sp_list) == self.n_tables
         batch, values_cat, offsets_cat, row_ids_cat, weights_cat = \
             self._concat_inputs(sp_list)
-        uniq, inverse, counts = torch.unique(
-            values_cat, return_inverse=True, return_counts=True)
         train = train and self.trainable
-        slots = self.storage.lookup_or_create(
-            uniq, counts, get_global_step(), train=train)
+        uniq, inverse, counts, slots = self._dedup_and_probe(
+            values_cat, train)
         if train and self._record_sparse_ids:
             self._recorded_ids.append(uniq.detach())
-        inverse = inverse.to(torch.int32)
         if not train:
             return self._forward(uniq, slots, inverse, offsets_cat,
                                  weights_cat, batch, out_dtype)
@@ -164,18 +161,31 @@ class EmbeddingCollection:
 
     _CHUNK = 128
 
+    def _dedup_and_probe(self, values_cat, train):
+        """unique + single hash probe. GPU training uses the fused hash
+        dedup (no sorts); other paths use torch.unique + the probe."""
+        if train and hasattr(self.storage, "dedup_lookup"):
+            return self.storage.dedup_lookup(values_cat, get_global_step())
+        uniq, inverse, counts = torch.unique(
+            values_cat, return_inverse=True, return_counts=True)
+        slots = self.storage.lookup_or_create(
+            uniq, counts, get_global_step(), train=train)
+        return uniq, inverse.to(torch.int32), counts, slots
+
     def _prep_backward(self, inverse, counts):
         """CSR over unique keys + fixed-size occurrence chunks so the
         backward is balanced under zipf-hot keys (no 10k-iteration
         threads)."""
         m = counts.numel()
         dev = self.device
-        order = torch.argsort(inverse).to(torch.int32)
         c32 = counts.to(torch.int32)
         bounds = torch.zeros(m + 1, dtype=torch.int32, device=dev)
         bounds[1:] = c32.cumsum(0)
         if self.device.type != "cuda":
+            order = torch.argsort(inverse.long()).to(torch.int32)
             return order, bounds, None, None
+        # sort-free CSR build (per-key cursors)
+        order = self.storage.ext.csr_order(inverse, bounds, m)
         nch = (counts + (self._CHUNK - 1)) // self._CHUNK
         chunk_u = torch.repeat_interleave(
             torch.arange(m, device=dev, dtype=torch.int64), nch)
@@ -210,14 +220,11 @@ class EmbeddingCollection:
             }
             self._matrix_cache[batch] = cache
         values_cat = ids.t().reshape(-1) + cache["tags"]
-        uniq, inverse, counts = torch.unique(
-            values_cat, return_inverse=True, return_counts=True)
         train = train and self.trainable
-        slots = self.storage.lookup_or_create(
-            uniq, counts, get_global_step(), train=train)
+        uniq, inverse, counts, slots = self._dedup_and_probe(
+            values_cat, train)
         if train and self._record_sparse_ids:
             self._recorded_ids.append(uniq.detach())
-        inverse = inverse.to(torch.int32)
         if not train:
             return self._forward(uniq, slots, inverse, cache["offsets"],
                                  None, batch, out_dtype)

@@ -119,25 +119,37 @@ class HbmDramStorage(HbmStorage):
         return self.slabs[name]
 
     # ---------------- lookup/create ----------------
-    def lookup_or_create(self, keys, counts, step, train=True):
+    def _watermark(self) -> int:
         # exact pre-call slot watermark: slots >= prev are new this call
         # (cold tier tolerates the D2H sync; the hot-only path never pays it
         # because new slots only exceed hot_rows once the hot tier is full)
-        prev = (int(self.slot_counter.cpu())
+        return (int(self.slot_counter.cpu())
                 if self._slots_hint >= self.hot_rows else 0)
+
+    def _init_cold_rows(self, keys, slots, prev):
+        """Host-initialize freshly-created cold rows (kernels skip them)."""
+        cold_new = (slots >= self.hot_rows) & (slots >= max(
+            self.hot_rows, prev))
+        if bool(cold_new.any()):
+            ks = keys[cold_new].cpu()
+            ss = (slots[cold_new].cpu().long() - self.hot_rows)
+            rows = self._default_rows_cpu(ks)
+            self.values_cold[ss] = self.default_values_cpu[rows]
+            for name, t in self.cold_slabs.items():
+                t[ss] = self._cold_slab_init[name]
+
+    def lookup_or_create(self, keys, counts, step, train=True):
+        prev = self._watermark()
         slots = super().lookup_or_create(keys, counts, step, train)
-        if train and prev >= 0:
-            # host-initialize freshly-created cold rows (kernel skips them)
-            cold_new = (slots >= self.hot_rows) & (slots >= max(
-                self.hot_rows, prev))
-            if bool(cold_new.any()):
-                ks = keys[cold_new].cpu()
-                ss = (slots[cold_new].cpu().long() - self.hot_rows)
-                rows = self._default_rows_cpu(ks)
-                self.values_cold[ss] = self.default_values_cpu[rows]
-                for name, t in self.cold_slabs.items():
-                    t[ss] = self._cold_slab_init[name]
+        if train:
+            self._init_cold_rows(keys, slots, prev)
         return slots
+
+    def dedup_lookup(self, values_cat, step):
+        prev = self._watermark()
+        uniq, inverse, counts, slots = super().dedup_lookup(values_cat, step)
+        self._init_cold_rows(uniq, slots, prev)
+        return uniq, inverse, counts, slots
 
     def _default_rows_cpu(self, keys):
         if self.key_bits > 0:

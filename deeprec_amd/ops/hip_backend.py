@@ -94,6 +94,12 @@ class HbmStorage:
                                    device=self.device)
         self.ht_version = torch.full((capacity,), -1, dtype=torch.int64,
                                      device=self.device)
+        # per-entry epoch stamp + compact index for the fused dedup path
+        self.ht_epoch = torch.zeros(capacity, dtype=torch.int32,
+                                    device=self.device)
+        self.ht_compact = torch.zeros(capacity, dtype=torch.int32,
+                                      device=self.device)
+        self._epoch = getattr(self, "_epoch", 0)
 
     def _alloc_slabs(self, max_slots: int):
         self.values = torch.empty(max_slots, self.dim, dtype=torch.float32,
@@ -188,6 +194,31 @@ class HbmStorage:
     def lookup(self, keys):
         slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)
         return slots
+
+    def dedup_lookup(self, values_cat: torch.Tensor, step: int):
+        """Fused unique+probe for a training step: raw (duplicated) keys in,
+        (uniq, inverse i32, counts i32, slots i32) out. One host sync (the
+        unique count) — the same sync torch.unique pays — and no sorts."""
+        nnz = values_cat.numel()
+        self._ensure_capacity(nnz)
+        self._epoch += 1
+        uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
+        self.ext.ht_dedup_a(values_cat, self.ht_keys, self.ht_freq,
+                            self.ht_version, self.ht_epoch, self.ht_compact,
+                            self._epoch, step, self.entry_counter, m_counter,
+                            uniq_buf, centry_buf, self.error_flag)
+        m = int(m_counter.cpu())
+        uniq = uniq_buf[:m]
+        slots = self.ext.ht_dedup_b(
+            centry_buf[:m], uniq, self.ht_slot, self.ht_freq,
+            self.slot_counter, self.max_slots, self.values,
+            self.default_values, self.dvd_per_table, self.key_bits,
+            self._init_limit(), self.filter_freq, self.error_flag)
+        inverse, counts = self.ext.ht_dedup_c(values_cat, self.ht_keys,
+                                              self.ht_compact, m)
+        return uniq, inverse, counts, slots
 
     def _use_no_permission(self) -> bool:
         return (self.filter_freq > 0 and
