@@ -95,7 +95,9 @@ __device__ __forceinline__ bf16x8 load_frag_col(const short* __restrict__ src,
 }
 
 // ------------------------------------------------------------------
-// fwd: C[M,N] = act(A[M,K] @ W[N,K]^T + bias); 4 waves/block stacked on M
+// fwd: C[M,N] = act(A[M,K] @ W[N,K]^T + bias); 4 waves/block stacked on M,
+// each wave computes 16 rows x 64 cols (4 n-tiles) so the A fragment is
+// loaded once per 4 MFMAs.
 // ------------------------------------------------------------------
 __global__ void k_linear_fwd(const short* __restrict__ A,
                              const short* __restrict__ W,
@@ -103,31 +105,41 @@ __global__ void k_linear_fwd(const short* __restrict__ A,
                              int K, int relu, short* __restrict__ C) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int tiles_n = (N + 15) / 16;
-  // block tile: 64 rows of M x 16 cols of N
-  const int m0 = (blockIdx.x / tiles_n) * 64 + wave * 16;
-  const int n0 = (blockIdx.x % tiles_n) * 16;
+  const int ntiles64 = (N + 63) / 64;
+  const int m0 = (blockIdx.x / ntiles64) * 64 + wave * 16;
+  const int n0 = (blockIdx.x % ntiles64) * 64;
   if (m0 >= M) return;
   const int row_a = m0 + (lane & 15);
   const int col_b = n0 + (lane & 15);
   const int kgrp = (lane >> 4) * 8;
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                  {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  const int nt = min(4, (N - n0 + 15) / 16);
   for (int k0 = 0; k0 < K; k0 += 32) {
     bf16x8 a = load_frag_row(A, row_a, k0 + kgrp, M, K);
-    bf16x8 b = load_frag_row(W, col_b, k0 + kgrp, N, K);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      if (t >= nt) break;
+      bf16x8 b = load_frag_row(W, col_b + t * 16, k0 + kgrp, N, K);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0,
+                                                       0);
+    }
   }
   // D: col = lane%16, row = 4*(lane/16) + i
-  const int cn = n0 + (lane & 15);
-  if (cn >= N) return;
-  const float bv = bias ? bias[cn] : 0.0f;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    int cm = m0 + (lane >> 4) * 4 + i;
-    if (cm >= M) continue;
-    float v = acc[i] + bv;
-    if (relu && v < 0.0f) v = 0.0f;
-    C[(int64_t)cm * N + cn] = f2bf_u16(v);
+  for (int t = 0; t < 4; ++t) {
+    if (t >= nt) break;
+    const int cn = n0 + t * 16 + (lane & 15);
+    if (cn >= N) continue;
+    const float bv = bias ? bias[cn] : 0.0f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int cm = m0 + (lane >> 4) * 4 + i;
+      if (cm >= M) continue;
+      float v = acc[t][i] + bv;
+      if (relu && v < 0.0f) v = 0.0f;
+      C[(int64_t)cm * N + cn] = f2bf_u16(v);
+    }
   }
 }
 
@@ -139,26 +151,37 @@ __global__ void k_linear_dx(const short* __restrict__ G,
                             short* __restrict__ dX) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int tiles_k = (K + 15) / 16;
-  const int m0 = (blockIdx.x / tiles_k) * 64 + wave * 16;
-  const int c0 = (blockIdx.x % tiles_k) * 16;
+  const int ktiles64 = (K + 63) / 64;
+  const int m0 = (blockIdx.x / ktiles64) * 64 + wave * 16;
+  const int c0 = (blockIdx.x % ktiles64) * 64;
   if (m0 >= M) return;
   const int row_g = m0 + (lane & 15);
   const int col_w = c0 + (lane & 15);
   const int kgrp = (lane >> 4) * 8;
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
+                  {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  const int kt = min(4, (K - c0 + 15) / 16);
   for (int n0 = 0; n0 < N; n0 += 32) {
     bf16x8 a = load_frag_row(G, row_g, n0 + kgrp, M, N);
-    bf16x8 b = load_frag_col(W, col_w, n0 + kgrp, N, K);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-  }
-  const int ck = c0 + (lane & 15);
-  if (ck >= K) return;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    int cm = m0 + (lane >> 4) * 4 + i;
-    if (cm >= M) continue;
-    dX[(int64_t)cm * K + ck] = f2bf_u16(acc[i]);
+    for (int t = 0; t < 4; ++t) {
+      if (t >= kt) break;
+      bf16x8 b = load_frag_col(W, col_w + t * 16, n0 + kgrp, N, K);
+      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0,
+                                                       0);
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    if (t >= kt) break;
+    const int ck = c0 + t * 16 + (lane & 15);
+    if (ck >= K) continue;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int cm = m0 + (lane >> 4) * 4 + i;
+      if (cm >= M) continue;
+      dX[(int64_t)cm * K + ck] = f2bf_u16(acc[t][i]);
+    }
   }
 }
 
@@ -418,7 +441,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w_bf16,
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
   int M = x.size(0), K = x.size(1), N = w_bf16.size(0);
   auto out = torch::empty({M, N}, x.options());
-  int tiles_n = (N + 15) / 16;
+  int tiles_n = (N + 63) / 64;
   int blocks = ((M + 63) / 64) * tiles_n;
   k_linear_fwd<<<blocks, 256, 0, dense_stream()>>>(
       bf_ptr(x), bf_ptr(w_bf16),
@@ -430,7 +453,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w_bf16,
 torch::Tensor linear_dx(torch::Tensor g, torch::Tensor w_bf16) {
   int M = g.size(0), N = g.size(1), K = w_bf16.size(1);
   auto dx = torch::empty({M, K}, g.options());
-  int tiles_k = (K + 15) / 16;
+  int tiles_k = (K + 63) / 64;
   int blocks = ((M + 63) / 64) * tiles_k;
   k_linear_dx<<<blocks, 256, 0, dense_stream()>>>(
       bf_ptr(g), bf_ptr(w_bf16), M, N, K, bf_ptr_mut(dx));
