@@ -37,10 +37,10 @@ class _FusedLinear(torch.autograd.Function):
         if ctx.relu:
             g = ext.relu_bwd(g, out)
         dx = ext.linear_dx(g, w16)
-        # transpose-then-row-load dW: two cheap contiguous-copy transposes
-        # buy fully vectorized fragments (vs 8 scalar loads per fragment)
-        dw, db = ext.linear_dw_nt(g.t().contiguous(), x16.t().contiguous(),
-                                  True)
+        # direct col-fragment dW: measured faster end-to-end than the
+        # transpose-then-row-load variant (torch .t().contiguous() costs
+        # ~12us/copy, more than the strided-fragment penalty it removes)
+        dw, db = ext.linear_dw(g, x16, True)
         return dx.to(ctx.x_dtype), dw, db, None
 
 

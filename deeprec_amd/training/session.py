@@ -215,3 +215,22 @@ class MonitoredTrainingSession:
         for h in self.hooks:
             h.after_run(self, results)
         return results
+
+
+class MemoryStatsHook(SessionRunHook):
+    """Log device memory accounting every N steps (the reference's
+    GPU-memory-optimization observability: allocator stats vs peak)."""
+
+    def __init__(self, every_n_steps: int = 500):
+        self.every_n = every_n_steps
+
+    def after_run(self, session, results):
+        if not torch.cuda.is_available():
+            return
+        step = get_global_step()
+        if step % self.every_n == 0:
+            alloc = torch.cuda.memory_allocated() / (1 << 20)
+            reserved = torch.cuda.memory_reserved() / (1 << 20)
+            peak = torch.cuda.max_memory_allocated() / (1 << 20)
+            log.info("memory MiB: allocated=%.1f reserved=%.1f peak=%.1f",
+                     alloc, reserved, peak)

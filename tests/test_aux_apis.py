@@ -148,3 +148,28 @@ def test_prefetch_iterator_cpu():
     for _ in range(3):
         dense, ids, labels = next(it)
         assert dense.shape == (8, 13) and ids.shape == (8, 26)
+
+
+def test_streaming_auc_and_accuracy():
+    from deeprec_amd.training.metrics import StreamingAccuracy, StreamingAUC
+    torch.manual_seed(0)
+    auc = StreamingAUC(num_thresholds=2000)
+    acc = StreamingAccuracy()
+    # separable data -> AUC near 1; random -> near 0.5
+    labels = torch.cat([torch.ones(500), torch.zeros(500)])
+    probs = torch.cat([torch.rand(500) * 0.4 + 0.6, torch.rand(500) * 0.4])
+    auc.update(probs, labels)
+    acc.update(probs, labels)
+    assert auc.result() > 0.95
+    assert acc.result() > 0.95
+    auc2 = StreamingAUC()
+    auc2.update(torch.rand(4000), (torch.rand(4000) < 0.5).float())
+    assert 0.4 < auc2.result() < 0.6
+
+
+def test_quantize_embeddings_roundtrip():
+    from tools.quantize_embeddings import dequantize_rows, quantize_rows
+    v = torch.randn(100, 16)
+    q, scale = quantize_rows(v)
+    v2 = dequantize_rows(q, scale)
+    assert (v - v2).abs().max() < v.abs().max() / 100
