@@ -68,9 +68,17 @@ def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,
     if isinstance(ev, ShardedEmbeddingVariable):
         return sharded_embedding_lookup_sparse(ev, sp_ids, combiner,
                                                out_dtype, train)
-    uniq, inverse, counts = torch.unique(
-        sp_ids.values, return_inverse=True, return_counts=True)
-    slots = ev.lookup_or_create(uniq, counts, train=train)
+    if train and ev.trainable and hasattr(ev.storage, "dedup_lookup"):
+        # fused hash dedup (sort-free) on the GPU training path
+        from deeprec_amd.embedding.variable import get_global_step
+        uniq, inverse, counts, slots = ev.storage.dedup_lookup(
+            sp_ids.values, get_global_step())
+        if ev._record_sparse_ids:
+            ev._recorded_ids.append(uniq.detach())
+    else:
+        uniq, inverse, counts = torch.unique(
+            sp_ids.values, return_inverse=True, return_counts=True)
+        slots = ev.lookup_or_create(uniq, counts, train=train)
     row_ids = sp_ids.row_ids()
     if not (train and ev.trainable):
         emb = ev.storage.gather(uniq, slots)
