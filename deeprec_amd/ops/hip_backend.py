@@ -209,7 +209,14 @@ class HbmStorage:
                             self.ht_version, self.ht_epoch, self.ht_compact,
                             self._epoch, step, self.entry_counter, m_counter,
                             uniq_buf, centry_buf, self.error_flag)
-        m = int(m_counter.cpu())
+        # one D2H sync (as torch.unique pays); piggyback the real counters
+        # so capacity hints don't inflate by nnz per step (which caused
+        # needless rehashes on high-uniqueness workloads)
+        c = torch.cat([m_counter, self.entry_counter,
+                       self.slot_counter]).cpu()
+        m = int(c[0])
+        self._entries_hint = int(c[1])
+        self._slots_hint = int(c[2]) + m  # pass B may admit up to m slots
         uniq = uniq_buf[:m]
         slots = self.ext.ht_dedup_b(
             centry_buf[:m], uniq, self.ht_slot, self.ht_freq,
