@@ -68,7 +68,9 @@ def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,
     if isinstance(ev, ShardedEmbeddingVariable):
         return sharded_embedding_lookup_sparse(ev, sp_ids, combiner,
                                                out_dtype, train)
-    if train and ev.trainable and hasattr(ev.storage, "dedup_lookup"):
+    import os
+    if (train and ev.trainable and hasattr(ev.storage, "dedup_lookup")
+            and not os.environ.get("DEEPREC_AMD_DISABLE_DEDUP")):
         # fused hash dedup (sort-free) on the GPU training path
         from deeprec_amd.embedding.variable import get_global_step
         uniq, inverse, counts, slots = ev.storage.dedup_lookup(
