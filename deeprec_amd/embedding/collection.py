@@ -164,12 +164,16 @@ class EmbeddingCollection:
     def _dedup_and_probe(self, values_cat, train):
         """unique + single hash probe. GPU training uses the fused hash
         dedup (no sorts); other paths use torch.unique + the probe."""
-        if train and hasattr(self.storage, "dedup_lookup"):
+        if (train and hasattr(self.storage, "dedup_lookup")
+                and self.storage.prefers_dedup()):
             return self.storage.dedup_lookup(values_cat, get_global_step())
         uniq, inverse, counts = torch.unique(
             values_cat, return_inverse=True, return_counts=True)
         slots = self.storage.lookup_or_create(
             uniq, counts, get_global_step(), train=train)
+        if hasattr(self.storage, "observe_uniq_ratio"):
+            self.storage.observe_uniq_ratio(uniq.numel(),
+                                            values_cat.numel())
         return uniq, inverse.to(torch.int32), counts, slots
 
     def _prep_backward(self, inverse, counts):

@@ -82,6 +82,11 @@ class HbmStorage:
         # host-side mirrors, refreshed lazily (avoid per-step D2H sync)
         self._entries_hint = 0
         self._slots_hint = 0
+        # adaptive dedup-vs-sort choice: hash dedup wins on duplication-
+        # heavy batches (zipf id streams); sort-based unique wins when keys
+        # are mostly distinct (long uniform sequences). Track the observed
+        # unique ratio and pick per step.
+        self._uniq_ratio = None
 
     # ---------------- allocation ----------------
     def _alloc_table(self, capacity: int):
@@ -195,6 +200,14 @@ class HbmStorage:
         slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)
         return slots
 
+    def prefers_dedup(self) -> bool:
+        return self._uniq_ratio is None or self._uniq_ratio < 0.5
+
+    def observe_uniq_ratio(self, m: int, nnz: int):
+        r = m / max(nnz, 1)
+        self._uniq_ratio = (r if self._uniq_ratio is None
+                            else 0.7 * self._uniq_ratio + 0.3 * r)
+
     def dedup_lookup(self, values_cat: torch.Tensor, step: int):
         """Fused unique+probe for a training step: raw (duplicated) keys in,
         (uniq, inverse i32, counts i32, slots i32) out. One host sync (the
@@ -217,6 +230,7 @@ class HbmStorage:
         m = int(c[0])
         self._entries_hint = int(c[1])
         self._slots_hint = int(c[2]) + m  # pass B may admit up to m slots
+        self.observe_uniq_ratio(m, nnz)
         uniq = uniq_buf[:m]
         slots = self.ext.ht_dedup_b(
             centry_buf[:m], uniq, self.ht_slot, self.ht_freq,
