@@ -43,9 +43,15 @@ class DIN(_SeqBase):
                  mlp_sizes=(256, 128, 64), device="cpu", bf16=True, **kw):
         super().__init__(embedding_dim, item_dim, device, bf16, name="din",
                          **kw)
-        self.att = nn.Sequential(
-            nn.Linear(item_dim * 4, att_hidden), nn.Sigmoid(),
-            nn.Linear(att_hidden, 1))
+        if self.bf16:
+            from deeprec_amd.ops.fused_mlp import FusedLinear
+            self.att = nn.Sequential(
+                FusedLinear(item_dim * 4, att_hidden, activation="sigmoid"),
+                FusedLinear(att_hidden, 1, activation=None))
+        else:
+            self.att = nn.Sequential(
+                nn.Linear(item_dim * 4, att_hidden), nn.Sigmoid(),
+                nn.Linear(att_hidden, 1))
         in_dim = NUM_DENSE + self.num_sparse * embedding_dim + item_dim * 2
         self.mlp = make_mlp(list(mlp_sizes) + [1], in_dim, device, self.bf16,
                             final_activation=False)
@@ -55,7 +61,7 @@ class DIN(_SeqBase):
         """seq [B,T,D], target [B,D] -> [B,D] attention-pooled."""
         t = target.unsqueeze(1).expand_as(seq)
         att_in = torch.cat([seq, t, seq - t, seq * t], dim=2)
-        scores = self.att(att_in).squeeze(2)          # [B, T]
+        scores = self.att(att_in).float().squeeze(2)  # [B, T]
         scores = scores.masked_fill(~mask, -1e9)
         w = torch.softmax(scores, dim=1)
         return (w.unsqueeze(2) * seq).sum(1)
@@ -84,9 +90,16 @@ class DIEN(_SeqBase):
         super().__init__(embedding_dim, item_dim, device, bf16, name="dien",
                          **kw)
         self.gru = nn.GRU(item_dim, gru_hidden, batch_first=True)
-        self.att = nn.Sequential(
-            nn.Linear(gru_hidden * 2, att_hidden), nn.Sigmoid(),
-            nn.Linear(att_hidden, 1))
+        if self.bf16:
+            from deeprec_amd.ops.fused_mlp import FusedLinear
+            self.att = nn.Sequential(
+                FusedLinear(gru_hidden * 2, att_hidden,
+                            activation="sigmoid"),
+                FusedLinear(att_hidden, 1, activation=None))
+        else:
+            self.att = nn.Sequential(
+                nn.Linear(gru_hidden * 2, att_hidden), nn.Sigmoid(),
+                nn.Linear(att_hidden, 1))
         self.augru_cell = nn.GRUCell(gru_hidden, gru_hidden)
         self.target_proj = nn.Linear(item_dim, gru_hidden)
         in_dim = (NUM_DENSE + self.num_sparse * embedding_dim
@@ -105,7 +118,7 @@ class DIEN(_SeqBase):
         tgt_h = self.target_proj(target)                 # [B,H]
         att_in = torch.cat(
             [h_seq, tgt_h.unsqueeze(1).expand_as(h_seq)], 2)
-        scores = self.att(att_in).squeeze(2)
+        scores = self.att(att_in).float().squeeze(2)
         scores = scores.masked_fill(mask == 0, -1e9)
         alpha = torch.softmax(scores, 1)                 # [B,T]
         # AUGRU: attention scales the update gate — implemented as

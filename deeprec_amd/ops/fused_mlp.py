@@ -9,22 +9,24 @@ layer instead of graph-level cast nodes.
 from __future__ import annotations
 
 import math
-from typing import List
+from typing import List, Optional
 
 import torch
 import torch.nn as nn
 
+_ACT_ID = {None: 0, "none": 0, "relu": 1, "sigmoid": 2}
+
 
 class _FusedLinear(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, relu):
+    def forward(ctx, x, weight, bias, act_id):
         from deeprec_amd.ops.build_ext import require_extension
         ext = require_extension()
         w16 = weight.detach().to(torch.bfloat16)
         x16 = x.to(torch.bfloat16).contiguous()
-        out = ext.linear_fwd(x16, w16, bias.detach().float(), relu)
+        out = ext.linear_fwd(x16, w16, bias.detach().float(), act_id)
         ctx.ext = ext
-        ctx.relu = relu
+        ctx.act_id = act_id
         ctx.x_dtype = x.dtype
         ctx.save_for_backward(x16, w16, out)
         return out
@@ -34,8 +36,8 @@ class _FusedLinear(torch.autograd.Function):
         x16, w16, out = ctx.saved_tensors
         ext = ctx.ext
         g = grad_out.to(torch.bfloat16).contiguous()
-        if ctx.relu:
-            g = ext.relu_bwd(g, out)
+        if ctx.act_id != 0:
+            g = ext.act_bwd(g, out, ctx.act_id)
         dx = ext.linear_dx(g, w16)
         # direct col-fragment dW: measured faster end-to-end than the
         # transpose-then-row-load variant (torch .t().contiguous() costs
@@ -45,14 +47,17 @@ class _FusedLinear(torch.autograd.Function):
 
 
 class FusedLinear(nn.Module):
-    """Linear(+ReLU) with fp32 master weight, bf16 MFMA compute on GPU."""
+    """Linear(+activation) with fp32 master weight, bf16 MFMA compute on
+    GPU. activation: None | 'relu' | 'sigmoid'. 3D inputs are flattened
+    over the leading dims."""
 
     def __init__(self, in_features: int, out_features: int,
-                 relu: bool = True):
+                 relu: bool = True, activation: Optional[str] = "unset"):
         super().__init__()
         self.in_features = in_features
         self.out_features = out_features
-        self.relu = relu
+        self.activation = ("relu" if relu else None) \
+            if activation == "unset" else activation
         self.weight = nn.Parameter(torch.empty(out_features, in_features))
         self.bias = nn.Parameter(torch.zeros(out_features))
         # torch Linear default init (kaiming uniform)
@@ -61,11 +66,19 @@ class FusedLinear(nn.Module):
         nn.init.uniform_(self.bias, -bound, bound)
 
     def forward(self, x):
-        if x.device.type == "cuda" and x.shape[0] % 16 == 0:
-            return _FusedLinear.apply(x, self.weight, self.bias, self.relu)
-        out = nn.functional.linear(x, self.weight.to(x.dtype),
-                                   self.bias.to(x.dtype))
-        return nn.functional.relu(out) if self.relu else out
+        lead = x.shape[:-1]
+        flat = x.reshape(-1, x.shape[-1])
+        if x.device.type == "cuda" and flat.shape[0] % 16 == 0:
+            out = _FusedLinear.apply(flat, self.weight, self.bias,
+                                     _ACT_ID[self.activation])
+        else:
+            out = nn.functional.linear(flat, self.weight.to(flat.dtype),
+                                       self.bias.to(flat.dtype))
+            if self.activation == "relu":
+                out = nn.functional.relu(out)
+            elif self.activation == "sigmoid":
+                out = torch.sigmoid(out)
+        return out.reshape(*lead, self.out_features)
 
 
 class _DotInteraction(torch.autograd.Function):

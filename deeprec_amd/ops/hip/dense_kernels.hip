@@ -102,7 +102,7 @@ __device__ __forceinline__ bf16x8 load_frag_col(const short* __restrict__ src,
 __global__ void k_linear_fwd(const short* __restrict__ A,
                              const short* __restrict__ W,
                              const float* __restrict__ bias, int M, int N,
-                             int K, int relu, short* __restrict__ C) {
+                             int K, int act, short* __restrict__ C) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int ntiles64 = (N + 63) / 64;
@@ -137,7 +137,8 @@ __global__ void k_linear_fwd(const short* __restrict__ A,
       int cm = m0 + (lane >> 4) * 4 + i;
       if (cm >= M) continue;
       float v = acc[t][i] + bv;
-      if (relu && v < 0.0f) v = 0.0f;
+      if (act == 1 && v < 0.0f) v = 0.0f;          // relu
+      else if (act == 2) v = 1.0f / (1.0f + __expf(-v));  // sigmoid
       C[(int64_t)cm * N + cn] = f2bf_u16(v);
     }
   }
@@ -316,10 +317,10 @@ __global__ void k_linear_dw_nt(const short* __restrict__ Gt,
   }
 }
 
-// relu backward mask: G = dY * (out > 0)
-__global__ void k_relu_bwd(const short* __restrict__ dY,
-                           const short* __restrict__ out, int64_t n,
-                           short* __restrict__ G) {
+// activation backward: G = dY * act_grad(out); act 1=relu, 2=sigmoid
+__global__ void k_act_bwd(const short* __restrict__ dY,
+                          const short* __restrict__ out, int64_t n, int act,
+                          short* __restrict__ G) {
   int64_t i = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 8;
   int64_t stride = gridDim.x * (int64_t)blockDim.x * 8;
   for (; i + 7 < n; i += stride) {
@@ -327,14 +328,26 @@ __global__ void k_relu_bwd(const short* __restrict__ dY,
     bf16x8 o = *reinterpret_cast<const bf16x8*>(out + i);
     bf16x8 r;
 #pragma unroll
-    for (int j = 0; j < 8; ++j)
-      r[j] = bf2f_u16(o[j]) > 0.0f ? g[j] : (short)0;
+    for (int j = 0; j < 8; ++j) {
+      if (act == 1) {
+        r[j] = bf2f_u16(o[j]) > 0.0f ? g[j] : (short)0;
+      } else {
+        float ov = bf2f_u16(o[j]);
+        r[j] = f2bf_u16(bf2f_u16(g[j]) * ov * (1.0f - ov));
+      }
+    }
     *reinterpret_cast<bf16x8*>(G + i) = r;
   }
   // tail
   if (blockIdx.x == 0 && threadIdx.x == 0) {
-    for (int64_t j = n & ~7LL; j < n; ++j)
-      G[j] = bf2f_u16(out[j]) > 0.0f ? dY[j] : (short)0;
+    for (int64_t j = n & ~7LL; j < n; ++j) {
+      if (act == 1) {
+        G[j] = bf2f_u16(out[j]) > 0.0f ? dY[j] : (short)0;
+      } else {
+        float ov = bf2f_u16(out[j]);
+        G[j] = f2bf_u16(bf2f_u16(dY[j]) * ov * (1.0f - ov));
+      }
+    }
   }
 }
 
@@ -437,7 +450,7 @@ static short* bf_ptr_mut(torch::Tensor& t) {
 }
 
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w_bf16,
-                         torch::Tensor bias, bool relu) {
+                         torch::Tensor bias, int64_t act) {
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
   int M = x.size(0), K = x.size(1), N = w_bf16.size(0);
   auto out = torch::empty({M, N}, x.options());
@@ -446,7 +459,7 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w_bf16,
   k_linear_fwd<<<blocks, 256, 0, dense_stream()>>>(
       bf_ptr(x), bf_ptr(w_bf16),
       bias.defined() && bias.numel() ? bias.data_ptr<float>() : nullptr, M,
-      N, K, relu ? 1 : 0, bf_ptr_mut(out));
+      N, K, (int)act, bf_ptr_mut(out));
   return out;
 }
 
@@ -504,12 +517,12 @@ std::tuple<torch::Tensor, torch::Tensor> linear_dw_nt(torch::Tensor gt,
   return {dw, db};
 }
 
-torch::Tensor relu_bwd(torch::Tensor dy, torch::Tensor out) {
+torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor out, int64_t act) {
   auto g = torch::empty_like(dy);
   int64_t n = dy.numel();
   int blocks = (int)std::min<int64_t>((n / 8 + 255) / 256, 4096);
-  k_relu_bwd<<<std::max(blocks, 1), 256, 0, dense_stream()>>>(
-      bf_ptr(dy), bf_ptr(out), n, bf_ptr_mut(g));
+  k_act_bwd<<<std::max(blocks, 1), 256, 0, dense_stream()>>>(
+      bf_ptr(dy), bf_ptr(out), n, (int)act, bf_ptr_mut(g));
   return g;
 }
 
@@ -547,5 +560,5 @@ void register_dense(py::module_& mod) {
   mod.def("linear_dx", &linear_dx);
   mod.def("linear_dw", &linear_dw);
   mod.def("linear_dw_nt", &linear_dw_nt);
-  mod.def("relu_bwd", &relu_bwd);
+  mod.def("act_bwd", &act_bwd);
 }

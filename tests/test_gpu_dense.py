@@ -53,15 +53,30 @@ def test_linear_dw(m, n, k):
     torch.testing.assert_close(db, ref_db, rtol=2e-2, atol=5e-2)
 
 
-def test_relu_bwd():
+def test_act_bwd():
     ext = _ext()
     dy = torch.randn(1000, 33, device=DEV).to(torch.bfloat16)
-    out = torch.randn(1000, 33, device=DEV).to(torch.bfloat16)
-    g = ext.relu_bwd(dy.reshape(-1).contiguous(),
-                     out.reshape(-1).contiguous())
+    out = torch.rand(1000, 33, device=DEV).to(torch.bfloat16) * 2 - 1
+    g = ext.act_bwd(dy.reshape(-1).contiguous(),
+                    out.reshape(-1).contiguous(), 1)
     ref = torch.where(out.reshape(-1).float() > 0,
                       dy.reshape(-1).float(), torch.zeros(1, device=DEV))
     torch.testing.assert_close(g.float(), ref)
+    g2 = ext.act_bwd(dy.reshape(-1).contiguous(),
+                     out.reshape(-1).contiguous(), 2)
+    ov = out.reshape(-1).float()
+    ref2 = dy.reshape(-1).float() * ov * (1 - ov)
+    torch.testing.assert_close(g2.float(), ref2, rtol=2e-2, atol=2e-2)
+
+
+def test_sigmoid_fwd():
+    ext = _ext()
+    x = torch.randn(64, 32, device=DEV).to(torch.bfloat16)
+    w = torch.randn(16, 32, device=DEV).to(torch.bfloat16)
+    b = torch.randn(16, device=DEV)
+    out = ext.linear_fwd(x, w, b, 2)
+    ref = torch.sigmoid(x.float() @ w.float().t() + b)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
 
 
 def test_fused_mlp_training_step():
