@@ -32,6 +32,22 @@ def make_mlp(sizes: Sequence[int], in_dim: int, device,
     return nn.Sequential(*layers)
 
 
+class Dice(nn.Module):
+    """Dice activation (reference: grappler dice_fusion.cc + DIN paper):
+    x * p + alpha * x * (1 - p), p = sigmoid(batchnorm(x))."""
+
+    def __init__(self, dim: int, eps: float = 1e-8):
+        super().__init__()
+        self.bn = nn.BatchNorm1d(dim, eps=eps, affine=False)
+        self.alpha = nn.Parameter(torch.zeros(dim))
+
+    def forward(self, x):
+        shape = x.shape
+        p = torch.sigmoid(self.bn(x.reshape(-1, shape[-1]).float()))
+        p = p.reshape(shape).to(x.dtype)
+        return x * p + self.alpha.to(x.dtype) * x * (1 - p)
+
+
 class RecModelBase(nn.Module):
     """Criteo-schema recommendation model base: one EmbeddingCollection for
     the 26 categorical features + helpers shared by the zoo."""
