@@ -65,7 +65,9 @@ def main():
                          learning_rate=0.001)
 
     if distributed:
-        from deeprec_amd.parallel import DenseGradAllreducer
+        from deeprec_amd.parallel import (DenseGradAllreducer,
+                                          broadcast_parameters)
+        broadcast_parameters(model.parameters())
         reducer = DenseGradAllreducer(model.parameters())
     else:
         reducer = None
