@@ -188,18 +188,10 @@ class EmbeddingCollection:
         if self.device.type != "cuda":
             order = torch.argsort(inverse.long()).to(torch.int32)
             return order, bounds, None, None
-        # sort-free CSR build (per-key cursors)
+        # sort-free CSR build (per-key cursors); the strided backward
+        # needs no host-built chunk arrays
         order = self.storage.ext.csr_order(inverse, bounds, m)
-        nch = (counts + (self._CHUNK - 1)) // self._CHUNK
-        chunk_u = torch.repeat_interleave(
-            torch.arange(m, device=dev, dtype=torch.int64), nch)
-        chunk_base = nch.cumsum(0) - nch
-        pos_in_u = (torch.arange(chunk_u.numel(), device=dev,
-                                 dtype=torch.int64)
-                    - chunk_base[chunk_u])
-        chunk_k0 = (bounds[:-1].to(torch.int64)[chunk_u]
-                    + pos_in_u * self._CHUNK).to(torch.int32)
-        return order, bounds, chunk_u.to(torch.int32), chunk_k0
+        return order, bounds, None, None
 
     def lookup_matrix(self, ids: torch.Tensor, out_dtype=None,
                       train: bool = True) -> torch.Tensor:
@@ -283,11 +275,10 @@ class EmbeddingCollection:
     def _backward(self, grad_out, order, bounds, chunk_u, chunk_k0,
                   row_ids_cat, weights_cat, row_coeff, m, batch):
         if self.device.type == "cuda":
-            return self.storage.ext.group_pooled_bwd_chunked(
+            return self.storage.ext.group_pooled_bwd_strided(
                 grad_out.contiguous(), order, bounds, row_ids_cat,
                 weights_cat if weights_cat is not None else torch.Tensor(),
-                row_coeff, chunk_u, chunk_k0, self._CHUNK, m, batch,
-                self.n_tables, self.dim)
+                row_coeff, m, batch, self.n_tables, self.dim)
         # CPU reference path
         g = grad_out.float().reshape(batch, self.n_tables, self.dim)
         grad_unique = torch.zeros(m, self.dim)

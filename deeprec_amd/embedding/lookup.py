@@ -79,14 +79,6 @@ def _chunked_ev_backward(ev, grad_out, inverse, offsets, row_ids, counts,
     bounds = torch.zeros(m + 1, dtype=torch.int32, device=dev)
     bounds[1:] = c32.cumsum(0)
     order = ext.csr_order(inverse, bounds, m)
-    nch = (counts.long() + (_CHUNK - 1)) // _CHUNK
-    chunk_u = torch.repeat_interleave(
-        torch.arange(m, device=dev, dtype=torch.int64), nch)
-    chunk_base = nch.cumsum(0) - nch
-    pos_in_u = (torch.arange(chunk_u.numel(), device=dev, dtype=torch.int64)
-                - chunk_base[chunk_u])
-    chunk_k0 = (bounds[:-1].long()[chunk_u] + pos_in_u * _CHUNK).to(
-        torch.int32)
     # per-row combiner coefficients
     lengths = (offsets[1:] - offsets[:-1]).float()
     if combiner == "sum":
@@ -102,11 +94,10 @@ def _chunked_ev_backward(ev, grad_out, inverse, offsets, row_ids, counts,
         row_coeff = torch.where(lengths > 0,
                                 1.0 / denom.clamp(min=1e-12),
                                 torch.zeros_like(lengths))
-    return ext.group_pooled_bwd_chunked(
+    return ext.group_pooled_bwd_strided(
         grad_out.contiguous(), order, bounds, row_ids.to(torch.int32),
         weights.float() if weights is not None else torch.Tensor(),
-        row_coeff, chunk_u.to(torch.int32), chunk_k0, _CHUNK, m,
-        offsets.numel() - 1, 1, ev.dim)
+        row_coeff, m, offsets.numel() - 1, 1, ev.dim)
 
 
 def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,

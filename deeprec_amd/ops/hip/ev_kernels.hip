@@ -628,6 +628,53 @@ __global__ void k_group_pooled_bwd_chunked(
   }
 }
 
+// Strided variant: no host-built chunk arrays. Thread (u, d, s) with
+// s in [0, SPLITS) covers occurrence windows [k0, k0+CHUNK) striding by
+// SPLITS*CHUNK — zipf-hot keys are split across SPLITS concurrent
+// accumulators (bounded atomics) while the host does zero glue work
+// (the chunk-array build was 6 kernels + a repeat_interleave sync/step).
+template <typename GradT, int SPLITS, int CHUNK>
+__global__ void k_group_pooled_bwd_strided(
+    const GradT* __restrict__ grad_out, const int32_t* __restrict__ order,
+    const int32_t* __restrict__ bounds, const int32_t* __restrict__ row_ids,
+    const float* __restrict__ weights, const float* __restrict__ row_coeff,
+    int m, int batch, int n_tables, int dim,
+    float* __restrict__ grad_unique) {
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)m * dim * SPLITS;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; t < total; t += stride) {
+    int d = (int)(t % dim);
+    int64_t ud = t / dim;
+    int u = (int)(ud % m);
+    int s = (int)(ud / m);
+    int beg = bounds[u], end = bounds[u + 1];
+    int cnt = end - beg;
+    if (s * CHUNK >= cnt) continue;   // this split has no window
+    float acc = 0.0f;
+    for (int k0 = beg + s * CHUNK; k0 < end; k0 += SPLITS * CHUNK) {
+      int k1 = min(k0 + CHUNK, end);
+      for (int k = k0; k < k1; ++k) {
+        int j = order[k];
+        int rid = row_ids[j];
+        int table = rid / batch;
+        int b = rid % batch;
+        float g;
+        int64_t gidx = ((int64_t)b * n_tables + table) * dim + d;
+        if constexpr (std::is_same_v<GradT, __hip_bfloat16>)
+          g = bf2f(grad_out[gidx]);
+        else
+          g = grad_out[gidx];
+        float w = weights ? weights[j] : 1.0f;
+        acc += w * row_coeff[rid] * g;
+      }
+    }
+    int64_t o = (int64_t)u * dim + d;
+    if (cnt <= CHUNK) grad_unique[o] = acc;   // single window: plain store
+    else atomicAdd(&grad_unique[o], acc);
+  }
+}
+
 // ---------------------------------------------------------------------
 // fused sparse optimizer applies (thread = (unique key i, dim d))
 // ---------------------------------------------------------------------
@@ -1129,6 +1176,39 @@ torch::Tensor group_pooled_bwd(torch::Tensor grad_out, torch::Tensor order,
   return grad_unique;
 }
 
+torch::Tensor group_pooled_bwd_strided(
+    torch::Tensor grad_out, torch::Tensor order, torch::Tensor bounds,
+    torch::Tensor row_ids, torch::Tensor weights, torch::Tensor row_coeff,
+    int64_t m, int64_t batch, int64_t n_tables, int64_t dim) {
+  constexpr int SPLITS = 8, CHUNK = 128;
+  auto grad_unique = torch::zeros(
+      {m, dim}, grad_out.options().dtype(torch::kFloat32));
+  int64_t total = m * dim * SPLITS;
+  if (total == 0) return grad_unique;
+  auto stream = current_stream();
+  const float* wptr =
+      weights.defined() && weights.numel() ? weights.data_ptr<float>()
+                                           : nullptr;
+  if (grad_out.scalar_type() == torch::kBFloat16) {
+    k_group_pooled_bwd_strided<__hip_bfloat16, SPLITS, CHUNK>
+        <<<n_blocks(total), kBlock, 0, stream>>>(
+            reinterpret_cast<const __hip_bfloat16*>(
+                grad_out.data_ptr<at::BFloat16>()),
+            order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
+            row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
+            (int)m, (int)batch, (int)n_tables, (int)dim,
+            grad_unique.data_ptr<float>());
+  } else {
+    k_group_pooled_bwd_strided<float, SPLITS, CHUNK>
+        <<<n_blocks(total), kBlock, 0, stream>>>(
+            grad_out.data_ptr<float>(), order.data_ptr<int32_t>(),
+            bounds.data_ptr<int32_t>(), row_ids.data_ptr<int32_t>(), wptr,
+            row_coeff.data_ptr<float>(), (int)m, (int)batch, (int)n_tables,
+            (int)dim, grad_unique.data_ptr<float>());
+  }
+  return grad_unique;
+}
+
 torch::Tensor group_pooled_bwd_chunked(
     torch::Tensor grad_out, torch::Tensor order, torch::Tensor bounds,
     torch::Tensor row_ids, torch::Tensor weights, torch::Tensor row_coeff,
@@ -1282,6 +1362,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("group_pooled_fwd_direct", &group_pooled_fwd_direct);
   mod.def("group_pooled_bwd", &group_pooled_bwd);
   mod.def("group_pooled_bwd_chunked", &group_pooled_bwd_chunked);
+  mod.def("group_pooled_bwd_strided", &group_pooled_bwd_strided);
   mod.def("apply_sgd", &apply_sgd);
   mod.def("apply_adagrad", &apply_adagrad);
   mod.def("apply_adagrad_decay", &apply_adagrad_decay);
