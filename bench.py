@@ -28,6 +28,10 @@ def main():
     p.add_argument("--device", type=str, default=None)
     p.add_argument("--optimizer", type=str, default="adamasync")
     p.add_argument("--no-bf16", action="store_true")
+    p.add_argument("--hip-graph", action="store_true", default=True,
+                   help="capture the steady-state step in a hipGraph "
+                        "(single-GPU only; replay eliminates launch gaps)")
+    p.add_argument("--no-hip-graph", dest="hip_graph", action="store_false")
     args = p.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -56,13 +60,24 @@ def main():
 
     torch.manual_seed(42 + rank)
     bf16 = (not args.no_bf16) and device.type == "cuda"
-    model = DLRM(device=device, bf16=bf16, sharded=distributed)
+    use_graph = (args.hip_graph and not distributed
+                 and device.type == "cuda"
+                 and args.optimizer == "adamasync")
+    ev_option = None
+    if use_graph:
+        from deeprec_amd.embedding.options import EmbeddingVariableOption
+        # no growth can happen inside a captured step: pre-size for the
+        # full synthetic id space
+        ev_option = EmbeddingVariableOption(init_capacity=1 << 23)
+    model = DLRM(device=device, bf16=bf16, sharded=distributed,
+                 ev_option=ev_option)
     ds = CriteoSyntheticDataset(batch_size=args.batch, device=device,
                                 seed=1234, rank=rank,
                                 matrix_format=True)
+    opt_kw = {"graph_safe": True} if use_graph else {}
     opt = make_optimizer(args.optimizer, params=model.parameters(),
                          embedding_variables=model.embedding_variables(),
-                         learning_rate=0.001)
+                         learning_rate=0.001, **opt_kw)
 
     if distributed:
         from deeprec_amd.parallel import (DenseGradAllreducer,
@@ -78,7 +93,7 @@ def main():
     n_total = args.warmup + args.steps
     batches = [ds.next_batch() for _ in range(min(n_total, 20))]
 
-    def one_step(i):
+    def eager_step(i):
         dense, sparse, labels = batches[i % len(batches)]
         logits = model(dense, sparse)
         loss = model.loss_fn(logits, labels)
@@ -89,8 +104,39 @@ def main():
         opt.step()
         return loss
 
-    for i in range(args.warmup):
-        one_step(i)
+    one_step = eager_step
+    if use_graph:
+        st = model.collection.storage
+        st.enable_graph_mode(expected_entries=1 << 22,
+                             expected_slots=1 << 22)
+        # optimizer slabs must exist before capture
+        st.get_slab("adam_m", model.collection.dim, 0.0)
+        st.get_slab("adam_v", model.collection.dim, 0.0)
+        model.collection.graph_mode = True
+        for i in range(args.warmup):
+            eager_step(i)
+        torch.cuda.synchronize()
+        # static input buffers + whole-step capture
+        sdense, sids, slabels = (t.clone() for t in batches[0])
+        graph = torch.cuda.CUDAGraph()
+        opt.zero_grad()
+        with torch.cuda.graph(graph):
+            logits = model(sdense, sids)
+            loss = model.loss_fn(logits, slabels)
+            loss.backward()
+            opt.step()
+        torch.cuda.synchronize()
+
+        def one_step(i):
+            dense, sparse, labels = batches[i % len(batches)]
+            sdense.copy_(dense, non_blocking=True)
+            sids.copy_(sparse, non_blocking=True)
+            slabels.copy_(labels, non_blocking=True)
+            graph.replay()
+            return loss
+    else:
+        for i in range(args.warmup):
+            one_step(i)
 
     if distributed:
         import torch.distributed as dist
@@ -106,6 +152,8 @@ def main():
         import torch.distributed as dist
         dist.barrier()
     elapsed = time.perf_counter() - t0
+    if use_graph:
+        model.collection.storage._check_error()  # no silent slab overflow
 
     if distributed:
         import torch.distributed as dist
@@ -137,6 +185,7 @@ def main():
                        "seq_len": 1,
                        "parallelism": f"dp{n_gpus}+ep{n_gpus}"
                        if distributed else "single",
+                       "hip_graph": use_graph,
                        "optimizer": args.optimizer,
                        "embedding_dim": 16, "num_tables": 26},
         }))

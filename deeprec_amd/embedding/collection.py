@@ -65,6 +65,7 @@ class EmbeddingCollection:
         self._anchor = torch.zeros((), device=self.device,
                                    requires_grad=trainable)
         self._matrix_cache = {}
+        self.graph_mode = False
         self._pending_grads: List = []
         self._recorded_ids: List[torch.Tensor] = []
         self._record_sparse_ids = False
@@ -164,6 +165,8 @@ class EmbeddingCollection:
     def _dedup_and_probe(self, values_cat, train):
         """unique + single hash probe. GPU training uses the fused hash
         dedup (no sorts); other paths use torch.unique + the probe."""
+        if self.graph_mode and train:
+            return self.storage.dedup_lookup_capture(values_cat)
         if (train and hasattr(self.storage, "dedup_lookup")
                 and self.storage.prefers_dedup()):
             return self.storage.dedup_lookup(values_cat, get_global_step())

@@ -220,6 +220,60 @@ __global__ void k_dedup_pass_a(
   }
 }
 
+// Graph-capture pass A: epoch and step advance on-device (bumped by
+// k_bump_epoch inside the captured step) so replays keep deduplicating
+// correctly with zero host involvement.
+__global__ void k_bump_epoch(int32_t* __restrict__ epoch_dev,
+                             int64_t* __restrict__ step_dev) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    epoch_dev[0] += 1;
+    step_dev[0] += 1;
+  }
+}
+
+__global__ void k_dedup_pass_a_dev(
+    const int64_t* __restrict__ keys, int nnz, int64_t* __restrict__ ht_keys,
+    int32_t* __restrict__ ht_freq, int64_t* __restrict__ ht_version,
+    int32_t* __restrict__ ht_epoch, int32_t* __restrict__ ht_compact,
+    int64_t cap_mask, const int32_t* __restrict__ epoch_dev,
+    const int64_t* __restrict__ step_dev,
+    int32_t* __restrict__ entry_counter, int32_t* __restrict__ m_counter,
+    int64_t* __restrict__ uniq_keys, int64_t* __restrict__ compact_entry,
+    int32_t* __restrict__ error_flag) {
+  const int epoch = *epoch_dev;
+  const int64_t step = *step_dev;
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; j < nnz; j += stride) {
+    const int64_t key = keys[j];
+    uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
+    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+      int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
+      int64_t cur = ht_keys[idx];
+      if (cur != key) {
+        if (cur != EMPTY_KEY) continue;
+        int64_t prev = (int64_t)atomicCAS(
+            (unsigned long long*)&ht_keys[idx],
+            (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+        if (prev != EMPTY_KEY && prev != key) continue;
+        if (prev == EMPTY_KEY) atomicAdd(entry_counter, 1);
+      }
+      atomicAdd(&ht_freq[idx], 1);
+      ht_version[idx] = step;
+      int old = atomicExch(&ht_epoch[idx], epoch);
+      if (old != epoch) {
+        int c = atomicAdd(m_counter, 1);
+        ht_compact[idx] = c;
+        uniq_keys[c] = key;
+        compact_entry[c] = idx;
+      }
+      goto next_j2;
+    }
+    atomicExch(error_flag, 2);
+  next_j2:;
+  }
+}
+
 // Pass B (per unique key): admission + default-value init, slots out.
 __global__ void k_dedup_pass_b(
     const int64_t* __restrict__ compact_entry,
@@ -232,6 +286,53 @@ __global__ void k_dedup_pass_b(
     int32_t* __restrict__ error_flag) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= m) return;
+  int64_t idx = compact_entry[c];
+  int32_t slot = ht_slot[idx];
+  if (slot < 0 && ht_freq[idx] >= filter_freq) {
+    slot = atomicAdd(slot_counter, 1);
+    if (slot >= max_slots) {
+      atomicExch(error_flag, 1);
+      out_slots[c] = -1;
+      return;
+    }
+    ht_slot[idx] = slot;
+    if (slot < init_limit) {
+      const int64_t key = uniq_keys[c];
+      int64_t dvrow;
+      if (key_bits > 0) {
+        int64_t mask = ((int64_t)1 << key_bits) - 1;
+        dvrow = (key >> key_bits) * default_value_dim +
+                (int64_t)((uint64_t)(key & mask) %
+                          (uint64_t)default_value_dim);
+      } else {
+        dvrow = (int64_t)((uint64_t)key % (uint64_t)default_value_dim);
+      }
+      const float* src = default_values + dvrow * dim;
+      float* dst = values + (int64_t)slot * dim;
+      for (int d = 0; d < dim; ++d) dst[d] = src[d];
+    }
+  }
+  out_slots[c] = slot;
+}
+
+// Padded pass B for hipGraph capture: grid covers n_cap (= nnz upper
+// bound); the true unique count comes from the device counter so no host
+// sync happens inside the captured step. Tail entries get slot -1.
+__global__ void k_dedup_pass_b_padded(
+    const int64_t* __restrict__ compact_entry,
+    const int64_t* __restrict__ uniq_keys, int n_cap,
+    const int32_t* __restrict__ m_dev, int32_t* __restrict__ ht_slot,
+    const int32_t* __restrict__ ht_freq, int32_t* __restrict__ slot_counter,
+    int max_slots, float* __restrict__ values,
+    const float* __restrict__ default_values, int dim,
+    int default_value_dim, int key_bits, int init_limit, int filter_freq,
+    int32_t* __restrict__ out_slots, int32_t* __restrict__ error_flag) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= n_cap) return;
+  if (c >= *m_dev) {
+    out_slots[c] = -1;
+    return;
+  }
   int64_t idx = compact_entry[c];
   int32_t slot = ht_slot[idx];
   if (slot < 0 && ht_freq[idx] >= filter_freq) {
@@ -762,6 +863,33 @@ __global__ void k_apply_adam(float* __restrict__ w, float* __restrict__ mom,
   APPLY_EPILOG
 }
 
+__global__ void k_apply_adam_dev(float* __restrict__ w,
+                                 float* __restrict__ mom,
+                                 float* __restrict__ vel,
+                                 const int32_t* __restrict__ slots,
+                                 const float* __restrict__ grad, int m,
+                                 int dim, float lr, float beta1, float beta2,
+                                 float epsilon,
+                                 const float* __restrict__ powers) {
+  const float lr_t =
+      lr * sqrtf(1.0f - powers[1]) / (1.0f - powers[0]);
+  APPLY_PROLOG
+  float mn = beta1 * mom[o] + (1.0f - beta1) * g;
+  float vn = beta2 * vel[o] + (1.0f - beta2) * g * g;
+  mom[o] = mn;
+  vel[o] = vn;
+  w[o] -= lr_t * mn / (sqrtf(vn) + epsilon);
+  APPLY_EPILOG
+}
+
+__global__ void k_update_powers(float* __restrict__ powers, float beta1,
+                                float beta2) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    powers[0] *= beta1;
+    powers[1] *= beta2;
+  }
+}
+
 __global__ void k_apply_adamw(float* __restrict__ w, float* __restrict__ mom,
                               float* __restrict__ vel,
                               const int32_t* __restrict__ slots,
@@ -894,6 +1022,53 @@ torch::Tensor ht_dedup_b(torch::Tensor compact_entry, torch::Tensor uniq_keys,
       values.size(1), (int)dvd_per_table, (int)key_bits, (int)init_limit,
       (int)filter_freq, slots.data_ptr<int32_t>(),
       error_flag.data_ptr<int32_t>());
+  return slots;
+}
+
+void bump_epoch(torch::Tensor epoch_dev, torch::Tensor step_dev) {
+  k_bump_epoch<<<1, 64, 0, current_stream()>>>(
+      epoch_dev.data_ptr<int32_t>(), step_dev.data_ptr<int64_t>());
+}
+
+torch::Tensor ht_dedup_a_dev(
+    torch::Tensor keys, torch::Tensor ht_keys, torch::Tensor ht_freq,
+    torch::Tensor ht_version, torch::Tensor ht_epoch,
+    torch::Tensor ht_compact, torch::Tensor epoch_dev,
+    torch::Tensor step_dev, torch::Tensor entry_counter,
+    torch::Tensor m_counter, torch::Tensor uniq_keys,
+    torch::Tensor compact_entry, torch::Tensor error_flag) {
+  int64_t nnz = keys.numel();
+  if (nnz == 0) return m_counter;
+  auto stream = current_stream();
+  k_dedup_pass_a_dev<<<n_blocks(nnz), kBlock, 0, stream>>>(
+      keys.data_ptr<int64_t>(), (int)nnz, ht_keys.data_ptr<int64_t>(),
+      ht_freq.data_ptr<int32_t>(), ht_version.data_ptr<int64_t>(),
+      ht_epoch.data_ptr<int32_t>(), ht_compact.data_ptr<int32_t>(),
+      ht_keys.numel() - 1, epoch_dev.data_ptr<int32_t>(),
+      step_dev.data_ptr<int64_t>(), entry_counter.data_ptr<int32_t>(),
+      m_counter.data_ptr<int32_t>(), uniq_keys.data_ptr<int64_t>(),
+      compact_entry.data_ptr<int64_t>(), error_flag.data_ptr<int32_t>());
+  return m_counter;
+}
+
+torch::Tensor ht_dedup_b_padded(
+    torch::Tensor compact_entry, torch::Tensor uniq_keys,
+    torch::Tensor m_dev, torch::Tensor ht_slot, torch::Tensor ht_freq,
+    torch::Tensor slot_counter, int64_t max_slots, torch::Tensor values,
+    torch::Tensor default_values, int64_t dvd_per_table, int64_t key_bits,
+    int64_t init_limit, int64_t filter_freq, torch::Tensor error_flag) {
+  int n_cap = uniq_keys.numel();
+  auto slots = torch::empty({n_cap}, ht_slot.options());
+  if (n_cap == 0) return slots;
+  auto stream = current_stream();
+  k_dedup_pass_b_padded<<<n_blocks(n_cap), kBlock, 0, stream>>>(
+      compact_entry.data_ptr<int64_t>(), uniq_keys.data_ptr<int64_t>(),
+      n_cap, m_dev.data_ptr<int32_t>(), ht_slot.data_ptr<int32_t>(),
+      ht_freq.data_ptr<int32_t>(), slot_counter.data_ptr<int32_t>(),
+      (int)max_slots, values.data_ptr<float>(),
+      default_values.data_ptr<float>(), values.size(1), (int)dvd_per_table,
+      (int)key_bits, (int)init_limit, (int)filter_freq,
+      slots.data_ptr<int32_t>(), error_flag.data_ptr<int32_t>());
   return slots;
 }
 
@@ -1302,6 +1477,27 @@ void apply_adam(torch::Tensor w, torch::Tensor mom, torch::Tensor vel,
       (float)beta1, (float)beta2, (float)epsilon);
 }
 
+void apply_adam_dev(torch::Tensor w, torch::Tensor mom, torch::Tensor vel,
+                    torch::Tensor slots, torch::Tensor grad, double lr,
+                    double beta1, double beta2, double epsilon,
+                    torch::Tensor powers) {
+  int m = slots.numel();
+  if (m == 0) return;
+  int dim = w.size(1);
+  auto stream = current_stream();
+  k_apply_adam_dev<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
+      w.data_ptr<float>(), mom.data_ptr<float>(), vel.data_ptr<float>(),
+      slots.data_ptr<int32_t>(), grad.data_ptr<float>(), m, dim, (float)lr,
+      (float)beta1, (float)beta2, (float)epsilon,
+      powers.data_ptr<float>());
+}
+
+void update_powers(torch::Tensor powers, double beta1, double beta2) {
+  auto stream = current_stream();
+  k_update_powers<<<1, 64, 0, stream>>>(powers.data_ptr<float>(),
+                                        (float)beta1, (float)beta2);
+}
+
 void apply_adamw(torch::Tensor w, torch::Tensor mom, torch::Tensor vel,
                  torch::Tensor slots, torch::Tensor grad, double lr_t,
                  double lr, double beta1, double beta2, double epsilon,
@@ -1350,6 +1546,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ht_lookup_insert", &ht_lookup_insert);
   mod.def("ht_dedup_a", &ht_dedup_a);
   mod.def("ht_dedup_b", &ht_dedup_b);
+  mod.def("ht_dedup_b_padded", &ht_dedup_b_padded);
+  mod.def("ht_dedup_a_dev", &ht_dedup_a_dev);
+  mod.def("bump_epoch", &bump_epoch);
   mod.def("ht_dedup_c", &ht_dedup_c);
   mod.def("csr_order", &csr_order);
   mod.def("ht_insert_bulk", &ht_insert_bulk);
@@ -1368,6 +1567,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("apply_adagrad_decay", &apply_adagrad_decay);
   mod.def("apply_adam", &apply_adam);
   mod.def("apply_adamw", &apply_adamw);
+  mod.def("apply_adam_dev", &apply_adam_dev);
+  mod.def("update_powers", &update_powers);
   mod.def("apply_rmsprop", &apply_rmsprop);
   mod.def("apply_ftrl", &apply_ftrl);
   mod.attr("EMPTY_KEY") = EMPTY_KEY;

@@ -200,6 +200,45 @@ class HbmStorage:
         slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)
         return slots
 
+    def enable_graph_mode(self, expected_entries: int, expected_slots: int):
+        """hipGraph capture prep: pre-size the table/slabs (no growth can
+        happen inside a captured step) and move the dedup epoch/step
+        counters to device scalars bumped inside the capture."""
+        if self.capacity < _pow2(int(expected_entries / _LOAD_FACTOR) + 1):
+            self._rehash(_pow2(int(expected_entries / _LOAD_FACTOR) + 1))
+        if self.max_slots < expected_slots:
+            self._grow_slots(expected_slots)
+        self.graph_mode = True
+        self._epoch_dev = torch.tensor([self._epoch], dtype=torch.int32,
+                                       device=self.device)
+        from deeprec_amd.embedding.variable import get_global_step
+        self._step_dev = torch.tensor([get_global_step()],
+                                      dtype=torch.int64, device=self.device)
+
+    def dedup_lookup_capture(self, values_cat: torch.Tensor):
+        """Sync-free, fixed-shape dedup+probe for hipGraph capture: all
+        outputs are nnz-padded; the true unique count lives in a device
+        counter consumed by the padded pass B (tail slots = -1)."""
+        nnz = values_cat.numel()
+        self.ext.bump_epoch(self._epoch_dev, self._step_dev)
+        uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
+        self.ext.ht_dedup_a_dev(values_cat, self.ht_keys, self.ht_freq,
+                                self.ht_version, self.ht_epoch,
+                                self.ht_compact, self._epoch_dev,
+                                self._step_dev, self.entry_counter,
+                                m_counter, uniq_buf, centry_buf,
+                                self.error_flag)
+        slots = self.ext.ht_dedup_b_padded(
+            centry_buf, uniq_buf, m_counter, self.ht_slot, self.ht_freq,
+            self.slot_counter, self.max_slots, self.values,
+            self.default_values, self.dvd_per_table, self.key_bits,
+            self._init_limit(), self.filter_freq, self.error_flag)
+        inverse, counts = self.ext.ht_dedup_c(values_cat, self.ht_keys,
+                                              self.ht_compact, nnz)
+        return uniq_buf, inverse, counts, slots
+
     def prefers_dedup(self) -> bool:
         return self._uniq_ratio is None or self._uniq_ratio < 0.5
 
@@ -438,7 +477,13 @@ def sparse_apply(name: str, storage: HbmStorage, slots, grad, hyper: dict):
     elif name == "adam_async":
         import math as _m
         b1, b2 = hyper["beta1"], hyper["beta2"]
-        if hyper.get("sparse_rmsprop"):
+        if hyper.get("powers_dev") is not None:
+            mom = storage.get_slab("adam_m", d, 0.0)
+            vel = storage.get_slab("adam_v", d, 0.0)
+            ext.apply_adam_dev(w, mom, vel, slots, grad, hyper["lr"],
+                               b1, b2, hyper["epsilon"],
+                               hyper["powers_dev"])
+        elif hyper.get("sparse_rmsprop"):
             vel = storage.get_slab("adam_v", d, 0.0)
             ext.apply_rmsprop(w, vel, slots, grad, hyper["lr"], b2,
                               hyper["epsilon"])

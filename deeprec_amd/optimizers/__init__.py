@@ -179,25 +179,47 @@ class AdamAsyncOptimizer(AdamOptimizer):
 
     def __init__(self, params=None, embedding_variables=None,
                  learning_rate=0.001, beta1=0.9, beta2=0.999, epsilon=1e-8,
-                 apply_sparse_rmsprop=False):
+                 apply_sparse_rmsprop=False, graph_safe=False):
         self.apply_sparse_rmsprop = apply_sparse_rmsprop
+        self.graph_safe = graph_safe  # hipGraph capture: powers on device
         self._beta_powers = {}
         super().__init__(params, embedding_variables, learning_rate,
                          beta1, beta2, epsilon)
 
     def _sparse_hyper(self, ev):
+        if self.graph_safe:
+            powers = self._beta_powers.get(ev.name)
+            if not torch.is_tensor(powers):
+                powers = torch.tensor([self.beta1, self.beta2],
+                                      device=ev.device)
+                self._beta_powers[ev.name] = powers
+            return dict(lr=self.lr, powers_dev=powers, beta1=self.beta1,
+                        beta2=self.beta2, epsilon=self.epsilon,
+                        sparse_rmsprop=self.apply_sparse_rmsprop)
         b1p, b2p = self._beta_powers.setdefault(
             ev.name, [self.beta1, self.beta2])
         return dict(lr=self.lr, beta1_power=b1p, beta2_power=b2p,
                     beta1=self.beta1, beta2=self.beta2, epsilon=self.epsilon,
                     sparse_rmsprop=self.apply_sparse_rmsprop)
 
+    def _make_dense(self, params):
+        return torch.optim.Adam(params, lr=self.lr,
+                                betas=(self.beta1, self.beta2),
+                                eps=self.epsilon,
+                                capturable=self.graph_safe)
+
     def _post_step(self):
         for ev in self.evs:
-            p = self._beta_powers.setdefault(ev.name,
-                                             [self.beta1, self.beta2])
-            p[0] *= self.beta1
-            p[1] *= self.beta2
+            p = self._beta_powers.setdefault(
+                ev.name, torch.tensor([self.beta1, self.beta2],
+                                      device=ev.device)
+                if self.graph_safe else [self.beta1, self.beta2])
+            if torch.is_tensor(p):
+                from deeprec_amd.ops.build_ext import require_extension
+                require_extension().update_powers(p, self.beta1, self.beta2)
+            else:
+                p[0] *= self.beta1
+                p[1] *= self.beta2
 
 
 class AdamWOptimizer(AdamOptimizer):
