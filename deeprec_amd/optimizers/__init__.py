@@ -210,10 +210,14 @@ class AdamAsyncOptimizer(AdamOptimizer):
 
     def _post_step(self):
         for ev in self.evs:
-            p = self._beta_powers.setdefault(
-                ev.name, torch.tensor([self.beta1, self.beta2],
-                                      device=ev.device)
-                if self.graph_safe else [self.beta1, self.beta2])
+            p = self._beta_powers.get(ev.name)
+            if p is None:
+                # NOTE: constructed lazily OUTSIDE any graph capture (the
+                # sparse-hyper call during warmup creates it first)
+                p = (torch.tensor([self.beta1, self.beta2],
+                                  device=ev.device)
+                     if self.graph_safe else [self.beta1, self.beta2])
+                self._beta_powers[ev.name] = p
             if torch.is_tensor(p):
                 from deeprec_amd.ops.build_ext import require_extension
                 require_extension().update_powers(p, self.beta1, self.beta2)
