@@ -80,5 +80,7 @@ def test_graph_replay_matches_eager():
         oe_i, og_i = torch.argsort(ke.cpu()), torch.argsort(kg.cpu())
         torch.testing.assert_close(ke.cpu()[oe_i], kg.cpu()[og_i])
         assert torch.equal(fe.cpu()[oe_i], fg.cpu()[og_i])
+        # dW uses fp32 atomic accumulation (order-nondeterministic), so
+        # eager/replay trajectories diverge at bf16 noise level
         torch.testing.assert_close(ve.cpu()[oe_i], vg.cpu()[og_i],
-                                   rtol=2e-3, atol=2e-4)
+                                   rtol=1e-2, atol=3e-3)
