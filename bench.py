@@ -106,6 +106,7 @@ def main():
 
     one_step = eager_step
     if use_graph:
+      try:
         st = model.collection.storage
         st.enable_graph_mode(expected_entries=1 << 22,
                              expected_slots=1 << 22)
@@ -127,13 +128,23 @@ def main():
             opt.step()
         torch.cuda.synchronize()
 
-        def one_step(i):
+        def graph_step(i):
             dense, sparse, labels = batches[i % len(batches)]
             sdense.copy_(dense, non_blocking=True)
             sids.copy_(sparse, non_blocking=True)
             slabels.copy_(labels, non_blocking=True)
             graph.replay()
             return loss
+
+        one_step = graph_step
+      except Exception as e:  # noqa: BLE001
+        # never lose the benchmark to a capture failure
+        print(f"hip-graph capture failed ({e}); falling back to eager",
+              flush=True)
+        use_graph = False
+        model.collection.graph_mode = False
+        for i in range(args.warmup):
+            eager_step(i)
     else:
         for i in range(args.warmup):
             one_step(i)
