@@ -278,10 +278,12 @@ class EmbeddingCollection:
     def _backward(self, grad_out, order, bounds, chunk_u, chunk_k0,
                   row_ids_cat, weights_cat, row_coeff, m, batch):
         if self.device.type == "cuda":
+            m_dev = (self.storage._last_m_dev
+                     if self.graph_mode else torch.Tensor())
             return self.storage.ext.group_pooled_bwd_strided(
                 grad_out.contiguous(), order, bounds, row_ids_cat,
                 weights_cat if weights_cat is not None else torch.Tensor(),
-                row_coeff, m, batch, self.n_tables, self.dim)
+                row_coeff, m, m_dev, batch, self.n_tables, self.dim)
         # CPU reference path
         g = grad_out.float().reshape(batch, self.n_tables, self.dim)
         grad_unique = torch.zeros(m, self.dim)

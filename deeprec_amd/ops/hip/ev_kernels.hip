@@ -739,8 +739,9 @@ __global__ void k_group_pooled_bwd_strided(
     const GradT* __restrict__ grad_out, const int32_t* __restrict__ order,
     const int32_t* __restrict__ bounds, const int32_t* __restrict__ row_ids,
     const float* __restrict__ weights, const float* __restrict__ row_coeff,
-    int m, int batch, int n_tables, int dim,
-    float* __restrict__ grad_unique) {
+    int m, const int32_t* __restrict__ m_dev, int batch, int n_tables,
+    int dim, float* __restrict__ grad_unique) {
+  if (m_dev) m = *m_dev;  // graph capture: true unique count on device
   int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t total = (int64_t)m * dim * SPLITS;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
@@ -1354,7 +1355,8 @@ torch::Tensor group_pooled_bwd(torch::Tensor grad_out, torch::Tensor order,
 torch::Tensor group_pooled_bwd_strided(
     torch::Tensor grad_out, torch::Tensor order, torch::Tensor bounds,
     torch::Tensor row_ids, torch::Tensor weights, torch::Tensor row_coeff,
-    int64_t m, int64_t batch, int64_t n_tables, int64_t dim) {
+    int64_t m, torch::Tensor m_dev, int64_t batch, int64_t n_tables,
+    int64_t dim) {
   constexpr int SPLITS = 8, CHUNK = 128;
   auto grad_unique = torch::zeros(
       {m, dim}, grad_out.options().dtype(torch::kFloat32));
@@ -1371,15 +1373,21 @@ torch::Tensor group_pooled_bwd_strided(
                 grad_out.data_ptr<at::BFloat16>()),
             order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
             row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
-            (int)m, (int)batch, (int)n_tables, (int)dim,
+            (int)m,
+            m_dev.defined() && m_dev.numel() ? m_dev.data_ptr<int32_t>()
+                                             : nullptr,
+            (int)batch, (int)n_tables, (int)dim,
             grad_unique.data_ptr<float>());
   } else {
     k_group_pooled_bwd_strided<float, SPLITS, CHUNK>
         <<<n_blocks(total), kBlock, 0, stream>>>(
             grad_out.data_ptr<float>(), order.data_ptr<int32_t>(),
             bounds.data_ptr<int32_t>(), row_ids.data_ptr<int32_t>(), wptr,
-            row_coeff.data_ptr<float>(), (int)m, (int)batch, (int)n_tables,
-            (int)dim, grad_unique.data_ptr<float>());
+            row_coeff.data_ptr<float>(), (int)m,
+            m_dev.defined() && m_dev.numel() ? m_dev.data_ptr<int32_t>()
+                                             : nullptr,
+            (int)batch, (int)n_tables, (int)dim,
+            grad_unique.data_ptr<float>());
   }
   return grad_unique;
 }

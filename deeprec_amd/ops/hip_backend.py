@@ -224,6 +224,7 @@ class HbmStorage:
         uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
         centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
         m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
+        self._last_m_dev = m_counter
         self.ext.ht_dedup_a_dev(values_cat, self.ht_keys, self.ht_freq,
                                 self.ht_version, self.ht_epoch,
                                 self.ht_compact, self._epoch_dev,
