@@ -512,17 +512,29 @@ __global__ void k_interact_bwd(const short* __restrict__ grad,
           *reinterpret_cast<const bf16x8*>(fsrc + t * 8);
     for (int t = lane; t < P; t += 64) g[t] = gsrc[t];
   }
+  // per-block pair-index LUT: the triangular index arithmetic in the
+  // inner loop cost more ALU than the actual FMAs. ALL threads (active or
+  // not) fill it and hit both barriers.
+  __shared__ short plut[40 * 40];
+  for (int t = threadIdx.x; t < F * F; t += blockDim.x) {
+    int i = t / F, j = t % F;
+    int lo = i < j ? i : j, hi = i < j ? j : i;
+    plut[t] = (short)(i == j ? -1
+                             : lo * F - (lo * (lo + 1)) / 2 + (hi - lo - 1));
+  }
   __syncthreads();
   if (!active) return;
   short* dst = reinterpret_cast<short*>(dfeats) + (int64_t)b * F * D;
   for (int t = lane; t < F * D; t += 64) {
     int i = t / D, d = t % D;
     float acc = 0.0f;
+    const short* prow = &plut[i * F];
+    const short* fd = f + d;
+#pragma unroll 4
     for (int j = 0; j < F; ++j) {
-      if (j == i) continue;
-      int lo = i < j ? i : j, hi = i < j ? j : i;
-      int p = lo * F - (lo * (lo + 1)) / 2 + (hi - lo - 1);
-      acc += bf2f_u16(g[p]) * bf2f_u16(f[j * D + d]);
+      int p = prow[j];
+      if (p < 0) continue;
+      acc += bf2f_u16(g[p]) * bf2f_u16(fd[j * D]);
     }
     dst[t] = f2bf_u16(acc);
   }
@@ -626,6 +638,7 @@ torch::Tensor interact_fwd(torch::Tensor feats, int64_t p_pad) {
 
 torch::Tensor interact_bwd(torch::Tensor grad, torch::Tensor feats) {
   int B = feats.size(0), F = feats.size(1), D = feats.size(2);
+  TORCH_CHECK(F <= 40, "interact_bwd: pair LUT supports F <= 40");
   int P = F * (F - 1) / 2;
   int P_pad = grad.size(1);
   auto dfeats = torch::empty_like(feats);
