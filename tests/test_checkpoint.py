@@ -192,3 +192,41 @@ def test_estimator_train_eval(tmp_path):
     assert latest_checkpoint(str(tmp_path)) is not None
     preds = next(est.predict(input_fn))
     assert preds.shape == (32,)
+
+
+def test_repartitioned_restore(tmp_path):
+    """Elastic capability: a checkpoint written by 2 shards restores into
+    1 (and vice versa) — the reference's 1000-bucket repartition-safe
+    scheme (Embedding-Variable-Export-Format.md)."""
+    from deeprec_amd.checkpoint.saver import Saver
+
+    # simulate two ranks' shard files: keys split by key % 2
+    ev_full = EmbeddingVariable("repart_ev", 4)
+    embedding_lookup_sparse(
+        ev_full, RaggedIds.from_dense(torch.arange(100).unsqueeze(1)),
+        combiner="sum")
+    keys, values, freqs, versions = ev_full.export()
+    import os
+
+    from safetensors.torch import save_file
+    path = os.path.join(str(tmp_path), "ckpt-1")
+    os.makedirs(path)
+    for r in range(2):
+        mask = (keys % 2) == r
+        save_file({"keys": keys[mask], "values": values[mask],
+                   "freqs": freqs[mask], "versions": versions[mask],
+                   "buckets": (keys[mask] % 1000).to(torch.int32)},
+                  os.path.join(path, f"ev-repart_ev-part{r}.safetensors"))
+    import json
+    json.dump({"global_step": 1},
+              open(os.path.join(path, "checkpoint.json"), "w"))
+
+    # restore the 2-shard checkpoint into a single EV
+    ev1 = EmbeddingVariable("repart_restored", 4)
+    ev1.name = "repart_ev"
+    saver = Saver(embedding_variables=[ev1])
+    saver.restore(path)
+    k1, v1, *_ = ev1.export()
+    o0, o1 = torch.argsort(keys), torch.argsort(k1)
+    torch.testing.assert_close(keys[o0], k1[o1])
+    torch.testing.assert_close(values[o0], v1[o1])
