@@ -58,6 +58,43 @@ class ParquetDataset:
                            for i, name in enumerate(rb.schema.names)}
 
 
+class CsvDataset:
+    """CSV batch reader with the reference modelzoo's Criteo column layout
+    (reference: build_model_input/parse_csv, modelzoo/dlrm/train.py:291):
+    label, I1..I13 continuous, C1..C26 categorical."""
+
+    def __init__(self, filenames, batch_size: int, column_names=None,
+                 num_epochs: int = 1, label_column: str = "clicked"):
+        import pandas  # noqa: F401
+        self.filenames = ([filenames] if isinstance(filenames, str)
+                          else list(filenames))
+        self.batch_size = batch_size
+        self.column_names = column_names
+        self.num_epochs = num_epochs
+        self.label_column = label_column
+
+    def __iter__(self):
+        import pandas as pd
+        for _ in range(self.num_epochs):
+            for fn in self.filenames:
+                for chunk in pd.read_csv(fn, names=self.column_names,
+                                         chunksize=self.batch_size):
+                    out = {}
+                    for col in chunk.columns:
+                        ser = chunk[col]
+                        if ser.dtype.kind in "iu":
+                            out[col] = torch.tensor(ser.to_numpy("int64"))
+                        elif ser.dtype.kind == "f":
+                            out[col] = torch.tensor(
+                                ser.to_numpy("float32"))
+                        else:  # hash string categoricals to int64 ids
+                            out[col] = torch.tensor(
+                                [hash(x) & ((1 << 48) - 1)
+                                 for x in ser.astype(str)],
+                                dtype=torch.int64)
+                    yield out
+
+
 class WorkQueue:
     """Elastic work distribution: a shared queue of work items (file paths
     / shard descriptors). Rank 0 owns the queue; take() hands the next item

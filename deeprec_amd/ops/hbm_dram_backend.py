@@ -19,9 +19,15 @@ re-designed for this engine's slot scheme:
   fp32 host applies on the pinned slabs (the cold tier is the slow path by
   construction);
 - promotion: at shrink/compaction the hottest keys by frequency are
-  re-packed into the HBM tier (LFU placement, reference CacheStrategy).
+  re-packed into the HBM tier (LFU placement, reference CacheStrategy);
+- SSD tier (reference: ssd_hash_kv.h mmap emb files): when
+  StorageOption.storage_path is set, the cold slabs are memory-mapped
+  files instead of pinned DRAM — the HBM_DRAM_SSD composition with the
+  OS page cache as the DRAM middle tier.
 """
 from __future__ import annotations
+
+import os
 
 import torch
 
@@ -61,11 +67,24 @@ class HbmDramStorage(HbmStorage):
         # keep the HBM slab at exactly the hot-tier budget
         if self.max_slots > self.hot_rows:
             self.values = self.values[: self.hot_rows].clone()
-        self.values_cold = torch.empty(1024, dim, dtype=torch.float32,
-                                       pin_memory=True)
+        self.storage_path = so.storage_path
+        if self.storage_path:
+            os.makedirs(self.storage_path, exist_ok=True)
+        self.values_cold = self._alloc_cold("values", 1024, dim)
         self.cold_slabs = {}
         self._cold_slab_init = {}
         self.default_values_cpu = self.default_values.cpu()
+
+    def _alloc_cold(self, name: str, rows: int, width: int,
+                    dtype=torch.float32) -> torch.Tensor:
+        """Cold-slab allocation: pinned DRAM, or a memory-mapped file under
+        storage_path (the SSD tier)."""
+        if not self.storage_path:
+            return torch.empty(rows, width, dtype=dtype, pin_memory=True)
+        fn = os.path.join(self.storage_path, f"{name}-{rows}.emb")
+        t = torch.from_file(fn, shared=True, size=rows * width,
+                            dtype=dtype)
+        return t.view(rows, width)
 
     # ---------------- tier plumbing ----------------
     def _init_limit(self) -> int:
@@ -82,14 +101,12 @@ class HbmDramStorage(HbmStorage):
         new_cap = cap
         while new_cap < cold_need:
             new_cap *= _GROW
-        nv = torch.empty(new_cap, self.dim, dtype=torch.float32,
-                         pin_memory=True)
+        nv = self._alloc_cold("values", new_cap, self.dim)
         nv[:cap] = self.values_cold
         self.values_cold = nv
         for name, t in list(self.cold_slabs.items()):
-            nt = torch.full((new_cap, t.shape[1]),
-                            self._cold_slab_init[name], dtype=t.dtype,
-                            pin_memory=True)
+            nt = self._alloc_cold(name, new_cap, t.shape[1], t.dtype)
+            nt.fill_(self._cold_slab_init[name])
             nt[: t.shape[0]] = t
             self.cold_slabs[name] = nt
 
@@ -103,9 +120,10 @@ class HbmDramStorage(HbmStorage):
 
     def _cold_slab(self, name, width, init_value, dtype=torch.float32):
         if name not in self.cold_slabs:
-            self.cold_slabs[name] = torch.full(
-                (self.values_cold.shape[0], width), init_value, dtype=dtype,
-                pin_memory=True)
+            t = self._alloc_cold(name, self.values_cold.shape[0], width,
+                                 dtype)
+            t.fill_(init_value)
+            self.cold_slabs[name] = t
             self._cold_slab_init[name] = init_value
         return self.cold_slabs[name]
 

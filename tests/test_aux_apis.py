@@ -173,3 +173,20 @@ def test_quantize_embeddings_roundtrip():
     q, scale = quantize_rows(v)
     v2 = dequantize_rows(q, scale)
     assert (v - v2).abs().max() < v.abs().max() / 100
+
+
+def test_csv_dataset(tmp_path):
+    from deeprec_amd.data.parquet import CsvDataset
+    fn = os.path.join(tmp_path, "train.csv")
+    with open(fn, "w") as f:
+        f.write("0,0.5,abc\n1,0.7,def\n1,0.1,abc\n")
+    ds = CsvDataset([fn], batch_size=2,
+                    column_names=["clicked", "I1", "C1"])
+    batches = list(ds)
+    assert len(batches) == 2
+    b = batches[0]
+    assert b["clicked"].dtype == torch.int64
+    assert b["I1"].dtype == torch.float32
+    assert b["C1"].dtype == torch.int64
+    # same string -> same id
+    assert int(batches[0]["C1"][0]) == int(batches[1]["C1"][0])
