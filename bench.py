@@ -114,7 +114,9 @@ def main():
         st.get_slab("adam_m", model.collection.dim, 0.0)
         st.get_slab("adam_v", model.collection.dim, 0.0)
         model.collection.graph_mode = True
-        for i in range(args.warmup):
+        # capture needs warmed state (optimizer slabs, device beta powers,
+        # dense adam state) — always run at least 2 eager steps
+        for i in range(max(args.warmup, 2)):
             eager_step(i)
         torch.cuda.synchronize()
         # static input buffers + whole-step capture
