@@ -13,7 +13,7 @@ from __future__ import annotations
 import json
 import os
 import threading
-from typing import Dict, List, Optional, Sequence
+from typing import List, Optional, Sequence
 
 import torch
 

@@ -24,8 +24,7 @@ import torch
 from deeprec_amd.embedding.options import EmbeddingVariableOption
 from deeprec_amd.embedding.ragged import RaggedIds
 from deeprec_amd.embedding.variable import EmbeddingVariable
-from deeprec_amd.embedding.lookup import (
-    embedding_lookup, embedding_lookup_sparse)
+from deeprec_amd.embedding.lookup import embedding_lookup
 
 
 class MultiHashVariable:

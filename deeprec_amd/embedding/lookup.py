@@ -16,7 +16,7 @@ fusion (ops/kv_variable_ops.cc:636).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import List, Sequence
 
 import torch
 

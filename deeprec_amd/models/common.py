@@ -12,7 +12,7 @@ from typing import List, Optional, Sequence
 import torch
 import torch.nn as nn
 
-from deeprec_amd.data.synthetic import NUM_DENSE, NUM_SPARSE
+from deeprec_amd.data.synthetic import NUM_SPARSE
 from deeprec_amd.embedding import EmbeddingVariableOption
 from deeprec_amd.embedding.collection import EmbeddingCollection
 

@@ -13,7 +13,7 @@ from typing import List, Optional
 import torch
 import torch.nn as nn
 
-from deeprec_amd.data.synthetic import NUM_DENSE, NUM_SPARSE
+from deeprec_amd.data.synthetic import NUM_SPARSE
 from deeprec_amd.embedding import (
     EmbeddingVariable, EmbeddingVariableOption,
     group_embedding_lookup_sparse,

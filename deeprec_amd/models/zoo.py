@@ -6,12 +6,12 @@ EmbeddingCollection sparse path and fused bf16 MLPs on GPU.
 from __future__ import annotations
 
 import math
-from typing import List, Optional
+from typing import List
 
 import torch
 import torch.nn as nn
 
-from deeprec_amd.data.synthetic import NUM_DENSE, NUM_SPARSE
+from deeprec_amd.data.synthetic import NUM_DENSE
 from deeprec_amd.models.common import RecModelBase, make_mlp
 
 

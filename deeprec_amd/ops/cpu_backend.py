@@ -13,8 +13,6 @@ non-goal beyond the Wide&Deep CPU plumbing config.
 from __future__ import annotations
 
 import math
-from typing import Optional
-
 import numpy as np
 import torch
 
@@ -165,7 +163,6 @@ class CpuStorage:
         return k % self.default_value_dim
 
     def _default_rows(self, keys):
-        import torch as _t
         if self.key_bits > 0:
             mask = (1 << self.key_bits) - 1
             return ((keys >> self.key_bits) * self.dvd_per_table

@@ -9,8 +9,6 @@ hard parts: host-coordinated rehash).
 from __future__ import annotations
 
 import math
-from typing import Optional
-
 import torch
 
 from deeprec_amd.embedding.options import (

@@ -12,7 +12,6 @@ sparse EV updates go through the fused HIP applies
 """
 from __future__ import annotations
 
-import math
 from typing import Iterable, List, Optional
 
 import torch

@@ -16,10 +16,8 @@ serving/processor/):
 from __future__ import annotations
 
 import glob
-import json
 import os
 import threading
-import time
 from typing import Callable, List, Optional
 
 import torch

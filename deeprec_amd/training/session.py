@@ -18,7 +18,7 @@ from typing import Callable, List, Optional
 
 import torch
 
-from deeprec_amd.embedding.variable import GLOBAL_STEP, get_global_step
+from deeprec_amd.embedding.variable import get_global_step
 
 log = logging.getLogger("deeprec_amd")
 
