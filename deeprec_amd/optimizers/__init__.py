@@ -209,7 +209,8 @@ class AdamAsyncOptimizer(AdamOptimizer):
             try:
                 return torch.optim.Adam(params, lr=self.lr,
                                         betas=(self.beta1, self.beta2),
-                                        eps=self.epsilon, fused=True)
+                                        eps=self.epsilon, fused=True,
+                                        capturable=True)
             except (RuntimeError, ValueError):
                 return torch.optim.Adam(params, lr=self.lr,
                                         betas=(self.beta1, self.beta2),
