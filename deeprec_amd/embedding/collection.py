@@ -158,7 +158,7 @@ class EmbeddingCollection:
         return _CollectionLookup.apply(
             self._anchor, self, uniq, slots, inverse, offsets_cat,
             row_ids_cat, order, bounds, chunk_u, chunk_k0, row_coeff,
-            weights_cat, batch, out_dtype)
+            weights_cat, batch, out_dtype, False)
 
     _CHUNK = 128
 
@@ -232,7 +232,7 @@ class EmbeddingCollection:
         return _CollectionLookup.apply(
             self._anchor, self, uniq, slots, inverse, cache["offsets"],
             cache["row_ids"], order, bounds, chunk_u, chunk_k0,
-            cache["row_coeff"], None, batch, out_dtype)
+            cache["row_coeff"], None, batch, out_dtype, True)
 
     def _forward(self, uniq, slots, inverse, offsets_cat, weights_cat, batch,
                  out_dtype, emb_override=None):
@@ -276,14 +276,16 @@ class EmbeddingCollection:
         return torch.cat(outs, dim=1)  # [B, N*D], table-major inner
 
     def _backward(self, grad_out, order, bounds, chunk_u, chunk_k0,
-                  row_ids_cat, weights_cat, row_coeff, m, batch):
+                  row_ids_cat, weights_cat, row_coeff, m, batch,
+                  identity_rows=False):
         if self.device.type == "cuda":
             m_dev = (self.storage._last_m_dev
                      if self.graph_mode else torch.Tensor())
             return self.storage.ext.group_pooled_bwd_strided(
                 grad_out.contiguous(), order, bounds, row_ids_cat,
                 weights_cat if weights_cat is not None else torch.Tensor(),
-                row_coeff, m, m_dev, batch, self.n_tables, self.dim)
+                row_coeff, m, m_dev, batch, self.n_tables, self.dim,
+                identity_rows)
         # CPU reference path
         g = grad_out.float().reshape(batch, self.n_tables, self.dim)
         grad_unique = torch.zeros(m, self.dim)
@@ -359,11 +361,12 @@ class _CollectionLookup(torch.autograd.Function):
     @staticmethod
     def forward(ctx, anchor, coll, uniq, slots, inverse, offsets_cat,
                 row_ids_cat, order, bounds, chunk_u, chunk_k0, row_coeff,
-                weights_cat, batch, out_dtype):
+                weights_cat, batch, out_dtype, identity_rows=False):
         out = coll._forward(uniq, slots, inverse, offsets_cat, weights_cat,
                             batch, out_dtype)
         ctx.coll = coll
         ctx.batch = batch
+        ctx.identity_rows = identity_rows
         ctx.save_for_backward(uniq, slots, order, bounds, row_ids_cat,
                               row_coeff)
         ctx.chunks = (chunk_u, chunk_k0)
@@ -378,6 +381,7 @@ class _CollectionLookup(torch.autograd.Function):
         grad_unique = coll._backward(grad_out, order, bounds, chunk_u,
                                      chunk_k0, row_ids_cat,
                                      ctx.weights_cat, row_coeff,
-                                     uniq.numel(), ctx.batch)
+                                     uniq.numel(), ctx.batch,
+                                     ctx.identity_rows)
         coll.accumulate_grad(slots, uniq, grad_unique)
-        return (torch.zeros_like(coll._anchor),) + (None,) * 14
+        return (torch.zeros_like(coll._anchor),) + (None,) * 15

@@ -97,7 +97,8 @@ def _chunked_ev_backward(ev, grad_out, inverse, offsets, row_ids, counts,
     return ext.group_pooled_bwd_strided(
         grad_out.contiguous(), order, bounds, row_ids.to(torch.int32),
         weights.float() if weights is not None else torch.Tensor(),
-        row_coeff, m, torch.Tensor(), offsets.numel() - 1, 1, ev.dim)
+        row_coeff, m, torch.Tensor(), offsets.numel() - 1, 1, ev.dim,
+        False)
 
 
 def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,

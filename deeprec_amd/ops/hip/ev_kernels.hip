@@ -734,7 +734,9 @@ __global__ void k_group_pooled_bwd_chunked(
 // SPLITS*CHUNK — zipf-hot keys are split across SPLITS concurrent
 // accumulators (bounded atomics) while the host does zero glue work
 // (the chunk-array build was 6 kernels + a repeat_interleave sync/step).
-template <typename GradT, int SPLITS, int CHUNK>
+// IDENTITY: the matrix fast path (one id per table per sample) has
+// row_ids == arange and unit combiner coefficients — skip both reads.
+template <typename GradT, int SPLITS, int CHUNK, bool IDENTITY>
 __global__ void k_group_pooled_bwd_strided(
     const GradT* __restrict__ grad_out, const int32_t* __restrict__ order,
     const int32_t* __restrict__ bounds, const int32_t* __restrict__ row_ids,
@@ -758,7 +760,7 @@ __global__ void k_group_pooled_bwd_strided(
       int k1 = min(k0 + CHUNK, end);
       for (int k = k0; k < k1; ++k) {
         int j = order[k];
-        int rid = row_ids[j];
+        int rid = IDENTITY ? j : row_ids[j];
         int table = rid / batch;
         int b = rid % batch;
         float g;
@@ -767,8 +769,12 @@ __global__ void k_group_pooled_bwd_strided(
           g = bf2f(grad_out[gidx]);
         else
           g = grad_out[gidx];
-        float w = weights ? weights[j] : 1.0f;
-        acc += w * row_coeff[rid] * g;
+        if constexpr (IDENTITY) {
+          acc += g;
+        } else {
+          float w = weights ? weights[j] : 1.0f;
+          acc += w * row_coeff[rid] * g;
+        }
       }
     }
     int64_t o = (int64_t)u * dim + d;
@@ -1356,7 +1362,7 @@ torch::Tensor group_pooled_bwd_strided(
     torch::Tensor grad_out, torch::Tensor order, torch::Tensor bounds,
     torch::Tensor row_ids, torch::Tensor weights, torch::Tensor row_coeff,
     int64_t m, torch::Tensor m_dev, int64_t batch, int64_t n_tables,
-    int64_t dim) {
+    int64_t dim, bool identity_rows) {
   constexpr int SPLITS = 8, CHUNK = 128;
   auto grad_unique = torch::zeros(
       {m, dim}, grad_out.options().dtype(torch::kFloat32));
@@ -1366,29 +1372,21 @@ torch::Tensor group_pooled_bwd_strided(
   const float* wptr =
       weights.defined() && weights.numel() ? weights.data_ptr<float>()
                                            : nullptr;
-  if (grad_out.scalar_type() == torch::kBFloat16) {
-    k_group_pooled_bwd_strided<__hip_bfloat16, SPLITS, CHUNK>
-        <<<n_blocks(total), kBlock, 0, stream>>>(
-            reinterpret_cast<const __hip_bfloat16*>(
-                grad_out.data_ptr<at::BFloat16>()),
-            order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
-            row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
-            (int)m,
-            m_dev.defined() && m_dev.numel() ? m_dev.data_ptr<int32_t>()
-                                             : nullptr,
-            (int)batch, (int)n_tables, (int)dim,
-            grad_unique.data_ptr<float>());
+  const int32_t* mdp =
+      m_dev.defined() && m_dev.numel() ? m_dev.data_ptr<int32_t>() : nullptr;
+  const bool is_bf16 = grad_out.scalar_type() == torch::kBFloat16;
+#define LAUNCH_BWD(T, PTR, IDENT)                                            k_group_pooled_bwd_strided<T, SPLITS, CHUNK, IDENT>                            <<<n_blocks(total), kBlock, 0, stream>>>(                                      PTR, order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),                row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),            (int)m, mdp, (int)batch, (int)n_tables, (int)dim,                          grad_unique.data_ptr<float>())
+  if (is_bf16) {
+    auto* gp = reinterpret_cast<const __hip_bfloat16*>(
+        grad_out.data_ptr<at::BFloat16>());
+    if (identity_rows) LAUNCH_BWD(__hip_bfloat16, gp, true);
+    else LAUNCH_BWD(__hip_bfloat16, gp, false);
   } else {
-    k_group_pooled_bwd_strided<float, SPLITS, CHUNK>
-        <<<n_blocks(total), kBlock, 0, stream>>>(
-            grad_out.data_ptr<float>(), order.data_ptr<int32_t>(),
-            bounds.data_ptr<int32_t>(), row_ids.data_ptr<int32_t>(), wptr,
-            row_coeff.data_ptr<float>(), (int)m,
-            m_dev.defined() && m_dev.numel() ? m_dev.data_ptr<int32_t>()
-                                             : nullptr,
-            (int)batch, (int)n_tables, (int)dim,
-            grad_unique.data_ptr<float>());
+    auto* gp = grad_out.data_ptr<float>();
+    if (identity_rows) LAUNCH_BWD(float, gp, true);
+    else LAUNCH_BWD(float, gp, false);
   }
+#undef LAUNCH_BWD
   return grad_unique;
 }
 
