@@ -22,36 +22,54 @@ BATCH_SIZES = {
 DEFAULT_BATCH = 8192
 
 
-def bench_model(name, steps, warmup, device):
+def bench_model(name, steps, warmup, device, use_graph=True):
     from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    from deeprec_amd.embedding.options import EmbeddingVariableOption
     from deeprec_amd.models import MODEL_REGISTRY, SEQUENCE_MODELS
     from deeprec_amd.optimizers import AdamAsyncOptimizer
 
     torch.manual_seed(0)
     is_seq = name in SEQUENCE_MODELS
+    use_graph = use_graph and not is_seq and device.type == "cuda"
     batch = BATCH_SIZES.get(name, DEFAULT_BATCH)
-    m = MODEL_REGISTRY[name](device=device, bf16=device.type == "cuda")
+    kw = {}
+    if use_graph:
+        kw["ev_option"] = EmbeddingVariableOption(init_capacity=1 << 23)
+    m = MODEL_REGISTRY[name](device=device, bf16=device.type == "cuda",
+                             **kw)
     ds = CriteoSyntheticDataset(batch_size=batch, seed=1, device=device,
                                 matrix_format=not is_seq)
     opt = AdamAsyncOptimizer(params=m.parameters(),
-                             embedding_variables=m.embedding_variables())
+                             embedding_variables=m.embedding_variables(),
+                             graph_safe=use_graph)
     if is_seq:
         batches = [ds.next_seq_batch(seq_len=50) for _ in range(8)]
     else:
         batches = [ds.next_batch() for _ in range(8)]
 
-    def step(i):
-        b = batches[i % len(batches)]
-        if is_seq:
-            dense, ids, seq, target, labels = b
-            logits = m(dense, ids[:, :m.num_sparse], seq, target)
-        else:
-            dense, ids, labels = b
-            logits = m(dense, ids)
-        loss = m.loss_fn(logits, labels)
-        opt.zero_grad()
-        loss.backward()
-        opt.step()
+    if use_graph:
+        from deeprec_amd.training.graph_step import GraphedTrainStep
+
+        def loss_fn(model, dense, ids, labels):
+            return model.loss_fn(model(dense, ids), labels)
+
+        gstep = GraphedTrainStep(m, opt, loss_fn, batches[0])
+
+        def step(i):
+            gstep(batches[i % len(batches)])
+    else:
+        def step(i):
+            b = batches[i % len(batches)]
+            if is_seq:
+                dense, ids, seq, target, labels = b
+                logits = m(dense, ids[:, :m.num_sparse], seq, target)
+            else:
+                dense, ids, labels = b
+                logits = m(dense, ids)
+            loss = m.loss_fn(logits, labels)
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
 
     for i in range(warmup):
         step(i)
