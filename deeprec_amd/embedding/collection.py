@@ -281,11 +281,17 @@ class EmbeddingCollection:
         if self.device.type == "cuda":
             m_dev = (self.storage._last_m_dev
                      if self.graph_mode else torch.Tensor())
+            # zipf batches (nnz >> m) want hot-key splitting; mostly-
+            # unique batches want 1 split (8x fewer idle probes)
+            nnz = row_ids_cat.numel()
+            # under capture m is nnz-padded, so the ratio test is wrong
+            # there; captured workloads are the zipf ones -> split
+            splits = 8 if (self.graph_mode or nnz > 2 * m) else 1
             return self.storage.ext.group_pooled_bwd_strided(
                 grad_out.contiguous(), order, bounds, row_ids_cat,
                 weights_cat if weights_cat is not None else torch.Tensor(),
                 row_coeff, m, m_dev, batch, self.n_tables, self.dim,
-                identity_rows)
+                identity_rows, splits)
         # CPU reference path
         g = grad_out.float().reshape(batch, self.n_tables, self.dim)
         grad_unique = torch.zeros(m, self.dim)

@@ -94,11 +94,13 @@ def _chunked_ev_backward(ev, grad_out, inverse, offsets, row_ids, counts,
         row_coeff = torch.where(lengths > 0,
                                 1.0 / denom.clamp(min=1e-12),
                                 torch.zeros_like(lengths))
+    nnz = inverse.numel()
+    splits = 8 if nnz > 2 * m else 1
     return ext.group_pooled_bwd_strided(
         grad_out.contiguous(), order, bounds, row_ids.to(torch.int32),
         weights.float() if weights is not None else torch.Tensor(),
         row_coeff, m, torch.Tensor(), offsets.numel() - 1, 1, ev.dim,
-        False)
+        False, splits)
 
 
 def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,
