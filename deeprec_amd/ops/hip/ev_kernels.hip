@@ -1358,35 +1358,72 @@ torch::Tensor group_pooled_bwd(torch::Tensor grad_out, torch::Tensor order,
   return grad_unique;
 }
 
+template <typename T, int SPL>
+static void launch_bwd_strided(const T* gp, torch::Tensor& order,
+                               torch::Tensor& bounds, torch::Tensor& row_ids,
+                               const float* wptr, torch::Tensor& row_coeff,
+                               int64_t m, const int32_t* mdp, int64_t batch,
+                               int64_t n_tables, int64_t dim,
+                               bool identity_rows, torch::Tensor& out,
+                               hipStream_t stream) {
+  constexpr int CHUNK = 128;
+  int64_t total = m * dim * SPL;
+  if (identity_rows) {
+    k_group_pooled_bwd_strided<T, SPL, CHUNK, true>
+        <<<n_blocks(total), kBlock, 0, stream>>>(
+            gp, order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
+            row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
+            (int)m, mdp, (int)batch, (int)n_tables, (int)dim,
+            out.data_ptr<float>());
+  } else {
+    k_group_pooled_bwd_strided<T, SPL, CHUNK, false>
+        <<<n_blocks(total), kBlock, 0, stream>>>(
+            gp, order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
+            row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
+            (int)m, mdp, (int)batch, (int)n_tables, (int)dim,
+            out.data_ptr<float>());
+  }
+}
+
 torch::Tensor group_pooled_bwd_strided(
     torch::Tensor grad_out, torch::Tensor order, torch::Tensor bounds,
     torch::Tensor row_ids, torch::Tensor weights, torch::Tensor row_coeff,
     int64_t m, torch::Tensor m_dev, int64_t batch, int64_t n_tables,
-    int64_t dim, bool identity_rows) {
-  constexpr int SPLITS = 8, CHUNK = 128;
+    int64_t dim, bool identity_rows, int64_t splits) {
+  // splits: 8 for duplication-heavy (zipf) batches — hot keys fan out
+  // across concurrent accumulators; 1 for mostly-unique batches (long
+  // sequences) where extra splits are 8x wasted bounds probes.
   auto grad_unique = torch::zeros(
       {m, dim}, grad_out.options().dtype(torch::kFloat32));
-  int64_t total = m * dim * SPLITS;
-  if (total == 0) return grad_unique;
+  if (m * dim == 0) return grad_unique;
   auto stream = current_stream();
   const float* wptr =
       weights.defined() && weights.numel() ? weights.data_ptr<float>()
                                            : nullptr;
   const int32_t* mdp =
       m_dev.defined() && m_dev.numel() ? m_dev.data_ptr<int32_t>() : nullptr;
-  const bool is_bf16 = grad_out.scalar_type() == torch::kBFloat16;
-#define LAUNCH_BWD(T, PTR, IDENT)                                            k_group_pooled_bwd_strided<T, SPLITS, CHUNK, IDENT>                            <<<n_blocks(total), kBlock, 0, stream>>>(                                      PTR, order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),                row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),            (int)m, mdp, (int)batch, (int)n_tables, (int)dim,                          grad_unique.data_ptr<float>())
-  if (is_bf16) {
+  if (grad_out.scalar_type() == torch::kBFloat16) {
     auto* gp = reinterpret_cast<const __hip_bfloat16*>(
         grad_out.data_ptr<at::BFloat16>());
-    if (identity_rows) LAUNCH_BWD(__hip_bfloat16, gp, true);
-    else LAUNCH_BWD(__hip_bfloat16, gp, false);
+    if (splits <= 1)
+      launch_bwd_strided<__hip_bfloat16, 1>(
+          gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
+          n_tables, dim, identity_rows, grad_unique, stream);
+    else
+      launch_bwd_strided<__hip_bfloat16, 8>(
+          gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
+          n_tables, dim, identity_rows, grad_unique, stream);
   } else {
     auto* gp = grad_out.data_ptr<float>();
-    if (identity_rows) LAUNCH_BWD(float, gp, true);
-    else LAUNCH_BWD(float, gp, false);
+    if (splits <= 1)
+      launch_bwd_strided<float, 1>(
+          gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
+          n_tables, dim, identity_rows, grad_unique, stream);
+    else
+      launch_bwd_strided<float, 8>(
+          gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
+          n_tables, dim, identity_rows, grad_unique, stream);
   }
-#undef LAUNCH_BWD
   return grad_unique;
 }
 
