@@ -31,8 +31,21 @@ class _SeqBase(RecModelBase):
         return [self.collection, self.item_ev]
 
     def seq_emb(self, seq_ids: torch.Tensor, train=True) -> torch.Tensor:
-        """[B, T] -> [B, T, item_dim]; id 0 = padding (masked by caller)."""
-        return embedding_lookup(self.item_ev, seq_ids, train=train).float()
+        """[B, T] -> [B, T, item_dim]; id 0 = padding -> zero embedding.
+
+        Padding is excluded from the lookup entirely (reference semantics:
+        masked positions train nothing). It also matters for performance:
+        ~half of all sequence positions are padding, and a single
+        200k-occurrence key would otherwise dominate the grad scatter."""
+        flat = seq_ids.reshape(-1)
+        mask = flat > 0
+        emb = torch.zeros(flat.numel(), self.item_dim,
+                          device=flat.device)
+        emb = emb.index_put(
+            (mask.nonzero().squeeze(1),),
+            embedding_lookup(self.item_ev, flat[mask],
+                             train=train).float())
+        return emb.reshape(*seq_ids.shape, self.item_dim)
 
 
 class DIN(_SeqBase):
