@@ -88,7 +88,8 @@ def _chunked_ev_backward(ev, grad_out, inverse, offsets, row_ids, counts,
             denom = lengths if combiner == "mean" else lengths.sqrt()
         else:
             acc = torch.zeros_like(lengths)
-            w = weights.float() if combiner == "mean"                 else weights.float() ** 2
+            w = (weights.float() if combiner == "mean"
+                 else weights.float() ** 2)
             acc.index_add_(0, row_ids.long(), w)
             denom = acc if combiner == "mean" else acc.sqrt()
         row_coeff = torch.where(lengths > 0,
