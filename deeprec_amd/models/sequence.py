@@ -102,7 +102,11 @@ class DIEN(_SeqBase):
                  bf16=True, **kw):
         super().__init__(embedding_dim, item_dim, device, bf16, name="dien",
                          **kw)
-        self.gru = nn.GRU(item_dim, gru_hidden, batch_first=True)
+        from deeprec_amd.ops.fused_gru import FusedGRU
+        # fused single-kernel recurrences (MIOpen's RNN path ran this at
+        # 44 ms/step; the python AUGRU loop added ~600 launches)
+        self.gru = FusedGRU(item_dim, gru_hidden)
+        self.augru = FusedGRU(gru_hidden, gru_hidden)
         if self.bf16:
             from deeprec_amd.ops.fused_mlp import FusedLinear
             self.att = nn.Sequential(
@@ -113,7 +117,6 @@ class DIEN(_SeqBase):
             self.att = nn.Sequential(
                 nn.Linear(gru_hidden * 2, att_hidden), nn.Sigmoid(),
                 nn.Linear(att_hidden, 1))
-        self.augru_cell = nn.GRUCell(gru_hidden, gru_hidden)
         self.target_proj = nn.Linear(item_dim, gru_hidden)
         in_dim = (NUM_DENSE + self.num_sparse * embedding_dim
                   + gru_hidden + item_dim)
@@ -127,20 +130,16 @@ class DIEN(_SeqBase):
         target = embedding_lookup(self.item_ev, target_ids,
                                   train=train).float()   # [B,Di]
         mask = (seq_ids > 0).float()
-        h_seq, _ = self.gru(seq)                         # [B,T,H]
+        h_seq = self.gru(seq)                            # [B,T,H]
         tgt_h = self.target_proj(target)                 # [B,H]
         att_in = torch.cat(
             [h_seq, tgt_h.unsqueeze(1).expand_as(h_seq)], 2)
         scores = self.att(att_in).float().squeeze(2)
         scores = scores.masked_fill(mask == 0, -1e9)
-        alpha = torch.softmax(scores, 1)                 # [B,T]
-        # AUGRU: attention scales the update gate — implemented as
-        # h_t = (1-a)*h_{t-1} + a*GRUCell(x_t, h_{t-1})
-        h = torch.zeros(seq.shape[0], h_seq.shape[2], device=seq.device)
-        for t in range(h_seq.shape[1]):
-            h_new = self.augru_cell(h_seq[:, t], h)
-            a = (alpha[:, t] * mask[:, t]).unsqueeze(1)
-            h = (1 - a) * h + a * h_new
+        alpha = torch.softmax(scores, 1) * mask          # [B,T]
+        # AUGRU: attention gates the update — h'=(1-a)h + a*GRU(h,x);
+        # a=0 at padding leaves h untouched, so [:, -1] is the final state
+        h = self.augru(h_seq, alpha)[:, -1]
         x = torch.cat([dense, emb.flatten(1).float(), h, target], 1)
         with self.amp():
             out = self.mlp(x.to(self.compute_dtype))
