@@ -256,67 +256,6 @@ __global__ void k_linear_dw(const short* __restrict__ G,
   }
 }
 
-// dW via pre-transposed operands: Gt [N, M] and At [K, M] row-major, so
-// every fragment is a contiguous 16 B row load (the load_frag_col version
-// was 8 scalar loads per fragment and dominated the backward at 31 us per
-// layer). dW[n,k] = sum_m Gt[n][m] * At[k][m]; fp32 atomic accumulation
-// over M chunks; fused dbias from Gt row sums (k-tile 0 only).
-__global__ void k_linear_dw_nt(const short* __restrict__ Gt,
-                               const short* __restrict__ At, int M, int N,
-                               int K, int chunk_rows, float* __restrict__ dW,
-                               float* __restrict__ dbias) {
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  const int tiles_k = (K + 15) / 16;
-  const int tiles_n = (N + 15) / 16;
-  const int tile_id = blockIdx.x % (tiles_n * tiles_k);
-  const int chunk = blockIdx.x / (tiles_n * tiles_k);
-  const int n0 = (tile_id / tiles_k) * 16;
-  const int k0 = (tile_id % tiles_k) * 16;
-  const int rows_per_wave = chunk_rows / 4;
-  const int mbeg = chunk * chunk_rows + wave * rows_per_wave;
-  const int mend = min(mbeg + rows_per_wave, M);
-
-  const int row_g = n0 + (lane & 15);
-  const int row_a = k0 + (lane & 15);
-  const int kgrp = (lane >> 4) * 8;
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  float bsum = 0.0f;
-  const bool do_bias = (dbias != nullptr) && (k0 == 0);
-  for (int m = mbeg; m < mend; m += 32) {
-    bf16x8 a = load_frag_row(Gt, row_g, m + kgrp, N, M);
-    bf16x8 b = load_frag_row(At, row_a, m + kgrp, K, M);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-    if (do_bias) {
-#pragma unroll
-      for (int i = 0; i < 8; ++i) bsum += bf2f_u16(a[i]);
-    }
-  }
-  const int ck = k0 + (lane & 15);
-  if (ck < K) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int cn = n0 + (lane >> 4) * 4 + i;
-      if (cn >= N) continue;
-      atomicAdd(&dW[(int64_t)cn * K + ck], acc[i]);
-    }
-  }
-  if (do_bias) {
-    __shared__ float bred[4][16];
-    float v = bsum;
-    for (int off = 16; off < 64; off += 16)
-      v += __shfl(bsum, (lane & 15) + off);
-    if (lane < 16) bred[wave][lane] = v;
-    __syncthreads();
-    if (wave == 0 && lane < 16) {
-      float t = bred[0][lane] + bred[1][lane] + bred[2][lane] +
-                bred[3][lane];
-      int cn = n0 + lane;
-      if (cn < N) atomicAdd(&dbias[cn], t);
-    }
-  }
-}
-
 // LDS-staged dW: block = (16-row N tile) x (64-col K group) x (128-row M
 // chunk). G[128,16] and A[128,64] chunks are staged with coalesced row
 // loads; the four waves compute the four 16x16 k-subtiles from LDS
@@ -592,27 +531,6 @@ std::tuple<torch::Tensor, torch::Tensor> linear_dw(torch::Tensor g,
   return {dw, db};
 }
 
-std::tuple<torch::Tensor, torch::Tensor> linear_dw_nt(torch::Tensor gt,
-                                                      torch::Tensor xt,
-                                                      bool want_bias) {
-  // gt: [N, M], xt: [K, M] (pre-transposed, contiguous)
-  int N = gt.size(0), M = gt.size(1), K = xt.size(0);
-  auto dw = torch::zeros({N, K}, gt.options().dtype(torch::kFloat32));
-  auto db = want_bias
-                ? torch::zeros({N}, gt.options().dtype(torch::kFloat32))
-                : torch::Tensor();
-  int tiles = ((N + 15) / 16) * ((K + 15) / 16);
-  int chunk_rows = 128;
-  while ((int64_t)tiles * ((M + chunk_rows - 1) / chunk_rows) > 8192 &&
-         chunk_rows < M)
-    chunk_rows *= 2;
-  int m_chunks = (M + chunk_rows - 1) / chunk_rows;
-  k_linear_dw_nt<<<tiles * m_chunks, 256, 0, dense_stream()>>>(
-      bf_ptr(gt), bf_ptr(xt), M, N, K, chunk_rows, dw.data_ptr<float>(),
-      want_bias ? db.data_ptr<float>() : nullptr);
-  return {dw, db};
-}
-
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor out, int64_t act) {
   auto g = torch::empty_like(dy);
   int64_t n = dy.numel();
@@ -656,6 +574,5 @@ void register_dense(py::module_& mod) {
   mod.def("linear_fwd", &linear_fwd);
   mod.def("linear_dx", &linear_dx);
   mod.def("linear_dw", &linear_dw);
-  mod.def("linear_dw_nt", &linear_dw_nt);
   mod.def("act_bwd", &act_bwd);
 }

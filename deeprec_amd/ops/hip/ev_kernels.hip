@@ -649,91 +649,6 @@ __global__ void k_group_pooled_fwd(
   }
 }
 
-// Atomic-free grouped backward: occurrences are CSR-grouped by unique key
-// (order = argsort(inverse), bounds = cumsum(counts)); thread (u, d) sums
-// its occurrences and writes grad_unique[u][d] once. row_coeff[N*B] is the
-// per-pooled-row combiner coefficient precomputed on device.
-template <typename GradT>
-__global__ void k_group_pooled_bwd(
-    const GradT* __restrict__ grad_out,  // [B, N*D]
-    const int32_t* __restrict__ order, const int32_t* __restrict__ bounds,
-    const int32_t* __restrict__ row_ids,  // [nnz] pooled row (t*B + b)
-    const float* __restrict__ weights, const float* __restrict__ row_coeff,
-    int m, int batch, int n_tables, int dim,
-    float* __restrict__ grad_unique) {
-  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-  int64_t total = (int64_t)m * dim;
-  int64_t stride = gridDim.x * (int64_t)blockDim.x;
-  for (; t < total; t += stride) {
-    int u = (int)(t / dim);
-    int d = (int)(t % dim);
-    float acc = 0.0f;
-    for (int k = bounds[u]; k < bounds[u + 1]; ++k) {
-      int j = order[k];
-      int rid = row_ids[j];
-      int table = rid / batch;
-      int b = rid % batch;
-      float g;
-      int64_t gidx = ((int64_t)b * n_tables + table) * dim + d;
-      if constexpr (std::is_same_v<GradT, __hip_bfloat16>)
-        g = bf2f(grad_out[gidx]);
-      else
-        g = grad_out[gidx];
-      float w = weights ? weights[j] : 1.0f;
-      acc += w * row_coeff[rid] * g;
-    }
-    grad_unique[t] = acc;
-  }
-}
-
-// Chunked variant: hot keys (zipf head) can have 10k+ occurrences, which
-// makes the one-thread-per-key loop tail-latency bound. Work is split into
-// fixed-size chunks of occurrences; chunks of the same key combine via
-// atomicAdd (contention is bounded by count/CHUNK, e.g. 10k/128 = 79).
-template <typename GradT>
-__global__ void k_group_pooled_bwd_chunked(
-    const GradT* __restrict__ grad_out, const int32_t* __restrict__ order,
-    const int32_t* __restrict__ bounds, const int32_t* __restrict__ row_ids,
-    const float* __restrict__ weights, const float* __restrict__ row_coeff,
-    const int32_t* __restrict__ chunk_u, const int32_t* __restrict__ chunk_k0,
-    int n_chunks, int chunk_size, int batch, int n_tables, int dim,
-    float* __restrict__ grad_unique) {
-  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
-  int64_t total = (int64_t)n_chunks * dim;
-  int64_t stride = gridDim.x * (int64_t)blockDim.x;
-  for (; t < total; t += stride) {
-    int c = (int)(t / dim);
-    int d = (int)(t % dim);
-    int u = chunk_u[c];
-    int k0 = chunk_k0[c];
-    int k1 = min(k0 + chunk_size, (int)bounds[u + 1]);
-    float acc = 0.0f;
-    for (int k = k0; k < k1; ++k) {
-      int j = order[k];
-      int rid = row_ids[j];
-      int table = rid / batch;
-      int b = rid % batch;
-      float g;
-      int64_t gidx = ((int64_t)b * n_tables + table) * dim + d;
-      if constexpr (std::is_same_v<GradT, __hip_bfloat16>)
-        g = bf2f(grad_out[gidx]);
-      else
-        g = grad_out[gidx];
-      float w = weights ? weights[j] : 1.0f;
-      acc += w * row_coeff[rid] * g;
-    }
-    int64_t o = (int64_t)u * dim + d;
-    bool single = (k0 == bounds[u] && k1 == bounds[u + 1]);
-    if (single) grad_unique[o] = acc;     // sole chunk: plain store
-    else atomicAdd(&grad_unique[o], acc);
-  }
-}
-
-// Strided variant: no host-built chunk arrays. Thread (u, d, s) with
-// s in [0, SPLITS) covers occurrence windows [k0, k0+CHUNK) striding by
-// SPLITS*CHUNK — zipf-hot keys are split across SPLITS concurrent
-// accumulators (bounded atomics) while the host does zero glue work
-// (the chunk-array build was 6 kernels + a repeat_interleave sync/step).
 // IDENTITY: the matrix fast path (one id per table per sample) has
 // row_ids == arange and unit combiner coefficients — skip both reads.
 template <typename GradT, int SPLITS, int CHUNK, bool IDENTITY>
@@ -1326,38 +1241,6 @@ torch::Tensor group_pooled_fwd_direct(
   return out;
 }
 
-torch::Tensor group_pooled_bwd(torch::Tensor grad_out, torch::Tensor order,
-                               torch::Tensor bounds, torch::Tensor row_ids,
-                               torch::Tensor weights, torch::Tensor row_coeff,
-                               int64_t m, int64_t batch, int64_t n_tables,
-                               int64_t dim) {
-  auto grad_unique = torch::empty(
-      {m, dim}, grad_out.options().dtype(torch::kFloat32));
-  int64_t total = m * dim;
-  if (total == 0) return grad_unique;
-  auto stream = current_stream();
-  const float* wptr =
-      weights.defined() && weights.numel() ? weights.data_ptr<float>()
-                                           : nullptr;
-  if (grad_out.scalar_type() == torch::kBFloat16) {
-    k_group_pooled_bwd<__hip_bfloat16>
-        <<<n_blocks(total), kBlock, 0, stream>>>(
-            reinterpret_cast<const __hip_bfloat16*>(
-                grad_out.data_ptr<at::BFloat16>()),
-            order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
-            row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
-            (int)m, (int)batch, (int)n_tables, (int)dim,
-            grad_unique.data_ptr<float>());
-  } else {
-    k_group_pooled_bwd<float><<<n_blocks(total), kBlock, 0, stream>>>(
-        grad_out.data_ptr<float>(), order.data_ptr<int32_t>(),
-        bounds.data_ptr<int32_t>(), row_ids.data_ptr<int32_t>(), wptr,
-        row_coeff.data_ptr<float>(), (int)m, (int)batch, (int)n_tables,
-        (int)dim, grad_unique.data_ptr<float>());
-  }
-  return grad_unique;
-}
-
 template <typename T, int SPL>
 static void launch_bwd_strided(const T* gp, torch::Tensor& order,
                                torch::Tensor& bounds, torch::Tensor& row_ids,
@@ -1423,43 +1306,6 @@ torch::Tensor group_pooled_bwd_strided(
       launch_bwd_strided<float, 8>(
           gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
           n_tables, dim, identity_rows, grad_unique, stream);
-  }
-  return grad_unique;
-}
-
-torch::Tensor group_pooled_bwd_chunked(
-    torch::Tensor grad_out, torch::Tensor order, torch::Tensor bounds,
-    torch::Tensor row_ids, torch::Tensor weights, torch::Tensor row_coeff,
-    torch::Tensor chunk_u, torch::Tensor chunk_k0, int64_t chunk_size,
-    int64_t m, int64_t batch, int64_t n_tables, int64_t dim) {
-  auto grad_unique = torch::zeros(
-      {m, dim}, grad_out.options().dtype(torch::kFloat32));
-  int n_chunks = chunk_u.numel();
-  int64_t total = (int64_t)n_chunks * dim;
-  if (total == 0) return grad_unique;
-  auto stream = current_stream();
-  const float* wptr =
-      weights.defined() && weights.numel() ? weights.data_ptr<float>()
-                                           : nullptr;
-  if (grad_out.scalar_type() == torch::kBFloat16) {
-    k_group_pooled_bwd_chunked<__hip_bfloat16>
-        <<<n_blocks(total), kBlock, 0, stream>>>(
-            reinterpret_cast<const __hip_bfloat16*>(
-                grad_out.data_ptr<at::BFloat16>()),
-            order.data_ptr<int32_t>(), bounds.data_ptr<int32_t>(),
-            row_ids.data_ptr<int32_t>(), wptr, row_coeff.data_ptr<float>(),
-            chunk_u.data_ptr<int32_t>(), chunk_k0.data_ptr<int32_t>(),
-            n_chunks, (int)chunk_size, (int)batch, (int)n_tables, (int)dim,
-            grad_unique.data_ptr<float>());
-  } else {
-    k_group_pooled_bwd_chunked<float>
-        <<<n_blocks(total), kBlock, 0, stream>>>(
-            grad_out.data_ptr<float>(), order.data_ptr<int32_t>(),
-            bounds.data_ptr<int32_t>(), row_ids.data_ptr<int32_t>(), wptr,
-            row_coeff.data_ptr<float>(), chunk_u.data_ptr<int32_t>(),
-            chunk_k0.data_ptr<int32_t>(), n_chunks, (int)chunk_size,
-            (int)batch, (int)n_tables, (int)dim,
-            grad_unique.data_ptr<float>());
   }
   return grad_unique;
 }
@@ -1604,8 +1450,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("pooled_bwd", &pooled_bwd);
   mod.def("group_pooled_fwd", &group_pooled_fwd);
   mod.def("group_pooled_fwd_direct", &group_pooled_fwd_direct);
-  mod.def("group_pooled_bwd", &group_pooled_bwd);
-  mod.def("group_pooled_bwd_chunked", &group_pooled_bwd_chunked);
   mod.def("group_pooled_bwd_strided", &group_pooled_bwd_strided);
   mod.def("apply_sgd", &apply_sgd);
   mod.def("apply_adagrad", &apply_adagrad);
