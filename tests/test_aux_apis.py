@@ -190,3 +190,65 @@ def test_csv_dataset(tmp_path):
     assert b["C1"].dtype == torch.int64
     # same string -> same id
     assert int(batches[0]["C1"][0]) == int(batches[1]["C1"][0])
+
+
+def test_sequence_feature_column():
+    from deeprec_amd import feature_column as fc
+    col = fc.embedding_column(
+        fc.sequence_categorical_column_with_embedding("hist"), dimension=4)
+    layer = fc.InputLayer([col])
+    feats = {"hist": torch.randint(1, 100, (6, 5))}  # [B, T]
+    out = layer(feats)
+    assert out.shape == (6, 5 * 4)  # flattened sequence embeddings
+
+
+def test_memory_stats_hook_noop_on_cpu():
+    from deeprec_amd.training.session import MemoryStatsHook
+    h = MemoryStatsHook(every_n_steps=1)
+    h.after_run(None, None)  # must not raise without a GPU
+
+
+def test_prefetch_close():
+    from deeprec_amd.data.prefetch import PrefetchIterator
+    from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    ds = CriteoSyntheticDataset(batch_size=4, seed=2, matrix_format=True)
+    it = PrefetchIterator(ds, depth=2)
+    next(iter(it))
+    it.close()
+
+
+def test_safe_embedding_lookup_negative_ids():
+    from deeprec_amd import (EmbeddingVariable, RaggedIds,
+                             safe_embedding_lookup_sparse)
+    ev = EmbeddingVariable("safe_ev", 4)
+    sp = RaggedIds.from_lists([[1, -1, 2], [-5], [3]])
+    out = safe_embedding_lookup_sparse(ev, sp, combiner="sum")
+    assert out.shape == (3, 4)
+    # row with only a negative id pools to zeros
+    assert torch.equal(out[1], torch.zeros(4))
+    # negative ids never enter the table
+    keys, *_ = ev.export()
+    assert (keys >= 0).all()
+
+
+def test_fused_gru_cpu_reference_trains():
+    from deeprec_amd.ops.fused_gru import FusedGRU
+    torch.manual_seed(0)
+    gru = FusedGRU(8, 16)
+    x = torch.randn(4, 6, 8, requires_grad=True)
+    alpha = torch.rand(4, 6)
+    out = gru(x, alpha)
+    assert out.shape == (4, 6, 16)
+    out.sum().backward()
+    assert x.grad is not None
+    assert gru.weight_hh_l0.grad is not None
+    # matches torch GRU when alpha is None and weights are shared
+    ref = torch.nn.GRU(8, 16, batch_first=True)
+    with torch.no_grad():
+        ref.weight_ih_l0.copy_(gru.weight_ih_l0)
+        ref.weight_hh_l0.copy_(gru.weight_hh_l0)
+        ref.bias_ih_l0.copy_(gru.bias_ih_l0)
+        ref.bias_hh_l0.copy_(gru.bias_hh_l0)
+    out2 = gru(x.detach(), None)
+    out_ref, _ = ref(x.detach())
+    torch.testing.assert_close(out2, out_ref, rtol=1e-4, atol=1e-5)
