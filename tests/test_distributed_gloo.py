@@ -208,3 +208,62 @@ def _body_dlrm_sharded_collection_step(rank, world):
 
 def test_dlrm_sharded_collection():
     _run_dist(_body_dlrm_sharded_collection_step, world_size=2, port=29532)
+
+
+def _body_sharded_collection_w4(rank, world):
+    """world_size=4 shard-merge equivalence (closer to the 8-GPU shape)."""
+    from deeprec_amd import EmbeddingVariableOption
+    from deeprec_amd.embedding.collection import EmbeddingCollection
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.parallel.sharded_collection import (
+        ShardedEmbeddingCollection)
+
+    def init(t):
+        gen = torch.Generator().manual_seed(9)
+        t.normal_(0, 1, generator=gen)
+
+    opt_ev = EmbeddingVariableOption(
+        init_option=InitializerOption(initializer=init, default_value_dim=4))
+    sc = ShardedEmbeddingCollection("sc4", ["a", "b", "c"], 4,
+                                    ev_option=opt_ev)
+    ref = EmbeddingCollection("ref4", ["a", "b", "c"], 4, ev_option=opt_ev)
+    o_s = AdagradOptimizer(embedding_variables=[sc], learning_rate=0.1)
+    o_r = AdagradOptimizer(embedding_variables=[ref], learning_rate=0.1)
+    for step in range(2):
+        g = torch.Generator().manual_seed(800 + step)
+        ids_all = torch.randint(0, 60, (4 * world, 3), generator=g)
+        out = sc.lookup_matrix(ids_all[rank * 4:(rank + 1) * 4])
+        (out ** 2).sum().backward()
+        o_s.step()
+        (ref.lookup_matrix(ids_all) ** 2).sum().backward()
+        o_r.step()
+    tabs_s, tabs_r = sc.export_tables(), ref.export_tables()
+    for name in tabs_s:
+        ks, vs, _, _ = tabs_s[name]
+        kr, vr, _, _ = tabs_r[name]
+        gk = [None] * world
+        gv = [None] * world
+        dist.all_gather_object(gk, ks)
+        dist.all_gather_object(gv, vs)
+        ka, va = torch.cat(gk), torch.cat(gv)
+        oi, ri = torch.argsort(ka), torch.argsort(kr)
+        torch.testing.assert_close(ka[oi], kr[ri])
+        torch.testing.assert_close(va[oi], vr[ri], rtol=1e-4, atol=1e-5)
+
+
+def test_sharded_collection_world4():
+    _run_dist(_body_sharded_collection_w4, world_size=4, port=29541)
+
+
+def _body_work_queue_dist(rank, world):
+    from deeprec_amd.data.parquet import WorkQueue
+    wq = WorkQueue([f"f{i}" for i in range(5)])
+    first = wq.take()
+    assert first == f"f{rank}"
+    second = wq.take()
+    expected = rank + world
+    assert second == (f"f{expected}" if expected < 5 else None)
+
+
+def test_work_queue_distributed():
+    _run_dist(_body_work_queue_dist, world_size=2, port=29542)
