@@ -108,8 +108,8 @@ def main():
     if use_graph:
       try:
         st = model.collection.storage
-        st.enable_graph_mode(expected_entries=1 << 22,
-                             expected_slots=1 << 22)
+        st.enable_graph_mode(expected_entries=1 << 23,
+                             expected_slots=1 << 23)
         # optimizer slabs must exist before capture
         st.get_slab("adam_m", model.collection.dim, 0.0)
         st.get_slab("adam_v", model.collection.dim, 0.0)

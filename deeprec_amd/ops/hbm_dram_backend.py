@@ -87,6 +87,11 @@ class HbmDramStorage(HbmStorage):
         return t.view(rows, width)
 
     # ---------------- tier plumbing ----------------
+    def enable_graph_mode(self, *a, **kw):
+        raise NotImplementedError(
+            "hipGraph capture is not supported with the HBM_DRAM tier "
+            "(cold-row initialization requires host work per step)")
+
     def _init_limit(self) -> int:
         return self.hot_rows
 

@@ -30,8 +30,8 @@ class GraphedTrainStep:
     def __init__(self, model, optimizer, loss_fn: Callable,
                  example_batch: Sequence[torch.Tensor],
                  warmup_steps: int = 2,
-                 expected_entries: int = 1 << 22,
-                 expected_slots: int = 1 << 22):
+                 expected_entries: int = 1 << 23,
+                 expected_slots: int = 1 << 23):
         self.model = model
         self.optimizer = optimizer
         self.loss_fn = loss_fn

@@ -20,8 +20,8 @@ def _make(seed, graph_mode):
                            learning_rate=0.001, graph_safe=graph_mode)
     if graph_mode:
         st = m.collection.storage
-        st.enable_graph_mode(expected_entries=1 << 17,
-                             expected_slots=1 << 17)
+        st.enable_graph_mode(expected_entries=1 << 18,
+                             expected_slots=1 << 18)
         st.get_slab("adam_m", m.collection.dim, 0.0)
         st.get_slab("adam_v", m.collection.dim, 0.0)
         m.collection.graph_mode = True
