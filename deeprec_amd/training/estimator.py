@@ -90,7 +90,11 @@ class Estimator:
 
     @torch.no_grad()
     def evaluate(self, input_fn: Callable, steps: int = 10) -> dict:
+        """loss / accuracy / streaming AUC over `steps` eval batches
+        (reference: modelzoo train.py eval loops report loss+acc+auc)."""
+        from deeprec_amd.training.metrics import StreamingAUC
         it = iter(input_fn())
+        auc = StreamingAUC()
         losses, n_correct, n_total = [], 0, 0
         for _ in range(steps):
             *features, labels = next(it)
@@ -98,11 +102,14 @@ class Estimator:
             if isinstance(logits, (list, tuple)):
                 logits = logits[0]
             losses.append(float(self.model.loss_fn(logits, labels)))
-            preds = (torch.sigmoid(logits) > 0.5).float()
+            probs = torch.sigmoid(logits)
+            auc.update(probs, labels)
+            preds = (probs > 0.5).float()
             n_correct += int((preds == labels).sum())
             n_total += labels.numel()
         return {"loss": sum(losses) / len(losses),
-                "accuracy": n_correct / max(n_total, 1)}
+                "accuracy": n_correct / max(n_total, 1),
+                "auc": auc.result()}
 
     @torch.no_grad()
     def predict(self, input_fn: Callable):
