@@ -14,9 +14,14 @@
 // CDNA4 ISA (A: row=lane%16, k=8*(lane/16)+i; B: col=lane%16, same k;
 // D: col=lane%16, row=4*(lane/16)+i) — verified on-device against a torch
 // fp32 reference (tests/test_gpu_dense.py, asymmetric random inputs).
-// Weights are L2-resident at these sizes, so operands are read through L2
-// directly (no LDS staging): the ~512 KB weight is fetched once and A rows
-// stream. Each 256-thread block = 4 waves = a 64x16 output tile.
+// Memory strategy per kernel (all measured on the DLRM layer mix):
+//   fwd: operands straight from L2 (weights are L2-resident, A rows
+//        stream); 64x64 block tiles maximize block count.
+//   dX:  W columns are strided -> LDS-staged transposed W tile for the
+//        wide-N layers (k_linear_dx_lds), direct for narrow ones.
+//   dW:  both operands are column reads -> transposed LDS chunks + multi-
+//        chunk register accumulation to bound the fp32 atomic flush
+//        volume (k_linear_dw_lds64t); 16-col variant for narrow N.
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -95,51 +100,62 @@ __device__ __forceinline__ bf16x8 load_frag_col(const short* __restrict__ src,
 }
 
 // ------------------------------------------------------------------
-// fwd: C[M,N] = act(A[M,K] @ W[N,K]^T + bias); 4 waves/block stacked on M,
-// each wave computes 16 rows x 64 cols (4 n-tiles) so the A fragment is
-// loaded once per 4 MFMAs.
+// fwd: C[M,N] = act(A[M,K] @ W[N,K]^T + bias); 4 waves/block stacked on M.
+// MT = m-subtiles per wave: each wave computes MT*16 rows x 64 cols, so
+// one set of 4 B fragments feeds 4*MT MFMAs. MT=1 maximizes block count
+// (latency hiding via parallelism); MT=2 halves B-fragment loads per
+// MFMA. (A 128-col NT=8 variant measured WORSE — 44us vs 29us on
+// 8192x512x480 — fewer blocks cost more than the halved A re-reads.)
 // ------------------------------------------------------------------
-__global__ void k_linear_fwd(const short* __restrict__ A,
-                             const short* __restrict__ W,
-                             const float* __restrict__ bias, int M, int N,
-                             int K, int act, short* __restrict__ C) {
+template <int MT>
+__global__ void k_linear_fwd_t(const short* __restrict__ A,
+                               const short* __restrict__ W,
+                               const float* __restrict__ bias, int M, int N,
+                               int K, int act, short* __restrict__ C) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int ntiles64 = (N + 63) / 64;
-  const int m0 = (blockIdx.x / ntiles64) * 64 + wave * 16;
+  const int m0 = (blockIdx.x / ntiles64) * 64 * MT + wave * 16 * MT;
   const int n0 = (blockIdx.x % ntiles64) * 64;
   if (m0 >= M) return;
-  const int row_a = m0 + (lane & 15);
   const int col_b = n0 + (lane & 15);
   const int kgrp = (lane >> 4) * 8;
-  f32x4 acc[4] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f},
-                  {0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  f32x4 acc[MT][4] = {};
   const int nt = min(4, (N - n0 + 15) / 16);
   for (int k0 = 0; k0 < K; k0 += 32) {
-    bf16x8 a = load_frag_row(A, row_a, k0 + kgrp, M, K);
+    bf16x8 a[MT];
+#pragma unroll
+    for (int r = 0; r < MT; ++r)
+      a[r] = load_frag_row(A, m0 + r * 16 + (lane & 15), k0 + kgrp, M, K);
 #pragma unroll
     for (int t = 0; t < 4; ++t) {
       if (t >= nt) break;
       bf16x8 b = load_frag_row(W, col_b + t * 16, k0 + kgrp, N, K);
-      acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[t], 0, 0,
-                                                       0);
+#pragma unroll
+      for (int r = 0; r < MT; ++r)
+        acc[r][t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[r], b,
+                                                            acc[r][t], 0, 0,
+                                                            0);
     }
   }
   // D: col = lane%16, row = 4*(lane/16) + i
 #pragma unroll
-  for (int t = 0; t < 4; ++t) {
-    if (t >= nt) break;
-    const int cn = n0 + t * 16 + (lane & 15);
-    if (cn >= N) continue;
-    const float bv = bias ? bias[cn] : 0.0f;
+  for (int r = 0; r < MT; ++r) {
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int cm = m0 + (lane >> 4) * 4 + i;
-      if (cm >= M) continue;
-      float v = acc[t][i] + bv;
-      if (act == 1 && v < 0.0f) v = 0.0f;          // relu
-      else if (act == 2) v = 1.0f / (1.0f + __expf(-v));  // sigmoid
-      C[(int64_t)cm * N + cn] = f2bf_u16(v);
+    for (int t = 0; t < 4; ++t) {
+      if (t >= nt) break;
+      const int cn = n0 + t * 16 + (lane & 15);
+      if (cn >= N) continue;
+      const float bv = bias ? bias[cn] : 0.0f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int cm = m0 + r * 16 + (lane >> 4) * 4 + i;
+        if (cm >= M) continue;
+        float v = acc[r][t][i] + bv;
+        if (act == 1 && v < 0.0f) v = 0.0f;          // relu
+        else if (act == 2) v = 1.0f / (1.0f + __expf(-v));  // sigmoid
+        C[(int64_t)cm * N + cn] = f2bf_u16(v);
+      }
     }
   }
 }
@@ -187,71 +203,78 @@ __global__ void k_linear_dx(const short* __restrict__ G,
 }
 
 // ------------------------------------------------------------------
-// dW[N,K] += Gc^T @ Ac over an M chunk; fp32 atomic accumulate.
-// grid.x = tiles_n * tiles_k * m_chunks; fused dbias column sums
-// (k-tile 0 only). dW/dbias must be zeroed by the caller.
+// dX via LDS-staged weights: the 64-col variant above builds its B
+// fragments from 8 strided global loads each (W columns); for wide N
+// that dominates (measured 37.5us on the 8192x512x480 layer). Here
+// W[:, c0:c0+64] is staged TRANSPOSED into LDS once per 256-row N
+// superchunk (row stride 264 shorts keeps the 16B fragment reads
+// aligned), and each wave computes 32 rows x 64 cols (8 MFMAs per pair
+// of G fragments).
 // ------------------------------------------------------------------
-__global__ void k_linear_dw(const short* __restrict__ G,
-                            const short* __restrict__ A, int M, int N, int K,
-                            int chunk_rows, float* __restrict__ dW,
-                            float* __restrict__ dbias) {
+constexpr int DX_NSUP = 256;  // N rows staged per superchunk (33 KB LDS)
+__global__ void k_linear_dx_lds(const short* __restrict__ G,
+                                const short* __restrict__ W, int M, int N,
+                                int K, short* __restrict__ dX) {
+  __shared__ short wl[64][DX_NSUP + 8];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int tiles_k = (K + 15) / 16;
-  const int tiles_n = (N + 15) / 16;
-  const int tile_id = blockIdx.x % (tiles_n * tiles_k);
-  const int chunk = blockIdx.x / (tiles_n * tiles_k);
-  const int n0 = (tile_id / tiles_k) * 16;
-  const int k0 = (tile_id % tiles_k) * 16;
-  // 4 waves split the chunk's rows
-  // NOTE: no early return — the fused-bias path below has a __syncthreads
-  // that every wave must reach; out-of-range waves just loop zero times
-  // (fragment loads are bounds-checked / zero-filled).
-  const int rows_per_wave = chunk_rows / 4;
-  const int mbeg = chunk * chunk_rows + wave * rows_per_wave;
-  const int mend = min(mbeg + rows_per_wave, M);
-
-  const int row_gt = n0 + (lane & 15);  // output row (N dim)
-  const int col_a = k0 + (lane & 15);
+  const int ktiles64 = (K + 63) / 64;
+  const int m0 = (blockIdx.x / ktiles64) * 128 + wave * 32;
+  const int c0 = (blockIdx.x % ktiles64) * 64;
+  const int row_g0 = m0 + (lane & 15);
+  const int row_g1 = m0 + 16 + (lane & 15);
   const int kgrp = (lane >> 4) * 8;
-  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  float bsum = 0.0f;
-  const bool do_bias = (dbias != nullptr) && (k0 == 0);
-  for (int m = mbeg; m < mend; m += 32) {
-    // A-operand = G^T: frag elem i = G[m + kgrp + i][row_gt]
-    bf16x8 a = load_frag_col(G, row_gt, m + kgrp, M, N);
-    bf16x8 b = load_frag_col(A, col_a, m + kgrp, M, K);
-    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
-    if (do_bias) {
+  const int kt = min(4, (K - c0 + 15) / 16);
+  f32x4 acc[2][4] = {};
+  for (int ns = 0; ns < N; ns += DX_NSUP) {
+    const int nlen = min(DX_NSUP, N - ns);
+    __syncthreads();  // previous superchunk's reads must finish
+    {
+      // stage W[ns..ns+nlen][c0..c0+64] -> wl[c][n]; rows past N are
+      // zero-filled by load_frag_row so fragment tails read zeros
+      const int r = threadIdx.x >> 3;
+      const int seg = threadIdx.x & 7;
+      const int nstg = (nlen + 31) & ~31;
+      for (int it = 0; it * 32 < nstg; ++it) {
+        int n = it * 32 + r;
+        bf16x8 v = load_frag_row(W, ns + n, c0 + seg * 8, N, K);
 #pragma unroll
-      for (int i = 0; i < 8; ++i) bsum += bf2f_u16(a[i]);
+        for (int j = 0; j < 8; ++j) wl[seg * 8 + j][n] = v[j];
+      }
     }
-  }
-  const int ck = k0 + (lane & 15);
-  if (ck < K) {
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      int cn = n0 + (lane >> 4) * 4 + i;
-      if (cn >= N) continue;
-      atomicAdd(&dW[(int64_t)cn * K + ck], acc[i]);
-    }
-  }
-  if (do_bias) {
-    // bsum holds this lane's partial over its G column (row_gt); lanes with
-    // the same (lane&15) share the column across kgrp groups — reduce via
-    // LDS per block then atomicAdd once per column per block.
-    __shared__ float bred[4][16];
-    // lanes 0-15 collect column partial sums from lanes 16-63
-    float v = bsum;
-    for (int off = 16; off < 64; off += 16)
-      v += __shfl(bsum, (lane & 15) + off);
-    if (lane < 16) bred[wave][lane] = v;
     __syncthreads();
-    if (wave == 0 && lane < 16) {
-      float t = bred[0][lane] + bred[1][lane] + bred[2][lane] +
-                bred[3][lane];
-      int cn = n0 + lane;
-      if (cn < N) atomicAdd(&dbias[cn], t);
+    if (m0 >= M) continue;
+    for (int n0 = 0; n0 < nlen; n0 += 32) {
+      bf16x8 a0 = load_frag_row(G, row_g0, ns + n0 + kgrp, M, N);
+      bf16x8 a1 = load_frag_row(G, row_g1, ns + n0 + kgrp, M, N);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        if (t >= kt) break;
+        bf16x8 b = *reinterpret_cast<const bf16x8*>(
+            &wl[t * 16 + (lane & 15)][n0 + kgrp]);
+        acc[0][t] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, b, acc[0][t], 0, 0,
+                                                    0);
+        acc[1][t] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, b, acc[1][t], 0, 0,
+                                                    0);
+      }
+    }
+  }
+  if (m0 >= M) return;
+#pragma unroll
+  for (int rg = 0; rg < 2; ++rg) {
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      if (t >= kt) break;
+      const int ck = c0 + t * 16 + (lane & 15);
+      if (ck >= K) continue;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        int cm = m0 + rg * 16 + (lane >> 4) * 4 + i;
+        if (cm >= M) continue;
+        dX[(int64_t)cm * K + ck] = f2bf_u16(acc[rg][t][i]);
+      }
     }
   }
 }
@@ -343,6 +366,172 @@ __global__ void k_linear_dw_lds(const short* __restrict__ G,
       int cn = n0 + lane;
       if (cn < N) atomicAdd(&dbias[cn], v);
     }
+  }
+}
+
+// 64-col-G variant: block = 64n x 64k x 128m. The 16-col version above
+// restages the A chunk once per 16-row N tile (N=512 -> 32x = 252 MB of
+// A traffic on the 8192x512x480 layer); staging G and A as 64-wide tiles
+// cuts that to N/64 restages (63 MB) + K/64 G restages. Wave w owns
+// output rows n0+16w..+16, all 64 k columns.
+__global__ void k_linear_dw_lds64(const short* __restrict__ G,
+                                  const short* __restrict__ A, int M, int N,
+                                  int K, float* __restrict__ dW,
+                                  float* __restrict__ dbias) {
+  constexpr int CH = 128;
+  __shared__ short a_l[CH][64 + 8];
+  __shared__ short g_l[CH][64 + 8];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  const int kgroups = (K + 63) / 64;
+  const int ngroups = (N + 63) / 64;
+  const int tile_id = blockIdx.x % (ngroups * kgroups);
+  const int chunk = blockIdx.x / (ngroups * kgroups);
+  const int n0 = (tile_id / kgroups) * 64;
+  const int k0 = (tile_id % kgroups) * 64;
+  const int mbeg = chunk * CH;
+  {
+    const int r = tid >> 3, seg = tid & 7;
+    for (int it = 0; it < CH / 32; ++it) {
+      int rr = it * 32 + r;
+      *reinterpret_cast<bf16x8*>(&a_l[rr][seg * 8]) =
+          load_frag_row(A, mbeg + rr, k0 + seg * 8, M, K);
+      *reinterpret_cast<bf16x8*>(&g_l[rr][seg * 8]) =
+          load_frag_row(G, mbeg + rr, n0 + seg * 8, M, N);
+    }
+  }
+  __syncthreads();
+  const int col_g = wave * 16 + (lane & 15);
+  const int kgrp = (lane >> 4) * 8;
+  f32x4 acc[4] = {};
+  float bsum = 0.0f;
+  const bool do_bias = (dbias != nullptr) && (k0 == 0);
+#pragma unroll
+  for (int ms = 0; ms < CH; ms += 32) {
+    bf16x8 gfrag;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) gfrag[i] = g_l[ms + kgrp + i][col_g];
+    if (do_bias) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i) bsum += bf2f_u16(gfrag[i]);
+    }
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      bf16x8 afrag;
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        afrag[i] = a_l[ms + kgrp + i][t * 16 + (lane & 15)];
+      acc[t] =
+          __builtin_amdgcn_mfma_f32_16x16x32_bf16(gfrag, afrag, acc[t], 0,
+                                                  0, 0);
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    const int ck = k0 + t * 16 + (lane & 15);
+    if (ck >= K) continue;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int cn = n0 + wave * 16 + (lane >> 4) * 4 + i;
+      if (cn >= N) continue;
+      atomicAdd(&dW[(int64_t)cn * K + ck], acc[t][i]);
+    }
+  }
+  if (do_bias) {
+    // each wave owns 16 distinct bias columns (n0 + 16w + lane%16):
+    // reduce the 4 kgrp groups' partials within the wave, one atomic each
+    float v = bsum;
+    for (int off = 16; off < 64; off += 16)
+      v += __shfl(bsum, (lane & 15) + off);
+    const int cn = n0 + wave * 16 + (lane & 15);
+    if (lane < 16 && cn < N) atomicAdd(&dbias[cn], v);
+  }
+}
+
+// Transposed-LDS dW: same 64n x 64k tile as lds64, but (a) A and G
+// chunks are stored TRANSPOSED ([col][row], row stride 136 shorts keeps
+// 16B alignment) so each MFMA fragment is ONE aligned b128 LDS read, and
+// (b) each block accumulates `cpb` consecutive 128-row M chunks in
+// registers before its single atomic flush. The per-chunk atomic flush
+// was the real bound: 64 chunks x N*K fp32 atomics = 63 MB of atomic
+// writes on the 8192x512x480 layer (~31 us of the 60 us).
+__global__ void k_linear_dw_lds64t(const short* __restrict__ G,
+                                   const short* __restrict__ A, int M, int N,
+                                   int K, int cpb,
+                                   float* __restrict__ dW,
+                                   float* __restrict__ dbias) {
+  constexpr int CH = 128;
+  __shared__ short a_t[64][CH + 8];
+  __shared__ short g_t[64][CH + 8];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tid = threadIdx.x;
+  const int kgroups = (K + 63) / 64;
+  const int ngroups = (N + 63) / 64;
+  const int tile_id = blockIdx.x % (ngroups * kgroups);
+  const int chunk0 = (blockIdx.x / (ngroups * kgroups)) * cpb;
+  const int n0 = (tile_id / kgroups) * 64;
+  const int k0 = (tile_id % kgroups) * 64;
+  const int col_g = wave * 16 + (lane & 15);
+  const int kgrp = (lane >> 4) * 8;
+  f32x4 acc[4] = {};
+  float bsum = 0.0f;
+  const bool do_bias = (dbias != nullptr) && (k0 == 0);
+  for (int c = 0; c < cpb; ++c) {
+    const int mbeg = (chunk0 + c) * CH;
+    if (mbeg >= M) break;
+    if (c > 0) __syncthreads();  // previous chunk's reads must finish
+    {
+      // coalesced global row loads, transposing scalar stores into LDS
+      const int r = tid >> 3, seg = tid & 7;
+      for (int it = 0; it < CH / 32; ++it) {
+        int rr = it * 32 + r;
+        bf16x8 va = load_frag_row(A, mbeg + rr, k0 + seg * 8, M, K);
+        bf16x8 vg = load_frag_row(G, mbeg + rr, n0 + seg * 8, M, N);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          a_t[seg * 8 + j][rr] = va[j];
+          g_t[seg * 8 + j][rr] = vg[j];
+        }
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int ms = 0; ms < CH; ms += 32) {
+      bf16x8 gfrag =
+          *reinterpret_cast<const bf16x8*>(&g_t[col_g][ms + kgrp]);
+      if (do_bias) {
+#pragma unroll
+        for (int i = 0; i < 8; ++i) bsum += bf2f_u16(gfrag[i]);
+      }
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+            &a_t[t * 16 + (lane & 15)][ms + kgrp]);
+        acc[t] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(gfrag, afrag, acc[t],
+                                                    0, 0, 0);
+      }
+    }
+  }
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    const int ck = k0 + t * 16 + (lane & 15);
+    if (ck >= K) continue;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int cn = n0 + wave * 16 + (lane >> 4) * 4 + i;
+      if (cn >= N) continue;
+      atomicAdd(&dW[(int64_t)cn * K + ck], acc[t][i]);
+    }
+  }
+  if (do_bias) {
+    float v = bsum;
+    for (int off = 16; off < 64; off += 16)
+      v += __shfl(bsum, (lane & 15) + off);
+    const int cn = n0 + wave * 16 + (lane & 15);
+    if (lane < 16 && cn < N) atomicAdd(&dbias[cn], v);
   }
 }
 
@@ -491,16 +680,24 @@ static short* bf_ptr_mut(torch::Tensor& t) {
 }
 
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w_bf16,
-                         torch::Tensor bias, int64_t act) {
+                         torch::Tensor bias, int64_t act,
+                         int64_t variant) {
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
   int M = x.size(0), K = x.size(1), N = w_bf16.size(0);
   auto out = torch::empty({M, N}, x.options());
   int tiles_n = (N + 63) / 64;
-  int blocks = ((M + 63) / 64) * tiles_n;
-  k_linear_fwd<<<blocks, 256, 0, dense_stream()>>>(
-      bf_ptr(x), bf_ptr(w_bf16),
-      bias.defined() && bias.numel() ? bias.data_ptr<float>() : nullptr, M,
-      N, K, (int)act, bf_ptr_mut(out));
+  const float* bp =
+      bias.defined() && bias.numel() ? bias.data_ptr<float>() : nullptr;
+  if (variant < 0) variant = 0;  // MT=1 measured fastest across the zoo
+  if (variant == 1) {
+    int blocks = ((M + 127) / 128) * tiles_n;
+    k_linear_fwd_t<2><<<blocks, 256, 0, dense_stream()>>>(
+        bf_ptr(x), bf_ptr(w_bf16), bp, M, N, K, (int)act, bf_ptr_mut(out));
+  } else {
+    int blocks = ((M + 63) / 64) * tiles_n;
+    k_linear_fwd_t<1><<<blocks, 256, 0, dense_stream()>>>(
+        bf_ptr(x), bf_ptr(w_bf16), bp, M, N, K, (int)act, bf_ptr_mut(out));
+  }
   return out;
 }
 
@@ -508,26 +705,50 @@ torch::Tensor linear_dx(torch::Tensor g, torch::Tensor w_bf16) {
   int M = g.size(0), N = g.size(1), K = w_bf16.size(1);
   auto dx = torch::empty({M, K}, g.options());
   int tiles_k = (K + 63) / 64;
-  int blocks = ((M + 63) / 64) * tiles_k;
-  k_linear_dx<<<blocks, 256, 0, dense_stream()>>>(
-      bf_ptr(g), bf_ptr(w_bf16), M, N, K, bf_ptr_mut(dx));
+  if (N >= 128) {
+    // wide GEMM-K: strided W-column loads dominate -> LDS-staged variant
+    int blocks = ((M + 127) / 128) * tiles_k;
+    k_linear_dx_lds<<<blocks, 256, 0, dense_stream()>>>(
+        bf_ptr(g), bf_ptr(w_bf16), M, N, K, bf_ptr_mut(dx));
+  } else {
+    int blocks = ((M + 63) / 64) * tiles_k;
+    k_linear_dx<<<blocks, 256, 0, dense_stream()>>>(
+        bf_ptr(g), bf_ptr(w_bf16), M, N, K, bf_ptr_mut(dx));
+  }
   return dx;
 }
 
 std::tuple<torch::Tensor, torch::Tensor> linear_dw(torch::Tensor g,
                                                    torch::Tensor x,
-                                                   bool want_bias) {
+                                                   bool want_bias,
+                                                   int64_t variant) {
   int M = g.size(0), N = g.size(1), K = x.size(1);
   // one zero-fill for dW + dbias (fill launches were ~9% of the step)
   auto ws = torch::zeros({(int64_t)N * K + (want_bias ? N : 0)},
                          g.options().dtype(torch::kFloat32));
   auto dw = ws.narrow(0, 0, (int64_t)N * K).view({N, K});
   auto db = want_bias ? ws.narrow(0, (int64_t)N * K, N) : torch::Tensor();
-  int tiles = ((N + 15) / 16) * ((K + 63) / 64);
   int m_chunks = (M + 127) / 128;
-  k_linear_dw_lds<<<tiles * m_chunks, 256, 0, dense_stream()>>>(
-      bf_ptr(g), bf_ptr(x), M, N, K, dw.data_ptr<float>(),
-      want_bias ? db.data_ptr<float>() : nullptr);
+  float* dbp = want_bias ? db.data_ptr<float>() : nullptr;
+  if (variant < 0) variant = N >= 64 ? 2 : 0;
+  if (variant == 2) {
+    // 64-wide tiles, transposed LDS, multi-chunk register accumulation:
+    // pick chunks-per-block so the grid stays ~1024 blocks while the
+    // atomic flush volume drops by cpb x
+    int tiles = ((N + 63) / 64) * ((K + 63) / 64);
+    int cpb = std::max(1, (int)((int64_t)m_chunks * tiles / 512));
+    int chunk_blocks = (m_chunks + cpb - 1) / cpb;
+    k_linear_dw_lds64t<<<tiles * chunk_blocks, 256, 0, dense_stream()>>>(
+        bf_ptr(g), bf_ptr(x), M, N, K, cpb, dw.data_ptr<float>(), dbp);
+  } else if (variant == 1) {
+    int tiles = ((N + 63) / 64) * ((K + 63) / 64);
+    k_linear_dw_lds64<<<tiles * m_chunks, 256, 0, dense_stream()>>>(
+        bf_ptr(g), bf_ptr(x), M, N, K, dw.data_ptr<float>(), dbp);
+  } else {
+    int tiles = ((N + 15) / 16) * ((K + 63) / 64);
+    k_linear_dw_lds<<<tiles * m_chunks, 256, 0, dense_stream()>>>(
+        bf_ptr(g), bf_ptr(x), M, N, K, dw.data_ptr<float>(), dbp);
+  }
   return {dw, db};
 }
 
@@ -571,8 +792,10 @@ torch::Tensor interact_bwd(torch::Tensor grad, torch::Tensor feats) {
 void register_dense(py::module_& mod) {
   mod.def("interact_fwd", &interact_fwd);
   mod.def("interact_bwd", &interact_bwd);
-  mod.def("linear_fwd", &linear_fwd);
+  mod.def("linear_fwd", &linear_fwd, py::arg("x"), py::arg("w"),
+          py::arg("bias"), py::arg("act"), py::arg("variant") = -1);
   mod.def("linear_dx", &linear_dx);
-  mod.def("linear_dw", &linear_dw);
+  mod.def("linear_dw", &linear_dw, py::arg("g"), py::arg("x"),
+          py::arg("want_bias"), py::arg("variant") = -1);
   mod.def("act_bwd", &act_bwd);
 }
