@@ -286,7 +286,7 @@ class EmbeddingCollection:
             nnz = row_ids_cat.numel()
             # under capture m is nnz-padded, so the ratio test is wrong
             # there; captured workloads are the zipf ones -> split
-            splits = 8 if (self.graph_mode or nnz > 2 * m) else 1
+            splits = 32 if (self.graph_mode or nnz > 2 * m) else 1
             return self.storage.ext.group_pooled_bwd_strided(
                 grad_out.contiguous(), order, bounds, row_ids_cat,
                 weights_cat if weights_cat is not None else torch.Tensor(),

@@ -96,7 +96,7 @@ def _chunked_ev_backward(ev, grad_out, inverse, offsets, row_ids, counts,
                                 1.0 / denom.clamp(min=1e-12),
                                 torch.zeros_like(lengths))
     nnz = inverse.numel()
-    splits = 8 if nnz > 2 * m else 1
+    splits = 32 if nnz > 2 * m else 1
     return ext.group_pooled_bwd_strided(
         grad_out.contiguous(), order, bounds, row_ids.to(torch.int32),
         weights.float() if weights is not None else torch.Tensor(),

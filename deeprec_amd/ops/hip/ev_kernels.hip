@@ -1241,7 +1241,7 @@ torch::Tensor group_pooled_fwd_direct(
   return out;
 }
 
-template <typename T, int SPL>
+template <typename T, int SPL, int CHUNK = 128>
 static void launch_bwd_strided(const T* gp, torch::Tensor& order,
                                torch::Tensor& bounds, torch::Tensor& row_ids,
                                const float* wptr, torch::Tensor& row_coeff,
@@ -1249,8 +1249,9 @@ static void launch_bwd_strided(const T* gp, torch::Tensor& order,
                                int64_t n_tables, int64_t dim,
                                bool identity_rows, torch::Tensor& out,
                                hipStream_t stream) {
-  constexpr int CHUNK = 128;
-  int64_t total = m * dim * SPL;
+  // grid-stride kernel: cap the grid so an nnz-padded m (graph capture)
+  // costs no extra block launches
+  int64_t total = std::min<int64_t>(m * dim * SPL, (int64_t)4096 * kBlock);
   if (identity_rows) {
     k_group_pooled_bwd_strided<T, SPL, CHUNK, true>
         <<<n_blocks(total), kBlock, 0, stream>>>(
@@ -1292,8 +1293,20 @@ torch::Tensor group_pooled_bwd_strided(
       launch_bwd_strided<__hip_bfloat16, 1>(
           gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
           n_tables, dim, identity_rows, grad_unique, stream);
-    else
+    else if (splits <= 8)
       launch_bwd_strided<__hip_bfloat16, 8>(
+          gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
+          n_tables, dim, identity_rows, grad_unique, stream);
+    else if (splits <= 16)
+      launch_bwd_strided<__hip_bfloat16, 16, 64>(
+          gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
+          n_tables, dim, identity_rows, grad_unique, stream);
+    else if (splits <= 32)
+      launch_bwd_strided<__hip_bfloat16, 32, 32>(
+          gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
+          n_tables, dim, identity_rows, grad_unique, stream);
+    else
+      launch_bwd_strided<__hip_bfloat16, 64, 16>(
           gp, order, bounds, row_ids, wptr, row_coeff, m, mdp, batch,
           n_tables, dim, identity_rows, grad_unique, stream);
   } else {
