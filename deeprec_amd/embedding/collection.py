@@ -165,6 +165,9 @@ class EmbeddingCollection:
     def _dedup_and_probe(self, values_cat, train):
         """unique + single hash probe. GPU training uses the fused hash
         dedup (no sorts); other paths use torch.unique + the probe."""
+        # a rank buffer from an earlier un-consumed dedup must never pair
+        # with this call's inverse (the sort path orders keys differently)
+        self.storage._last_rank = None
         if self.graph_mode and train:
             return self.storage.dedup_lookup_capture(values_cat)
         if (train and hasattr(self.storage, "dedup_lookup")
@@ -191,9 +194,15 @@ class EmbeddingCollection:
         if self.device.type != "cuda":
             order = torch.argsort(inverse.long()).to(torch.int32)
             return order, bounds, None, None
-        # sort-free CSR build (per-key cursors); the strided backward
-        # needs no host-built chunk arrays
-        order = self.storage.ext.csr_order(inverse, bounds, m)
+        # sort-free CSR build; when the fused dedup ran, pass C already
+        # recorded each occurrence's rank within its key, so order is a
+        # direct scatter (no second atomic-cursor pass over nnz)
+        rank = getattr(self.storage, "_last_rank", None)
+        self.storage._last_rank = None
+        if rank is not None and rank.numel() == inverse.numel():
+            order = self.storage.ext.csr_scatter(inverse, rank, bounds)
+        else:  # torch.unique (sort) path
+            order = self.storage.ext.csr_order(inverse, bounds, m)
         return order, bounds, None, None
 
     def lookup_matrix(self, ids: torch.Tensor, out_dtype=None,

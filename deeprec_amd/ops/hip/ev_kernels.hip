@@ -367,7 +367,8 @@ __global__ void k_dedup_pass_c(
     const int64_t* __restrict__ keys, int nnz,
     const int64_t* __restrict__ ht_keys,
     const int32_t* __restrict__ ht_compact, int64_t cap_mask,
-    int32_t* __restrict__ inverse, int32_t* __restrict__ counts) {
+    int32_t* __restrict__ inverse, int32_t* __restrict__ counts,
+    int32_t* __restrict__ rank) {
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; j < nnz; j += stride) {
@@ -378,11 +379,25 @@ __global__ void k_dedup_pass_c(
       if (ht_keys[idx] == key) {
         int c = ht_compact[idx];
         inverse[j] = c;
-        atomicAdd(&counts[c], 1);
+        // the counter value doubles as this occurrence's rank within its
+        // key, letting the CSR order build become a direct scatter
+        // (k_csr_scatter) instead of a second atomic-cursor pass
+        rank[j] = atomicAdd(&counts[c], 1);
         break;
       }
     }
   }
+}
+
+// order[bounds[c] + rank[j]] = j — direct scatter using pass C's ranks.
+__global__ void k_csr_scatter(const int32_t* __restrict__ inverse,
+                              const int32_t* __restrict__ rank, int nnz,
+                              const int32_t* __restrict__ bounds,
+                              int32_t* __restrict__ order) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; j < nnz; j += stride)
+    order[bounds[inverse[j]] + rank[j]] = (int32_t)j;
 }
 
 // CSR order build: order[bounds[c] + pos++] = j (pos via per-key cursor).
@@ -994,19 +1009,34 @@ torch::Tensor ht_dedup_b_padded(
   return slots;
 }
 
-std::tuple<torch::Tensor, torch::Tensor> ht_dedup_c(
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ht_dedup_c(
     torch::Tensor keys, torch::Tensor ht_keys, torch::Tensor ht_compact,
     int64_t m) {
   int64_t nnz = keys.numel();
   auto inverse = torch::empty({nnz}, ht_compact.options());
-  auto counts = torch::zeros({m}, ht_compact.options());
-  if (nnz == 0) return {inverse, counts};
+  auto rank = torch::empty({nnz}, ht_compact.options());
+  auto counts = torch::empty({m}, ht_compact.options());
+  if (nnz == 0) return {inverse, counts, rank};
   auto stream = current_stream();
+  hipMemsetAsync(counts.data_ptr<int32_t>(), 0, m * sizeof(int32_t),
+                 stream);
   k_dedup_pass_c<<<n_blocks(nnz), kBlock, 0, stream>>>(
       keys.data_ptr<int64_t>(), (int)nnz, ht_keys.data_ptr<int64_t>(),
       ht_compact.data_ptr<int32_t>(), ht_keys.numel() - 1,
-      inverse.data_ptr<int32_t>(), counts.data_ptr<int32_t>());
-  return {inverse, counts};
+      inverse.data_ptr<int32_t>(), counts.data_ptr<int32_t>(),
+      rank.data_ptr<int32_t>());
+  return {inverse, counts, rank};
+}
+
+torch::Tensor csr_scatter(torch::Tensor inverse, torch::Tensor rank,
+                          torch::Tensor bounds) {
+  int64_t nnz = inverse.numel();
+  auto order = torch::empty({nnz}, inverse.options());
+  if (nnz == 0) return order;
+  k_csr_scatter<<<n_blocks(nnz), kBlock, 0, current_stream()>>>(
+      inverse.data_ptr<int32_t>(), rank.data_ptr<int32_t>(), (int)nnz,
+      bounds.data_ptr<int32_t>(), order.data_ptr<int32_t>());
+  return order;
 }
 
 torch::Tensor csr_order(torch::Tensor inverse, torch::Tensor bounds,
@@ -1277,10 +1307,12 @@ torch::Tensor group_pooled_bwd_strided(
   // splits: 8 for duplication-heavy (zipf) batches — hot keys fan out
   // across concurrent accumulators; 1 for mostly-unique batches (long
   // sequences) where extra splits are 8x wasted bounds probes.
-  auto grad_unique = torch::zeros(
+  auto grad_unique = torch::empty(
       {m, dim}, grad_out.options().dtype(torch::kFloat32));
   if (m * dim == 0) return grad_unique;
   auto stream = current_stream();
+  hipMemsetAsync(grad_unique.data_ptr<float>(), 0,
+                 (size_t)m * dim * sizeof(float), stream);
   const float* wptr =
       weights.defined() && weights.numel() ? weights.data_ptr<float>()
                                            : nullptr;
@@ -1455,6 +1487,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("bump_epoch", &bump_epoch);
   mod.def("ht_dedup_c", &ht_dedup_c);
   mod.def("csr_order", &csr_order);
+  mod.def("csr_scatter", &csr_scatter);
   mod.def("ht_insert_bulk", &ht_insert_bulk);
   mod.def("ht_lookup", &ht_lookup);
   mod.def("ht_export", &ht_export);

@@ -234,8 +234,9 @@ class HbmStorage:
             self.slot_counter, self.max_slots, self.values,
             self.default_values, self.dvd_per_table, self.key_bits,
             self._init_limit(), self.filter_freq, self.error_flag)
-        inverse, counts = self.ext.ht_dedup_c(values_cat, self.ht_keys,
-                                              self.ht_compact, nnz)
+        inverse, counts, rank = self.ext.ht_dedup_c(
+            values_cat, self.ht_keys, self.ht_compact, nnz)
+        self._last_rank = rank  # consumed by collection._prep_backward
         return uniq_buf, inverse, counts, slots
 
     def prefers_dedup(self) -> bool:
@@ -275,8 +276,9 @@ class HbmStorage:
             self.slot_counter, self.max_slots, self.values,
             self.default_values, self.dvd_per_table, self.key_bits,
             self._init_limit(), self.filter_freq, self.error_flag)
-        inverse, counts = self.ext.ht_dedup_c(values_cat, self.ht_keys,
-                                              self.ht_compact, m)
+        inverse, counts, rank = self.ext.ht_dedup_c(
+            values_cat, self.ht_keys, self.ht_compact, m)
+        self._last_rank = rank  # consumed by collection._prep_backward
         return uniq, inverse, counts, slots
 
     def _use_no_permission(self) -> bool:
