@@ -535,6 +535,17 @@ __global__ void k_linear_dw_lds64t(const short* __restrict__ G,
   }
 }
 
+// zero-fill (see ev_kernels k_zero_f32: memset nodes don't replay in
+// captured graphs)
+__global__ void k_zero_f32d(float* __restrict__ p, int64_t n) {
+  int64_t i = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 4;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x * 4;
+  for (; i + 3 < n; i += stride)
+    *reinterpret_cast<float4*>(p + i) = make_float4(0.f, 0.f, 0.f, 0.f);
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t j = n & ~3LL; j < n; ++j) p[j] = 0.0f;
+}
+
 // activation backward: G = dY * act_grad(out); act 1=relu, 2=sigmoid
 __global__ void k_act_bwd(const short* __restrict__ dY,
                           const short* __restrict__ out, int64_t n, int act,
@@ -723,12 +734,13 @@ std::tuple<torch::Tensor, torch::Tensor> linear_dw(torch::Tensor g,
                                                    bool want_bias,
                                                    int64_t variant) {
   int M = g.size(0), N = g.size(1), K = x.size(1);
-  // one async memset for dW + dbias (torch fill kernels cost ~4.7us of
-  // launch+elementwise overhead each; the DMA memset is ~1us)
+  // one fill-kernel zero for dW + dbias (cheaper than torch's fill
+  // machinery; hipMemsetAsync is avoided: memset nodes recorded during
+  // hipGraph capture were observed not to replay)
   int64_t ws_len = (int64_t)N * K + (want_bias ? N : 0);
   auto ws = torch::empty({ws_len}, g.options().dtype(torch::kFloat32));
-  hipMemsetAsync(ws.data_ptr<float>(), 0, ws_len * sizeof(float),
-                 dense_stream());
+  k_zero_f32d<<<(int)std::min<int64_t>((ws_len / 4 + 255) / 256, 1024),
+                256, 0, dense_stream()>>>(ws.data_ptr<float>(), ws_len);
   auto dw = ws.narrow(0, 0, (int64_t)N * K).view({N, K});
   auto db = want_bias ? ws.narrow(0, (int64_t)N * K, N) : torch::Tensor();
   int m_chunks = (M + 127) / 128;

@@ -362,6 +362,18 @@ __global__ void k_dedup_pass_b_padded(
   out_slots[c] = slot;
 }
 
+// Zero-fill (float4-wide). Used instead of hipMemsetAsync because memset
+// nodes recorded during hipGraph capture were observed NOT to replay
+// (counts/grad buffers accumulated across replays -> OOB scatters).
+__global__ void k_zero_f32(float* __restrict__ p, int64_t n) {
+  int64_t i = (blockIdx.x * (int64_t)blockDim.x + threadIdx.x) * 4;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x * 4;
+  for (; i + 3 < n; i += stride)
+    *reinterpret_cast<float4*>(p + i) = make_float4(0.f, 0.f, 0.f, 0.f);
+  if (blockIdx.x == 0 && threadIdx.x == 0)
+    for (int64_t j = n & ~3LL; j < n; ++j) p[j] = 0.0f;
+}
+
 // Pass C (per occurrence): inverse + per-batch counts.
 __global__ void k_dedup_pass_c(
     const int64_t* __restrict__ keys, int nnz,
@@ -1018,8 +1030,8 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ht_dedup_c(
   auto counts = torch::empty({m}, ht_compact.options());
   if (nnz == 0) return {inverse, counts, rank};
   auto stream = current_stream();
-  hipMemsetAsync(counts.data_ptr<int32_t>(), 0, m * sizeof(int32_t),
-                 stream);
+  k_zero_f32<<<n_blocks((m + 3) / 4), kBlock, 0, stream>>>(
+      reinterpret_cast<float*>(counts.data_ptr<int32_t>()), m);
   k_dedup_pass_c<<<n_blocks(nnz), kBlock, 0, stream>>>(
       keys.data_ptr<int64_t>(), (int)nnz, ht_keys.data_ptr<int64_t>(),
       ht_compact.data_ptr<int32_t>(), ht_keys.numel() - 1,
@@ -1311,8 +1323,8 @@ torch::Tensor group_pooled_bwd_strided(
       {m, dim}, grad_out.options().dtype(torch::kFloat32));
   if (m * dim == 0) return grad_unique;
   auto stream = current_stream();
-  hipMemsetAsync(grad_unique.data_ptr<float>(), 0,
-                 (size_t)m * dim * sizeof(float), stream);
+  k_zero_f32<<<n_blocks((m * dim + 3) / 4), kBlock, 0, stream>>>(
+      grad_unique.data_ptr<float>(), m * dim);
   const float* wptr =
       weights.defined() && weights.numel() ? weights.data_ptr<float>()
                                            : nullptr;

@@ -84,3 +84,38 @@ def test_graph_replay_matches_eager():
         # eager/replay trajectories diverge at bf16 noise level
         torch.testing.assert_close(ve.cpu()[oe_i], vg.cpu()[og_i],
                                    rtol=1e-2, atol=3e-3)
+
+
+@pytest.mark.gpu
+def test_graph_replay_many_fresh_batches():
+    """Regression: 10+ replays over FRESH id batches. Catches per-replay
+    re-zeroing bugs (hipMemsetAsync nodes recorded during capture were
+    observed NOT to replay -> counts/grad accumulation -> OOB scatter)."""
+    from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    from deeprec_amd.embedding.options import EmbeddingVariableOption
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.optimizers import AdamAsyncOptimizer
+    from deeprec_amd.training.graph_step import GraphedTrainStep
+
+    torch.manual_seed(0)
+    dev = torch.device("cuda")
+    m = DLRM(device=dev, bf16=True,
+             ev_option=EmbeddingVariableOption(init_capacity=1 << 20))
+    ds = CriteoSyntheticDataset(batch_size=4096, seed=11, device=dev,
+                                matrix_format=True)
+    opt = AdamAsyncOptimizer(params=m.parameters(),
+                             embedding_variables=m.embedding_variables(),
+                             graph_safe=True)
+
+    def loss_fn(model, dense, ids, labels):
+        return model.loss_fn(model(dense, ids), labels)
+
+    step = GraphedTrainStep(m, opt, loss_fn, ds.next_batch(),
+                            expected_entries=1 << 21,
+                            expected_slots=1 << 21)
+    assert step.graph is not None, "capture must succeed"
+    for _ in range(12):
+        loss = step(ds.next_batch())
+    torch.cuda.synchronize()
+    m.collection.storage._check_error()
+    assert torch.isfinite(loss)

@@ -81,6 +81,18 @@ def bench_model(name, steps, warmup, device, use_graph=True):
     if device.type == "cuda":
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
+    # tear the captured graph down NOW: a dozen live hipGraphs destroyed
+    # in interpreter-exit order crash the runtime (observed memory fault
+    # at teardown), and per-model cleanup also releases the 8M-slot EVs
+    if use_graph:
+        gstep.graph = None
+        del gstep
+    del m, opt, ds, batches
+    import gc
+    gc.collect()
+    if device.type == "cuda":
+        torch.cuda.synchronize()
+        torch.cuda.empty_cache()
     return batch, steps * batch / dt, dt / steps * 1000
 
 
@@ -109,13 +121,15 @@ def main():
             rows.append((name, 0, 0.0, 0.0))
         from deeprec_amd.embedding.variable import reset_registry
         reset_registry()
+        if args.out:  # write incrementally: a late crash keeps results
+            with open(args.out, "w") as f:
+                f.write(f"# Model-zoo throughput ({device})\n\n")
+                f.write("| model | batch | samples/sec | ms/step |\n")
+                f.write("|---|---|---|---|\n")
+                for name_, batch_, sps_, ms_ in rows:
+                    f.write(f"| {name_} | {batch_} | {sps_:.0f} "
+                            f"| {ms_:.3f} |\n")
     if args.out:
-        with open(args.out, "w") as f:
-            f.write(f"# Model-zoo throughput ({device})\n\n")
-            f.write("| model | batch | samples/sec | ms/step |\n")
-            f.write("|---|---|---|---|\n")
-            for name, batch, sps, ms in rows:
-                f.write(f"| {name} | {batch} | {sps:.0f} | {ms:.3f} |\n")
         print("wrote", args.out)
 
 
