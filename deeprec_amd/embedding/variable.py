@@ -123,6 +123,12 @@ class EmbeddingVariable:
         return self.storage.shrink(
             step if step is not None else get_global_step())
 
+    def rebalance(self) -> int:
+        """LFU hot/cold repack for multi-tier storages; 0 elsewhere."""
+        if hasattr(self.storage, "rebalance"):
+            return self.storage.rebalance()
+        return 0
+
     def start_sparse_recording(self):
         self._record_sparse_ids = True
 

@@ -271,6 +271,58 @@ class HbmDramStorage(HbmStorage):
                      if slab_rows else None)
         return n_evicted
 
+    def rebalance(self) -> int:
+        """LFU promotion: repack so the hottest keys (by the engine's
+        per-entry frequency counters) occupy the HBM tier and the long
+        tail lives in DRAM/SSD (reference capability: CacheStrategy /
+        LFUCache promotion, multi_tier_storage.h). Host-coordinated and
+        off the hot path — run it periodically (RebalanceHook), like
+        shrink/compaction. Returns the number of rows that changed tier."""
+        names = list(self.slabs.keys())
+        keys, slots, freqs, versions = self._export_entries()
+        adm = slots >= 0
+        n = int(adm.sum())
+        if n == 0:
+            return 0
+        f_adm = freqs[adm]
+        s_adm = slots[adm]
+        order = torch.argsort(f_adm.long(), descending=True, stable=True)
+        ranks = torch.empty(n, dtype=torch.int32, device=self.device)
+        ranks[order] = torch.arange(n, dtype=torch.int32,
+                                    device=self.device)
+        moved = int(((s_adm < self.hot_rows)
+                     != (ranks < self.hot_rows)).sum())
+        if moved == 0:
+            return 0
+        # materialize current rows (old placement), then rewrite in rank
+        # order: ranks [0, hot_rows) -> HBM, the rest -> cold
+        values = self.materialize(keys[adm], s_adm)
+        slab_rows = dict(zip(names, self.export_slabs(names))) \
+            if names else {}
+        vals_ranked = values[order]
+        hot_n = min(n, self.hot_rows)
+        self.values[:hot_n] = vals_ranked[:hot_n]
+        if n > hot_n:
+            self._grow_slots(n)
+            self.values_cold[: n - hot_n] = vals_ranked[hot_n:].cpu()
+        for name, rows in slab_rows.items():
+            rows_ranked = rows[order.cpu()]
+            self.slabs[name][:hot_n] = rows_ranked[:hot_n].to(self.device)
+            if n > hot_n:
+                self.cold_slabs[name][: n - hot_n] = rows_ranked[hot_n:]
+        new_slots = torch.full_like(slots, -1)
+        new_slots[adm] = ranks
+        self._alloc_table(self.capacity)
+        self.entry_counter.zero_()
+        self.slot_counter.fill_(n)
+        self.ext.ht_insert_bulk(keys, new_slots, freqs, versions,
+                                self.ht_keys, self.ht_slot, self.ht_freq,
+                                self.ht_version, self.entry_counter,
+                                self.error_flag)
+        self._check_error()
+        self._sync_counters()
+        return moved
+
     def export(self, include_filtered: bool = False):
         keys, slots, freqs, versions = self._export_entries()
         adm = slots >= 0

@@ -217,6 +217,28 @@ class MonitoredTrainingSession:
         return results
 
 
+class RebalanceHook(SessionRunHook):
+    """Every `every_steps`, repack multi-tier EmbeddingVariables so the
+    hottest keys (by engine frequency counters) sit in the HBM tier
+    (reference: CacheStrategy-driven promotion in multi_tier_storage)."""
+
+    def __init__(self, every_steps: int = 10000, variables=None):
+        self.every = max(1, every_steps)
+        self.variables = variables
+        self._n = 0
+
+    def after_run(self, results):
+        self._n += 1
+        if self._n % self.every:
+            return
+        from deeprec_amd.embedding.variable import all_embedding_variables
+        evs = self.variables if self.variables is not None             else all_embedding_variables()
+        moved = sum(ev.rebalance() for ev in evs
+                    if hasattr(ev, "rebalance"))
+        if moved:
+            log.info("rebalance: %d rows changed tier", moved)
+
+
 class MemoryStatsHook(SessionRunHook):
     """Log device memory accounting every N steps (the reference's
     GPU-memory-optimization observability: allocator stats vs peak)."""

@@ -106,3 +106,52 @@ def test_multitier_ssd_path(tmp_path):
     # values persisted through the mmap round-trip
     out1 = embedding_lookup(ev, torch.arange(500, device=DEV))
     assert torch.isfinite(out1).all()
+
+
+@pytest.mark.gpu
+def test_lfu_rebalance_promotes_hot_keys():
+    """After rebalance(), the highest-frequency keys must occupy the HBM
+    tier (slot < hot_rows) and all values must survive the repack."""
+    from deeprec_amd.embedding.options import (EmbeddingVariableOption,
+                                               StorageOption, StorageType)
+    from deeprec_amd.ops.hbm_dram_backend import HbmDramStorage
+
+    torch.manual_seed(0)
+    dim = 8
+    hot_rows = 32
+    opt = EmbeddingVariableOption(storage_option=StorageOption(
+        storage_type=StorageType.HBM_DRAM,
+        storage_size=[hot_rows * dim * 4]))
+    st = HbmDramStorage(dim, opt, device="cuda")
+    n = 96
+    keys = torch.arange(n, dtype=torch.int64, device="cuda")
+    # slot order WITHIN one batched insert is race-ordered, so force the
+    # misplacement deterministically: insert 64 keys first (they own the
+    # whole 32-row hot tier + 32 cold), then the 32 soon-to-be-hot keys
+    # (guaranteed cold slots 64..95)
+    st.lookup_or_create(keys[:64],
+                        torch.ones(64, dtype=torch.int32, device="cuda"),
+                        step=0)
+    st.lookup_or_create(keys[64:],
+                        torch.ones(32, dtype=torch.int32, device="cuda"),
+                        step=0)
+    slots0 = st.lookup(keys)
+    assert bool((slots0[64:] >= hot_rows).all())
+    before = st.materialize(keys, slots0).cpu()
+    # make the LAST 32 keys the hottest (high counts -> high freq)
+    hotkeys = keys[64:]
+    for _ in range(5):
+        st.lookup_or_create(hotkeys,
+                            torch.full((32,), 50, dtype=torch.int32,
+                                       device="cuda"), step=1)
+    moved = st.rebalance()
+    assert moved == 64  # 32 promoted + 32 demoted
+    slots1 = st.lookup(keys)
+    # hottest keys now in the HBM tier
+    assert bool((slots1[64:] < hot_rows).all())
+    # values survive the repack exactly
+    after = st.materialize(keys, slots1).cpu()
+    torch.testing.assert_close(before, after)
+    # frequencies preserved too
+    f = st.frequencies(keys)
+    assert int(f[64:].min()) > int(f[:64].max())
