@@ -77,7 +77,8 @@ class DLRM(nn.Module):
                 f"{name_prefix}/sparse",
                 [f"C{i+1}" for i in range(num_sparse)], embedding_dim,
                 ev_option=ev_option, combiners=["mean"] * num_sparse,
-                device=self.device_)
+                device=self.device_,
+                comm_dtype=torch.bfloat16 if self.bf16 else None)
             self.evs = []
         elif use_collection:
             from deeprec_amd.embedding.collection import EmbeddingCollection

@@ -33,9 +33,14 @@ class ShardedEmbeddingCollection:
                  embedding_dim: int,
                  ev_option: Optional[EmbeddingVariableOption] = None,
                  combiners=None, device=None, value_dtype=torch.float32,
-                 generator=None, trainable: bool = True):
+                 generator=None, trainable: bool = True,
+                 comm_dtype: Optional[torch.dtype] = None):
         self.world = comm.world_size()
         self.rank = comm.rank()
+        # comm_dtype=torch.bfloat16 halves the xGMI bytes of the row-return
+        # all-to-all; lossless end-to-end for bf16 models (the pooled rows
+        # are cast to bf16 for the MLP anyway). Gradients stay fp32.
+        self.comm_dtype = comm_dtype
         self.local = EmbeddingCollection(
             f"{name}/part_{self.rank}", table_names, embedding_dim,
             ev_option, combiners, device, value_dtype, generator, trainable)
@@ -164,7 +169,14 @@ def _exchange_lookup(sev: ShardedEmbeddingCollection, uniq, counts, train):
                                            train=train)
     emb2 = coll.storage.gather(uniq2, slots2)        # [m2, D] fp32
     emb_out = emb2[inv2]                              # [n_recv, D]
-    emb_back = comm.all_to_all_single(emb_out.contiguous(), out_sp, in_sp)
+    if sev.comm_dtype is not None:
+        # reduced-precision row exchange; int16 view keeps the transport
+        # dtype universally supported (gloo included)
+        payload = emb_out.to(sev.comm_dtype).view(torch.int16).contiguous()
+        emb_back = comm.all_to_all_single(payload, out_sp, in_sp)             .view(sev.comm_dtype).float()
+    else:
+        emb_back = comm.all_to_all_single(emb_out.contiguous(), out_sp,
+                                          in_sp)
     emb = torch.empty_like(emb_back)
     emb[order_o] = emb_back                           # uniq order
     ctx = (order_o, inv2, slots2, uniq2, in_sp, out_sp)

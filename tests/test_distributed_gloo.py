@@ -267,3 +267,27 @@ def _body_work_queue_dist(rank, world):
 
 def test_work_queue_distributed():
     _run_dist(_body_work_queue_dist, world_size=2, port=29542)
+
+
+def _body_bf16_exchange(rank, world):
+    from deeprec_amd.parallel.sharded_collection import (
+        ShardedEmbeddingCollection)
+
+    torch.manual_seed(7)
+    names = [f"t{i}" for i in range(4)]
+    ids = torch.randint(0, 50, (16, 4))
+    g1 = torch.Generator().manual_seed(5)
+    g2 = torch.Generator().manual_seed(5)
+    sev32 = ShardedEmbeddingCollection("x32", names, 8, generator=g1)
+    sev16 = ShardedEmbeddingCollection("x16", names, 8, generator=g2,
+                                       comm_dtype=torch.bfloat16)
+    out32 = sev32.lookup_matrix(ids, train=True)
+    out16 = sev16.lookup_matrix(ids, train=True)
+    # identical math up to the bf16 transport quantization of the rows
+    torch.testing.assert_close(out16, out32, rtol=1e-2, atol=1e-2)
+    out16.sum().backward()
+    dist.barrier()
+
+
+def test_bf16_row_exchange_world2():
+    _run_dist(_body_bf16_exchange, world_size=2, port=29543)
