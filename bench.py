@@ -84,6 +84,8 @@ def main():
                                           broadcast_parameters)
         broadcast_parameters(model.parameters())
         reducer = DenseGradAllreducer(model.parameters())
+        # overlap the dense all-reduce with the sparse applies
+        opt.pre_dense_step = reducer.wait
     else:
         reducer = None
 
@@ -100,7 +102,7 @@ def main():
         opt.zero_grad()
         loss.backward()
         if reducer is not None:
-            reducer.allreduce()
+            reducer.allreduce(async_op=True)
         opt.step()
         return loss
 

@@ -33,6 +33,9 @@ class Optimizer:
         self.evs = list(embedding_variables or [])
         params = [p for p in (params or []) if p.requires_grad]
         self._dense = self._make_dense(params) if params else None
+        # distributed seam: set to DenseGradAllreducer.wait to overlap an
+        # async dense-grad all-reduce with the sparse applies
+        self.pre_dense_step = None
         self._step_count = 0
 
     # -- subclass hooks --
@@ -61,8 +64,8 @@ class Optimizer:
 
     def step(self, increment_global_step: bool = True):
         self._step_count += 1
-        if self._dense is not None:
-            self._dense.step()
+        # sparse applies first: they are independent of the dense grads,
+        # so an in-flight async all-reduce overlaps with them
         for ev in self.evs:
             if not ev.trainable:
                 ev.consume_grads()
@@ -70,6 +73,10 @@ class Optimizer:
             hyper = self._sparse_hyper(ev)
             for slots, keys, grad in ev.consume_grads():
                 self._apply_sparse(ev, slots, grad, hyper)
+        if self.pre_dense_step is not None:
+            self.pre_dense_step()
+        if self._dense is not None:
+            self._dense.step()
         self._post_step()
         if increment_global_step:
             GLOBAL_STEP.increment()

@@ -49,8 +49,13 @@ class DenseGradAllreducer:
         self.params: List[torch.Tensor] = [
             p for p in params if p.requires_grad]
         self.bucket_cap = int(bucket_cap_mb * (1 << 20))
+        self._pending = []
 
-    def allreduce(self):
+    def allreduce(self, async_op: bool = False):
+        """Reduce all dense grads. With async_op=True the collectives are
+        launched without waiting — call wait() (or set it as the
+        optimizer's pre_dense_step) before the dense optimizer step, so
+        the all-reduce overlaps with the sparse applies."""
         if not is_initialized() or world_size() == 1:
             return
         w = world_size()
@@ -61,15 +66,28 @@ class DenseGradAllreducer:
             bucket.append(p.grad)
             nbytes += p.grad.numel() * p.grad.element_size()
             if nbytes >= self.bucket_cap:
-                self._reduce_bucket(bucket, w)
+                self._reduce_bucket(bucket, w, async_op)
                 bucket, nbytes = [], 0
         if bucket:
-            self._reduce_bucket(bucket, w)
+            self._reduce_bucket(bucket, w, async_op)
 
-    @staticmethod
-    def _reduce_bucket(grads, w):
+    def wait(self):
+        for handle, flat, grads, w in self._pending:
+            if handle is not None:
+                handle.wait()
+            flat.div_(w)
+            for g, r in zip(
+                    grads, torch._utils._unflatten_dense_tensors(flat,
+                                                                 grads)):
+                g.copy_(r)
+        self._pending = []
+
+    def _reduce_bucket(self, grads, w, async_op=False):
         flat = torch._utils._flatten_dense_tensors(grads)
-        dist.all_reduce(flat)
+        handle = dist.all_reduce(flat, async_op=async_op)
+        if async_op:
+            self._pending.append((handle, flat, grads, w))
+            return
         flat.div_(w)
         for g, r in zip(grads,
                         torch._utils._unflatten_dense_tensors(flat, grads)):

@@ -291,3 +291,22 @@ def _body_bf16_exchange(rank, world):
 
 def test_bf16_row_exchange_world2():
     _run_dist(_body_bf16_exchange, world_size=2, port=29543)
+
+
+def _body_async_allreduce(rank, world):
+    from deeprec_amd.parallel import DenseGradAllreducer
+
+    torch.manual_seed(rank)
+    p = torch.nn.Parameter(torch.randn(1000))
+    p.grad = torch.full((1000,), float(rank + 1))
+    red = DenseGradAllreducer([p])
+    red.allreduce(async_op=True)
+    red.wait()
+    # mean of (1, 2) = 1.5 for world=2
+    expect = sum(range(1, world + 1)) / world
+    assert torch.allclose(p.grad, torch.full((1000,), expect))
+    dist.barrier()
+
+
+def test_async_allreduce_world2():
+    _run_dist(_body_async_allreduce, world_size=2, port=29544)
