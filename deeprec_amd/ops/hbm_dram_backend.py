@@ -284,9 +284,14 @@ class HbmDramStorage(HbmStorage):
         n = int(adm.sum())
         if n == 0:
             return 0
-        f_adm = freqs[adm]
+        from deeprec_amd.embedding.options import CacheStrategy
+        lru = (self.ev_option.storage_option is not None and
+               self.ev_option.storage_option.cache_strategy
+               == CacheStrategy.LRU)
+        # LFU ranks by frequency counters, LRU by last-touched step
+        score = versions[adm] if lru else freqs[adm].long()
         s_adm = slots[adm]
-        order = torch.argsort(f_adm.long(), descending=True, stable=True)
+        order = torch.argsort(score, descending=True, stable=True)
         ranks = torch.empty(n, dtype=torch.int32, device=self.device)
         ranks[order] = torch.arange(n, dtype=torch.int32,
                                     device=self.device)

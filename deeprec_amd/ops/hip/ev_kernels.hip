@@ -867,17 +867,22 @@ __global__ void k_apply_rmsprop(float* __restrict__ w, float* __restrict__ vel,
   APPLY_EPILOG
 }
 
+// l2_shrinkage != 0 is the FtrlV2 variant (reference:
+// KvResourceSparseApplyFtrlV2): the linear term sees
+// g + 2*l2_shrinkage*w while the accumulator sees plain g^2.
 __global__ void k_apply_ftrl(float* __restrict__ w, float* __restrict__ n,
                              float* __restrict__ z,
                              const int32_t* __restrict__ slots,
                              const float* __restrict__ grad, int m, int dim,
-                             float lr, float l1, float l2, float lr_power) {
+                             float lr, float l1, float l2, float lr_power,
+                             float l2_shrinkage) {
   APPLY_PROLOG
   float wv = w[o];
   float n_old = n[o];
   float n_new = n_old + g * g;
   float sigma = (powf(n_new, -lr_power) - powf(n_old, -lr_power)) / lr;
-  float z_new = z[o] + g - sigma * wv;
+  float gs = g + 2.0f * l2_shrinkage * wv;
+  float z_new = z[o] + gs - sigma * wv;
   n[o] = n_new;
   z[o] = z_new;
   float quad = powf(n_new, -lr_power) / lr + 2.0f * l2;
@@ -1474,7 +1479,7 @@ void apply_rmsprop(torch::Tensor w, torch::Tensor vel, torch::Tensor slots,
 
 void apply_ftrl(torch::Tensor w, torch::Tensor n, torch::Tensor z,
                 torch::Tensor slots, torch::Tensor grad, double lr, double l1,
-                double l2, double lr_power) {
+                double l2, double lr_power, double l2_shrinkage) {
   int m = slots.numel();
   if (m == 0) return;
   int dim = w.size(1);
@@ -1482,7 +1487,7 @@ void apply_ftrl(torch::Tensor w, torch::Tensor n, torch::Tensor z,
   k_apply_ftrl<<<n_blocks((int64_t)m * dim), kBlock, 0, stream>>>(
       w.data_ptr<float>(), n.data_ptr<float>(), z.data_ptr<float>(),
       slots.data_ptr<int32_t>(), grad.data_ptr<float>(), m, dim, (float)lr,
-      (float)l1, (float)l2, (float)lr_power);
+      (float)l1, (float)l2, (float)lr_power, (float)l2_shrinkage);
 }
 
 void register_dense(pybind11::module_& mod);  // dense_kernels.hip

@@ -499,7 +499,8 @@ def sparse_apply(name: str, storage: HbmStorage, slots, grad, hyper: dict):
         n = storage.get_slab("ftrl_accum", d, 0.1)
         z = storage.get_slab("ftrl_linear", d, 0.0)
         ext.apply_ftrl(w, n, z, slots, grad, hyper["lr"], hyper["l1"],
-                       hyper["l2"], hyper["lr_power"])
+                       hyper["l2"], hyper["lr_power"],
+                       hyper.get("l2_shrinkage", 0.0))
     else:
         raise ValueError(f"unknown sparse optimizer {name!r}")
 

@@ -107,9 +107,9 @@ def apply_adamw(storage, slots, grad, lr, step_t, weight_decay=0.01,
 
 
 def apply_ftrl(storage, slots, grad, lr, l1=0.0, l2=0.0,
-               lr_power=-0.5, **_):
+               lr_power=-0.5, l2_shrinkage=0.0, **_):
     """FTRL-proximal (reference: KvSparseApplyFtrlOp,
-    training_ali_ops.cc:431)."""
+    training_ali_ops.cc:431); l2_shrinkage != 0 is the FtrlV2 variant."""
     s, g = _adm(storage, slots, grad)
     n = storage.get_slab("ftrl_accum", storage.dim, 0.1)
     z = storage.get_slab("ftrl_linear", storage.dim, 0.0)
@@ -117,7 +117,7 @@ def apply_ftrl(storage, slots, grad, lr, l1=0.0, l2=0.0,
     n_old = n[s]
     n_new = n_old + g * g
     sigma = (n_new.pow(-lr_power) - n_old.pow(-lr_power)) / lr
-    z_new = z[s] + g - sigma * w
+    z_new = z[s] + (g + 2.0 * l2_shrinkage * w) - sigma * w
     n[s], z[s] = n_new, z_new
     quad = n_new.pow(-lr_power) / lr + 2.0 * l2
     w_new = torch.where(

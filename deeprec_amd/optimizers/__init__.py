@@ -275,11 +275,13 @@ class FtrlOptimizer(Optimizer):
                  learning_rate=0.01, learning_rate_power=-0.5,
                  initial_accumulator_value=0.1,
                  l1_regularization_strength=0.0,
-                 l2_regularization_strength=0.0):
+                 l2_regularization_strength=0.0,
+                 l2_shrinkage_regularization_strength=0.0):
         self.lr_power = learning_rate_power
         self.initial_accumulator_value = initial_accumulator_value
         self.l1 = l1_regularization_strength
         self.l2 = l2_regularization_strength
+        self.l2_shrinkage = l2_shrinkage_regularization_strength
         super().__init__(params, embedding_variables, learning_rate)
 
     def _make_dense(self, params):
@@ -288,7 +290,7 @@ class FtrlOptimizer(Optimizer):
 
     def _sparse_hyper(self, ev):
         return dict(lr=self.lr, l1=self.l1, l2=self.l2,
-                    lr_power=self.lr_power)
+                    lr_power=self.lr_power, l2_shrinkage=self.l2_shrinkage)
 
 
 def make_optimizer(name: str, params=None, embedding_variables=None,

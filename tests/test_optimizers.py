@@ -116,3 +116,34 @@ def test_multiple_lookups_same_ev_per_step():
     opt.step()
     w1 = ev.gather(torch.tensor([7]))
     torch.testing.assert_close(w1, w0 - 2.0)  # two applies of grad 1
+
+
+def test_ftrl_v2_l2_shrinkage():
+    """FtrlV2: l2_shrinkage feeds the linear term but not the accumulator
+    (reference: KvResourceSparseApplyFtrlV2)."""
+    import torch
+    from deeprec_amd.embedding import EmbeddingVariable
+    from deeprec_amd.optimizers import FtrlOptimizer
+
+    torch.manual_seed(0)
+
+    def run(shrink):
+        from deeprec_amd.embedding.variable import reset_registry
+        reset_registry()
+        ev = EmbeddingVariable(f"ftrlv2_{shrink}", 4, device="cpu")
+        opt = FtrlOptimizer(embedding_variables=[ev], learning_rate=0.1,
+                            l2_shrinkage_regularization_strength=shrink)
+        ids = torch.tensor([1, 2, 3])
+        for _ in range(3):
+            from deeprec_amd.embedding import embedding_lookup
+            out = embedding_lookup(ev, ids, train=True)
+            loss = (out ** 2).sum()
+            loss.backward()
+            opt.step()
+        k, v, _, _ = ev.export()
+        return v[torch.argsort(k)]
+
+    v0 = run(0.0)
+    v1 = run(0.5)
+    assert not torch.allclose(v0, v1), \
+        "l2_shrinkage must change the trajectory"
