@@ -535,6 +535,56 @@ __global__ void k_linear_dw_lds64t(const short* __restrict__ G,
   }
 }
 
+// ------------------------------------------------------------------
+// Fused row L2-normalize: y = x / sqrt(max(sum(x^2), eps)); backward
+// dx = (dy - y * <y, dy>) * inv_norm. One wave per row, shfl reduction
+// (reference capability: FusedL2Normalize[Grad],
+// fused_l2_normalize_op.cc:236,493 — CPU AVX there, CDNA4 wave here).
+// ------------------------------------------------------------------
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off; off >>= 1) v += __shfl_down(v, off);
+  return __shfl(v, 0);
+}
+
+__global__ void k_l2norm_fwd(const short* __restrict__ X, int M, int N,
+                             float eps, short* __restrict__ Y,
+                             float* __restrict__ inv_norms) {
+  const int lane = threadIdx.x & 63;
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= M) return;
+  const short* x = X + (int64_t)row * N;
+  float ss = 0.0f;
+  for (int j = lane; j < N; j += 64) {
+    float v = bf2f_u16(x[j]);
+    ss += v * v;
+  }
+  float inv = rsqrtf(fmaxf(wave_sum(ss), eps));
+  short* y = Y + (int64_t)row * N;
+  for (int j = lane; j < N; j += 64)
+    y[j] = f2bf_u16(bf2f_u16(x[j]) * inv);
+  if (lane == 0) inv_norms[row] = inv;
+}
+
+__global__ void k_l2norm_bwd(const short* __restrict__ dY,
+                             const short* __restrict__ Y,
+                             const float* __restrict__ inv_norms, int M,
+                             int N, short* __restrict__ dX) {
+  const int lane = threadIdx.x & 63;
+  const int row = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (row >= M) return;
+  const short* dy = dY + (int64_t)row * N;
+  const short* y = Y + (int64_t)row * N;
+  float dot = 0.0f;
+  for (int j = lane; j < N; j += 64)
+    dot += bf2f_u16(y[j]) * bf2f_u16(dy[j]);
+  dot = wave_sum(dot);
+  const float inv = inv_norms[row];
+  short* dx = dX + (int64_t)row * N;
+  for (int j = lane; j < N; j += 64)
+    dx[j] = f2bf_u16((bf2f_u16(dy[j]) - bf2f_u16(y[j]) * dot) * inv);
+}
+
 // zero-fill (see ev_kernels k_zero_f32: memset nodes don't replay in
 // captured graphs)
 __global__ void k_zero_f32d(float* __restrict__ p, int64_t n) {
@@ -804,7 +854,30 @@ torch::Tensor interact_bwd(torch::Tensor grad, torch::Tensor feats) {
   return dfeats;
 }
 
+std::tuple<torch::Tensor, torch::Tensor> l2norm_fwd(torch::Tensor x,
+                                                    double eps) {
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+  int M = x.size(0), N = x.size(1);
+  auto y = torch::empty_like(x);
+  auto inv = torch::empty({M}, x.options().dtype(torch::kFloat32));
+  k_l2norm_fwd<<<(M + 3) / 4, 256, 0, dense_stream()>>>(
+      bf_ptr(x), M, N, (float)eps, bf_ptr_mut(y), inv.data_ptr<float>());
+  return {y, inv};
+}
+
+torch::Tensor l2norm_bwd(torch::Tensor dy, torch::Tensor y,
+                         torch::Tensor inv) {
+  int M = y.size(0), N = y.size(1);
+  auto dx = torch::empty_like(y);
+  k_l2norm_bwd<<<(M + 3) / 4, 256, 0, dense_stream()>>>(
+      bf_ptr(dy.contiguous()), bf_ptr(y), inv.data_ptr<float>(), M, N,
+      bf_ptr_mut(dx));
+  return dx;
+}
+
 void register_dense(py::module_& mod) {
+  mod.def("l2norm_fwd", &l2norm_fwd);
+  mod.def("l2norm_bwd", &l2norm_bwd);
   mod.def("interact_fwd", &interact_fwd);
   mod.def("interact_bwd", &interact_bwd);
   mod.def("linear_fwd", &linear_fwd, py::arg("x"), py::arg("w"),

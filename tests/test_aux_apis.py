@@ -252,3 +252,37 @@ def test_fused_gru_cpu_reference_trains():
     out2 = gru(x.detach(), None)
     out_ref, _ = ref(x.detach())
     torch.testing.assert_close(out2, out_ref, rtol=1e-4, atol=1e-5)
+
+
+def test_fused_l2_normalize_cpu():
+    from deeprec_amd.ops.fused_norm import fused_l2_normalize
+
+    x = torch.randn(10, 8)
+    y = fused_l2_normalize(x)
+    torch.testing.assert_close(
+        y, torch.nn.functional.normalize(x, dim=-1), rtol=1e-5, atol=1e-6)
+
+
+def test_kafka_dataset_offsets_and_resume(tmp_path):
+    from deeprec_amd.data.kafka import KafkaDataset
+
+    log = tmp_path / "clicks-0.log"
+    log.write_text("\n".join(f'{{"id": {i}}}' for i in range(10)) + "\n")
+    import json as _json
+    ds = KafkaDataset(["clicks:0:0"], servers=f"file://{tmp_path}",
+                      message_parser=_json.loads, batch_size=4)
+    batches = list(ds)
+    assert [m["id"] for b in batches for m in b] == list(range(10))
+
+    # resume from a checkpointed offset
+    ds2 = KafkaDataset(["clicks:0:0"], servers=f"file://{tmp_path}",
+                       message_parser=_json.loads, batch_size=4)
+    it = iter(ds2)
+    next(it)                      # consume 4
+    ck = tmp_path / "offsets.json"
+    ds2.save(str(ck))
+    ds3 = KafkaDataset(["clicks:0:0"], servers=f"file://{tmp_path}",
+                       message_parser=_json.loads, batch_size=100)
+    ds3.restore(str(ck))
+    rest = next(iter(ds3))
+    assert [m["id"] for m in rest] == list(range(4, 10))

@@ -177,3 +177,22 @@ def test_fused_gru_weight_grads():
     for pg, pc in zip(gru_g.parameters(), gru_c.parameters()):
         torch.testing.assert_close(pg.grad.cpu(), pc.grad, rtol=5e-2,
                                    atol=1e-1)
+
+
+@pytest.mark.gpu
+def test_fused_l2_normalize_matches_torch():
+    from deeprec_amd.ops.fused_norm import fused_l2_normalize
+
+    torch.manual_seed(0)
+    for m, n in [(64, 16), (257, 100), (1000, 64)]:
+        x = torch.randn(m, n, device="cuda", requires_grad=True)
+        y = fused_l2_normalize(x)
+        ref = torch.nn.functional.normalize(x.detach().double(),
+                                            dim=-1).float()
+        torch.testing.assert_close(y.float(), ref, rtol=1e-2, atol=1e-2)
+        g = torch.randn_like(y, dtype=torch.float32)
+        y.backward(g)
+        xr = x.detach().double().requires_grad_(True)
+        torch.nn.functional.normalize(xr, dim=-1).backward(g.double())
+        torch.testing.assert_close(x.grad.float(), xr.grad.float(),
+                                   rtol=2e-2, atol=2e-2)

@@ -125,3 +125,35 @@ def test_batcher_propagates_errors():
             fut.result(timeout=10)
     finally:
         batcher.close()
+
+
+def test_feature_store_publish_and_lookup(tmp_path):
+    from deeprec_amd.serving.feature_store import (
+        FileFeatureStore, LocalFeatureStore, publish_checkpoint,
+        store_backed_lookup)
+
+    model = _make_ckpt(tmp_path)
+    # exported truth from the live model
+    tables = model.collection.export_tables()
+    store = LocalFeatureStore()
+    ckpt = sorted(tmp_path.glob("ckpt-*"))[0]
+    n = publish_checkpoint(store, str(ckpt))
+    assert n > 0 and len(store.tables()) > 0
+
+    name = store.tables()[0]
+    # find the matching exported table rows
+    for tname, (k, v, f, ver) in tables.items():
+        if tname in name or name in tname:
+            got = store_backed_lookup(store, name, k[:5], v.shape[1])
+            torch.testing.assert_close(got, v[:5], rtol=1e-6, atol=1e-6)
+            break
+
+    # file-backed round trip
+    fstore = FileFeatureStore(str(tmp_path / "fs"))
+    publish_checkpoint(fstore, str(ckpt))
+    fstore.flush()
+    fstore2 = FileFeatureStore(str(tmp_path / "fs"))
+    assert sorted(fstore2.tables()) == sorted(fstore.tables())
+    missing = store_backed_lookup(fstore2, name,
+                                  torch.tensor([10 ** 12]), 4, default=0.5)
+    assert torch.allclose(missing, torch.full((1, 4), 0.5))
