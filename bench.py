@@ -78,7 +78,6 @@ def main():
     opt = make_optimizer(args.optimizer, params=model.parameters(),
                          embedding_variables=model.embedding_variables(),
                          learning_rate=0.001, **opt_kw)
-
     if distributed:
         from deeprec_amd.parallel import (DenseGradAllreducer,
                                           broadcast_parameters)
@@ -88,6 +87,11 @@ def main():
         opt.pre_dense_step = reducer.wait
     else:
         reducer = None
+    if bf16:
+        # AFTER the parameter broadcast: the bf16 shadows snapshot the
+        # weights at enable time
+        from deeprec_amd.ops.fused_mlp import enable_weight_cache
+        opt.post_step_hook = enable_weight_cache(model)
 
     # pre-generate batches outside the timed region (CPU RNG is not the
     # system under test); embedding ids differ per step so hash-table and

@@ -36,6 +36,8 @@ class Optimizer:
         # distributed seam: set to DenseGradAllreducer.wait to overlap an
         # async dense-grad all-reduce with the sparse applies
         self.pre_dense_step = None
+        # post-step seam (e.g. fused_mlp.enable_weight_cache's refresh)
+        self.post_step_hook = None
         self._step_count = 0
 
     # -- subclass hooks --
@@ -77,6 +79,8 @@ class Optimizer:
             self.pre_dense_step()
         if self._dense is not None:
             self._dense.step()
+        if self.post_step_hook is not None:
+            self.post_step_hook()
         self._post_step()
         if increment_global_step:
             GLOBAL_STEP.increment()

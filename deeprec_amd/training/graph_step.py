@@ -36,6 +36,9 @@ class GraphedTrainStep:
         self.optimizer = optimizer
         self.loss_fn = loss_fn
         self.graph = None
+        from deeprec_amd.ops.fused_mlp import enable_weight_cache
+        if optimizer.post_step_hook is None:
+            optimizer.post_step_hook = enable_weight_cache(model)
         self._eager(example_batch)  # ensure slabs exist before presizing
 
         try:
