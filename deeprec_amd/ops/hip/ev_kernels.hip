@@ -204,14 +204,19 @@ __global__ void k_dedup_pass_a(
         if (prev != EMPTY_KEY && prev != key) continue;
         if (prev == EMPTY_KEY) atomicAdd(entry_counter, 1);
       }
-      atomicAdd(&ht_freq[idx], 1);  // exact occurrence counting
-      ht_version[idx] = step;
-      int old = atomicExch(&ht_epoch[idx], epoch);
-      if (old != epoch) {  // first toucher this step
-        int c = atomicAdd(m_counter, 1);
-        ht_compact[idx] = c;
-        uniq_keys[c] = key;
-        compact_entry[c] = idx;
+      // freq/version updates moved to pass B (one per UNIQUE key via
+      // pass C's exact counts): hot keys were serializing thousands of
+      // per-occurrence atomics on one cache line here. The plain read
+      // short-circuits the epoch claim for already-claimed keys; racing
+      // first-touchers are resolved by the atomicExch.
+      if (ht_epoch[idx] != epoch) {
+        int old = atomicExch(&ht_epoch[idx], epoch);
+        if (old != epoch) {  // first toucher this step
+          int c = atomicAdd(m_counter, 1);
+          ht_compact[idx] = c;
+          uniq_keys[c] = key;
+          compact_entry[c] = idx;
+        }
       }
       goto next_j;
     }
@@ -258,14 +263,15 @@ __global__ void k_dedup_pass_a_dev(
         if (prev != EMPTY_KEY && prev != key) continue;
         if (prev == EMPTY_KEY) atomicAdd(entry_counter, 1);
       }
-      atomicAdd(&ht_freq[idx], 1);
-      ht_version[idx] = step;
-      int old = atomicExch(&ht_epoch[idx], epoch);
-      if (old != epoch) {
-        int c = atomicAdd(m_counter, 1);
-        ht_compact[idx] = c;
-        uniq_keys[c] = key;
-        compact_entry[c] = idx;
+      // freq/version handled per-unique in pass B (see k_dedup_pass_a)
+      if (ht_epoch[idx] != epoch) {
+        int old = atomicExch(&ht_epoch[idx], epoch);
+        if (old != epoch) {
+          int c = atomicAdd(m_counter, 1);
+          ht_compact[idx] = c;
+          uniq_keys[c] = key;
+          compact_entry[c] = idx;
+        }
       }
       goto next_j2;
     }
@@ -274,11 +280,15 @@ __global__ void k_dedup_pass_a_dev(
   }
 }
 
-// Pass B (per unique key): admission + default-value init, slots out.
+// Pass B (per unique key): freq/version bookkeeping (one write per
+// unique key, using pass C's exact per-batch counts — runs AFTER pass
+// C), then admission + default-value init, slots out.
 __global__ void k_dedup_pass_b(
     const int64_t* __restrict__ compact_entry,
     const int64_t* __restrict__ uniq_keys, int m,
-    int32_t* __restrict__ ht_slot, const int32_t* __restrict__ ht_freq,
+    int32_t* __restrict__ ht_slot, int32_t* __restrict__ ht_freq,
+    int64_t* __restrict__ ht_version,
+    const int32_t* __restrict__ batch_counts, int64_t step,
     int32_t* __restrict__ slot_counter, int max_slots,
     float* __restrict__ values, const float* __restrict__ default_values,
     int dim, int default_value_dim, int key_bits, int init_limit,
@@ -287,6 +297,8 @@ __global__ void k_dedup_pass_b(
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= m) return;
   int64_t idx = compact_entry[c];
+  ht_freq[idx] += batch_counts[c];
+  ht_version[idx] = step;
   int32_t slot = ht_slot[idx];
   if (slot < 0 && ht_freq[idx] >= filter_freq) {
     slot = atomicAdd(slot_counter, 1);
@@ -322,7 +334,10 @@ __global__ void k_dedup_pass_b_padded(
     const int64_t* __restrict__ compact_entry,
     const int64_t* __restrict__ uniq_keys, int n_cap,
     const int32_t* __restrict__ m_dev, int32_t* __restrict__ ht_slot,
-    const int32_t* __restrict__ ht_freq, int32_t* __restrict__ slot_counter,
+    int32_t* __restrict__ ht_freq, int64_t* __restrict__ ht_version,
+    const int32_t* __restrict__ batch_counts,
+    const int64_t* __restrict__ step_dev,
+    int32_t* __restrict__ slot_counter,
     int max_slots, float* __restrict__ values,
     const float* __restrict__ default_values, int dim,
     int default_value_dim, int key_bits, int init_limit, int filter_freq,
@@ -334,6 +349,8 @@ __global__ void k_dedup_pass_b_padded(
     return;
   }
   int64_t idx = compact_entry[c];
+  ht_freq[idx] += batch_counts[c];
+  ht_version[idx] = *step_dev;
   int32_t slot = ht_slot[idx];
   if (slot < 0 && ht_freq[idx] >= filter_freq) {
     slot = atomicAdd(slot_counter, 1);
@@ -959,6 +976,8 @@ torch::Tensor ht_dedup_a(torch::Tensor keys, torch::Tensor ht_keys,
 
 torch::Tensor ht_dedup_b(torch::Tensor compact_entry, torch::Tensor uniq_keys,
                          torch::Tensor ht_slot, torch::Tensor ht_freq,
+                         torch::Tensor ht_version,
+                         torch::Tensor batch_counts, int64_t step,
                          torch::Tensor slot_counter, int64_t max_slots,
                          torch::Tensor values, torch::Tensor default_values,
                          int64_t dvd_per_table, int64_t key_bits,
@@ -971,7 +990,8 @@ torch::Tensor ht_dedup_b(torch::Tensor compact_entry, torch::Tensor uniq_keys,
   k_dedup_pass_b<<<n_blocks(m), kBlock, 0, stream>>>(
       compact_entry.data_ptr<int64_t>(), uniq_keys.data_ptr<int64_t>(), m,
       ht_slot.data_ptr<int32_t>(), ht_freq.data_ptr<int32_t>(),
-      slot_counter.data_ptr<int32_t>(), (int)max_slots,
+      ht_version.data_ptr<int64_t>(), batch_counts.data_ptr<int32_t>(),
+      step, slot_counter.data_ptr<int32_t>(), (int)max_slots,
       values.data_ptr<float>(), default_values.data_ptr<float>(),
       values.size(1), (int)dvd_per_table, (int)key_bits, (int)init_limit,
       (int)filter_freq, slots.data_ptr<int32_t>(),
@@ -1008,6 +1028,8 @@ torch::Tensor ht_dedup_a_dev(
 torch::Tensor ht_dedup_b_padded(
     torch::Tensor compact_entry, torch::Tensor uniq_keys,
     torch::Tensor m_dev, torch::Tensor ht_slot, torch::Tensor ht_freq,
+    torch::Tensor ht_version, torch::Tensor batch_counts,
+    torch::Tensor step_dev,
     torch::Tensor slot_counter, int64_t max_slots, torch::Tensor values,
     torch::Tensor default_values, int64_t dvd_per_table, int64_t key_bits,
     int64_t init_limit, int64_t filter_freq, torch::Tensor error_flag) {
@@ -1018,7 +1040,9 @@ torch::Tensor ht_dedup_b_padded(
   k_dedup_pass_b_padded<<<n_blocks(n_cap), kBlock, 0, stream>>>(
       compact_entry.data_ptr<int64_t>(), uniq_keys.data_ptr<int64_t>(),
       n_cap, m_dev.data_ptr<int32_t>(), ht_slot.data_ptr<int32_t>(),
-      ht_freq.data_ptr<int32_t>(), slot_counter.data_ptr<int32_t>(),
+      ht_freq.data_ptr<int32_t>(), ht_version.data_ptr<int64_t>(),
+      batch_counts.data_ptr<int32_t>(), step_dev.data_ptr<int64_t>(),
+      slot_counter.data_ptr<int32_t>(),
       (int)max_slots, values.data_ptr<float>(),
       default_values.data_ptr<float>(), values.size(1), (int)dvd_per_table,
       (int)key_bits, (int)init_limit, (int)filter_freq,

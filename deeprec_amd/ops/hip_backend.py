@@ -229,14 +229,18 @@ class HbmStorage:
                                 self._step_dev, self.entry_counter,
                                 m_counter, uniq_buf, centry_buf,
                                 self.error_flag)
-        slots = self.ext.ht_dedup_b_padded(
-            centry_buf, uniq_buf, m_counter, self.ht_slot, self.ht_freq,
-            self.slot_counter, self.max_slots, self.values,
-            self.default_values, self.dvd_per_table, self.key_bits,
-            self._init_limit(), self.filter_freq, self.error_flag)
+        # pass C first: its exact per-batch counts feed pass B's single
+        # per-unique freq update (pass A is claim-only — hot keys no
+        # longer serialize per-occurrence atomics)
         inverse, counts, rank = self.ext.ht_dedup_c(
             values_cat, self.ht_keys, self.ht_compact, nnz)
         self._last_rank = rank  # consumed by collection._prep_backward
+        slots = self.ext.ht_dedup_b_padded(
+            centry_buf, uniq_buf, m_counter, self.ht_slot, self.ht_freq,
+            self.ht_version, counts, self._step_dev,
+            self.slot_counter, self.max_slots, self.values,
+            self.default_values, self.dvd_per_table, self.key_bits,
+            self._init_limit(), self.filter_freq, self.error_flag)
         return uniq_buf, inverse, counts, slots
 
     def prefers_dedup(self) -> bool:
@@ -271,14 +275,16 @@ class HbmStorage:
         self._slots_hint = int(c[2]) + m  # pass B may admit up to m slots
         self.observe_uniq_ratio(m, nnz)
         uniq = uniq_buf[:m]
-        slots = self.ext.ht_dedup_b(
-            centry_buf[:m], uniq, self.ht_slot, self.ht_freq,
-            self.slot_counter, self.max_slots, self.values,
-            self.default_values, self.dvd_per_table, self.key_bits,
-            self._init_limit(), self.filter_freq, self.error_flag)
+        # pass C first: exact counts feed pass B's per-unique freq update
         inverse, counts, rank = self.ext.ht_dedup_c(
             values_cat, self.ht_keys, self.ht_compact, m)
         self._last_rank = rank  # consumed by collection._prep_backward
+        slots = self.ext.ht_dedup_b(
+            centry_buf[:m], uniq, self.ht_slot, self.ht_freq,
+            self.ht_version, counts, step,
+            self.slot_counter, self.max_slots, self.values,
+            self.default_values, self.dvd_per_table, self.key_bits,
+            self._init_limit(), self.filter_freq, self.error_flag)
         return uniq, inverse, counts, slots
 
     def _use_no_permission(self) -> bool:
