@@ -34,13 +34,18 @@ class ShardedEmbeddingCollection:
                  ev_option: Optional[EmbeddingVariableOption] = None,
                  combiners=None, device=None, value_dtype=torch.float32,
                  generator=None, trainable: bool = True,
-                 comm_dtype: Optional[torch.dtype] = None):
+                 comm_dtype: Optional[torch.dtype] = None,
+                 node_size: Optional[int] = None):
         self.world = comm.world_size()
         self.rank = comm.rank()
         # comm_dtype=torch.bfloat16 halves the xGMI bytes of the row-return
         # all-to-all; lossless end-to-end for bf16 models (the pooled rows
         # are cast to bf16 for the MLP anyway). Gradients stay fp32.
         self.comm_dtype = comm_dtype
+        # node_size (= GPUs per node, e.g. LOCAL_WORLD_SIZE) switches the
+        # exchanges to the two-hop node-aware all-to-all when the job
+        # spans nodes (parallel/hierarchical.py)
+        self.node_size = node_size
         self.local = EmbeddingCollection(
             f"{name}/part_{self.rank}", table_names, embedding_dim,
             ev_option, combiners, device, value_dtype, generator, trainable)
@@ -146,6 +151,15 @@ class ShardedEmbeddingCollection:
             weights_cat, batch, out_dtype)
 
 
+def _a2a(sev: ShardedEmbeddingCollection, tensor, in_sp, out_sp):
+    if sev.node_size and sev.world > sev.node_size:
+        from deeprec_amd.parallel.hierarchical import hierarchical_all_to_all
+        out, oc = hierarchical_all_to_all(tensor, in_sp, sev.node_size)
+        assert oc == list(out_sp)
+        return out
+    return comm.all_to_all_single(tensor, in_sp, out_sp)
+
+
 def _exchange_lookup(sev: ShardedEmbeddingCollection, uniq, counts, train):
     """Route unique composite keys to owners, lookup there, return
     ([m, D] fp32 embedding rows in uniq order, owner-side context)."""
@@ -159,8 +173,8 @@ def _exchange_lookup(sev: ShardedEmbeddingCollection, uniq, counts, train):
     send_splits = torch.bincount(owner, minlength=w)
     recv_splits = comm.exchange_counts(send_splits)
     in_sp, out_sp = send_splits.tolist(), recv_splits.tolist()
-    recv_keys = comm.all_to_all_single(send_keys, in_sp, out_sp)
-    recv_counts = comm.all_to_all_single(send_counts, in_sp, out_sp)
+    recv_keys = _a2a(sev, send_keys, in_sp, out_sp)
+    recv_counts = _a2a(sev, send_counts, in_sp, out_sp)
     uniq2, inv2 = torch.unique(recv_keys, return_inverse=True)
     counts2 = torch.zeros(uniq2.numel(), dtype=recv_counts.dtype,
                           device=uniq.device)
@@ -211,8 +225,7 @@ class _ShardedCollectionLookup(torch.autograd.Function):
                                      chunk_k0, row_ids_cat, ctx.weights_cat,
                                      row_coeff, uniq.numel(), ctx.batch)
         grad_send = grad_unique[order_o]
-        grad_recv = comm.all_to_all_single(grad_send.contiguous(),
-                                           in_sp, out_sp)
+        grad_recv = _a2a(sev, grad_send.contiguous(), in_sp, out_sp)
         grad2 = torch.zeros(uniq2.numel(), grad_recv.shape[1],
                             device=grad_recv.device, dtype=grad_recv.dtype)
         grad2.index_add_(0, inv2.long(), grad_recv)

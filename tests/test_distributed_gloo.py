@@ -310,3 +310,50 @@ def _body_async_allreduce(rank, world):
 
 def test_async_allreduce_world2():
     _run_dist(_body_async_allreduce, world_size=2, port=29544)
+
+
+def _body_hierarchical_a2a(rank, world):
+    from deeprec_amd.parallel import comm
+    from deeprec_amd.parallel.hierarchical import hierarchical_all_to_all
+
+    torch.manual_seed(100 + rank)
+    for trial in range(3):
+        splits = torch.randint(0, 7, (world,)).tolist()
+        if trial == 2:
+            splits[rank] = 0  # empty self-block edge case
+        rows = torch.randn(sum(splits), 5) + rank * 100
+        flat_counts = comm.exchange_counts(
+            torch.tensor(splits, dtype=torch.int64))
+        flat = comm.all_to_all_single(rows, splits, flat_counts.tolist())
+        hier, hc = hierarchical_all_to_all(rows.clone(), splits,
+                                           node_size=2)
+        assert hc == flat_counts.tolist()
+        torch.testing.assert_close(hier, flat)
+    dist.barrier()
+
+
+def test_hierarchical_a2a_world4():
+    _run_dist(_body_hierarchical_a2a, world_size=4, port=29545)
+
+
+def _body_sharded_hierarchical(rank, world):
+    from deeprec_amd.parallel.sharded_collection import (
+        ShardedEmbeddingCollection)
+
+    torch.manual_seed(3)
+    names = [f"t{i}" for i in range(4)]
+    ids = torch.randint(0, 60, (16, 4))
+    g1 = torch.Generator().manual_seed(9)
+    g2 = torch.Generator().manual_seed(9)
+    flat = ShardedEmbeddingCollection("hflat", names, 8, generator=g1)
+    hier = ShardedEmbeddingCollection("hhier", names, 8, generator=g2,
+                                      node_size=2)
+    o1 = flat.lookup_matrix(ids, train=True)
+    o2 = hier.lookup_matrix(ids, train=True)
+    torch.testing.assert_close(o1, o2)
+    (o1.sum() + o2.sum()).backward()
+    dist.barrier()
+
+
+def test_sharded_hierarchical_world4():
+    _run_dist(_body_sharded_hierarchical, world_size=4, port=29546)
