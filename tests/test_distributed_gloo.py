@@ -357,3 +357,48 @@ def _body_sharded_hierarchical(rank, world):
 
 def test_sharded_hierarchical_world4():
     _run_dist(_body_sharded_hierarchical, world_size=4, port=29546)
+
+
+def _body_elastic_resize(rank, world):
+    """world=2 trains, coordinator proposes shrink to 1, both checkpoint;
+    the restore-under-new-world path is covered by the repartition tests."""
+    import tempfile
+
+    from deeprec_amd.checkpoint.saver import Saver
+    from deeprec_amd.embedding import EmbeddingVariable, embedding_lookup
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.parallel.elastic import (ElasticAgent,
+                                              ElasticController,
+                                              elastic_step_hook)
+
+    port = 29650
+    ctrl = ElasticController(port=port) if rank == 0 else None
+    dist.barrier()  # store server up before agents connect
+    agent = ElasticAgent(port=port)
+
+    torch.manual_seed(rank)
+    ev = EmbeddingVariable(f"elastic_{rank}", 4, device="cpu")
+    opt = AdagradOptimizer(embedding_variables=[ev], learning_rate=0.1)
+    saver = Saver(embedding_variables=[ev], rank=rank, world_size=world)
+    tmp = tempfile.mkdtemp(prefix=f"elastic{rank}_")
+
+    stopped_at = None
+    for step in range(6):
+        ids = torch.randint(0, 40, (8,))
+        out = embedding_lookup(ev, ids, train=True)
+        (out ** 2).sum().backward()
+        opt.step()
+        if rank == 0 and step == 2:
+            ctrl.propose_resize(1)
+        dist.barrier()  # all ranks see the event at the same boundary
+        if elastic_step_hook(agent, saver, tmp, global_step=step):
+            stopped_at = step
+            break
+    assert stopped_at == 2  # the step the event was proposed
+    import glob as g
+    assert g.glob(tmp + "/ckpt-*"), "scale event must leave a checkpoint"
+    dist.barrier()
+
+
+def test_elastic_resize_world2():
+    _run_dist(_body_elastic_resize, world_size=2, port=29547)
