@@ -6,3 +6,6 @@ from deeprec_amd.embedding.variable import (  # noqa: F401
 from deeprec_amd.embedding.lookup import (  # noqa: F401
     embedding_lookup, embedding_lookup_sparse,
     safe_embedding_lookup_sparse, group_embedding_lookup_sparse)
+from deeprec_amd.embedding.hash_table import (  # noqa: F401,E402
+    BloomFilterAdmitStrategy, CounterAdmitStrategy,
+    DistributedHashTable, HashTable)

@@ -286,3 +286,39 @@ def test_kafka_dataset_offsets_and_resume(tmp_path):
     ds3.restore(str(ck))
     rest = next(iter(ds3))
     assert [m["id"] for m in rest] == list(range(4, 10))
+
+
+def test_hash_table_api():
+    from deeprec_amd.embedding.hash_table import (
+        BloomFilterAdmitStrategy, DistributedHashTable, HashTable)
+    from deeprec_amd.optimizers import AdagradOptimizer
+
+    torch.manual_seed(0)
+    ht = HashTable([8], name="ht_plain")
+    keys = torch.tensor([3, 7, 3, 11])
+    rows = ht.lookup(keys)
+    assert rows.shape == (4, 8) and ht.size() == 3
+    # trains through the standard optimizer via the EV escape hatch
+    opt = AdagradOptimizer(embedding_variables=[ht.embedding_variable()],
+                          learning_rate=0.1)
+    before = ht.lookup(keys, admit=False).detach().clone()
+    (ht.lookup(keys) ** 2).sum().backward()
+    opt.step()
+    assert not torch.allclose(before, ht.lookup(keys, admit=False))
+
+    # bloom admission: below min_frequency nothing is admitted
+    htf = HashTable([4], name="ht_bloom",
+                    admit_strategy=BloomFilterAdmitStrategy(min_frequency=3))
+    k = torch.tensor([5])
+    htf.lookup(k)
+    assert htf.size() == 0
+    htf.lookup(k)
+    htf.lookup(k)
+    assert htf.size() == 1
+
+    dht = DistributedHashTable([4], num_partitions=2, name="dht")
+    dkeys = torch.arange(10)
+    out = dht.lookup(dkeys)
+    assert out.shape == (10, 4) and dht.size() == 10
+    # routing is stable: same keys hit the same partitions
+    torch.testing.assert_close(out, dht.lookup(dkeys, admit=False))
