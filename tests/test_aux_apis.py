@@ -322,3 +322,43 @@ def test_hash_table_api():
     assert out.shape == (10, 4) and dht.size() == 10
     # routing is stable: same keys hit the same partitions
     torch.testing.assert_close(out, dht.lookup(dkeys, admit=False))
+
+
+def test_rebalance_hook_noop_on_single_tier():
+    from deeprec_amd.embedding import EmbeddingVariable
+    from deeprec_amd.training.session import RebalanceHook
+
+    from deeprec_amd.embedding.variable import reset_registry
+    reset_registry()
+    ev = EmbeddingVariable("reb_cpu", 4, device="cpu")
+    hook = RebalanceHook(every_steps=1, variables=[ev])
+    hook.after_run(None)  # single-tier storage: rebalance() returns 0
+    assert ev.rebalance() == 0
+
+
+def test_estimator_evaluate_reports_auc():
+    import torch as t
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.training.estimator import Estimator
+
+    def model_fn(params):
+        m = DLRM(device="cpu", bf16=False, num_sparse=3,
+                 mlp_bot=(16,), mlp_top=(16,))
+        o = AdagradOptimizer(params=m.parameters(),
+                             embedding_variables=m.embedding_variables(),
+                             learning_rate=0.05)
+        return m, o
+
+    def input_fn():
+        t.manual_seed(0)
+        while True:
+            dense = t.randn(32, 13)
+            ids = t.randint(0, 50, (32, 3))
+            labels = (dense[:, 0] > 0).float()
+            yield dense, ids, labels
+
+    est = Estimator(model_fn)
+    metrics = est.evaluate(input_fn, steps=5)
+    assert set(metrics) == {"loss", "accuracy", "auc"}
+    assert 0.0 <= metrics["auc"] <= 1.0
