@@ -119,6 +119,9 @@ class EmbeddingVariable:
     def restore(self, keys, values, freqs=None, versions=None):
         self.storage.import_(keys, values, freqs, versions)
 
+    def memory_usage(self) -> dict:
+        return self.storage.memory_usage()
+
     def shrink(self, step: Optional[int] = None) -> int:
         return self.storage.shrink(
             step if step is not None else get_global_step())

@@ -332,6 +332,16 @@ class CpuStorage:
                 for name, rows in slab_rows.items():
                     self.get_slab(name, rows.shape[1], 0.0)[s] = rows[i]
 
+    def memory_usage(self) -> dict:
+        values = self.values.numel() * self.values.element_size()
+        slabs = sum(t.numel() * t.element_size()
+                    for t in self.slabs.values())
+        n = self.n_entries
+        table = n * (8 + 4 + 4 + 8 + 1)  # key/slot/freq/version/alive
+        return {"table_bytes": table, "values_bytes": values,
+                "slab_bytes": slabs,
+                "total_bytes": table + values + slabs}
+
     def size(self) -> int:
         n = self.n_entries
         return int((self.entry_alive[:n] & (self.entry_slot[:n] >= 0)).sum())

@@ -328,6 +328,15 @@ class HbmDramStorage(HbmStorage):
         self._sync_counters()
         return moved
 
+    def memory_usage(self) -> dict:
+        out = super().memory_usage()
+        cold = (self.values_cold.numel() * self.values_cold.element_size()
+                + sum(t.numel() * t.element_size()
+                      for t in self.cold_slabs.values()))
+        out["cold_bytes"] = cold
+        out["total_bytes"] += cold
+        return out
+
     def export(self, include_filtered: bool = False):
         keys, slots, freqs, versions = self._export_entries()
         adm = slots >= 0

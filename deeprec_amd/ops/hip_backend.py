@@ -416,6 +416,21 @@ class HbmStorage:
         self._check_error()
         self._sync_counters()
 
+    def memory_usage(self) -> dict:
+        """Byte accounting (reference capability:
+        embedding_variable_memory_test.cc): hash-table arrays, value slab,
+        optimizer slabs."""
+        table = sum(t.numel() * t.element_size()
+                    for t in (self.ht_keys, self.ht_slot, self.ht_freq,
+                              self.ht_version, self.ht_epoch,
+                              self.ht_compact))
+        values = self.values.numel() * self.values.element_size()
+        slabs = sum(t.numel() * t.element_size()
+                    for t in self.slabs.values())
+        return {"table_bytes": table, "values_bytes": values,
+                "slab_bytes": slabs,
+                "total_bytes": table + values + slabs}
+
     def size(self) -> int:
         _, slots, _, _ = self._export_entries()
         return int((slots >= 0).sum())

@@ -82,3 +82,37 @@ def exchange_counts(counts: torch.Tensor) -> torch.Tensor:
     if w == 1:
         return counts.clone()
     return all_to_all_single(counts.contiguous(), [1] * w, [1] * w)
+
+
+def padded_all_to_all(inp: torch.Tensor, in_splits, cap: int):
+    """Fixed-shape all-to-all: every peer slot padded to `cap` rows.
+
+    The building block for capturing the DISTRIBUTED step in a hipGraph
+    (RCCL collectives are capturable, but only with static shapes): both
+    the payload (w*cap rows each way) and the true counts (w ints) have
+    shapes independent of the data. Rows beyond a slot's true count are
+    zeros on the wire and must be ignored via the returned counts.
+
+    Returns (out [w*cap, ...] with peer p's rows at [p*cap, p*cap+n_p),
+    out_counts int64 [w]). Raises if any split exceeds cap.
+    """
+    w = world_size()
+    if max(in_splits, default=0) > cap:
+        raise ValueError(f"split {max(in_splits)} exceeds pad cap {cap}")
+    if w == 1:
+        out = torch.zeros((cap,) + tuple(inp.shape[1:]), dtype=inp.dtype,
+                          device=inp.device)
+        out[: inp.shape[0]] = inp
+        return out, torch.tensor(list(in_splits), dtype=torch.int64)
+    send = torch.zeros((w * cap,) + tuple(inp.shape[1:]),
+                       dtype=inp.dtype, device=inp.device)
+    off = 0
+    for p, n in enumerate(in_splits):
+        send[p * cap: p * cap + n] = inp[off: off + n]
+        off += n
+    counts = torch.tensor(list(in_splits), dtype=torch.int64,
+                          device=inp.device if dist.get_backend() != "gloo"
+                          else "cpu")
+    out_counts = exchange_counts(counts.cpu()).to(counts.device)
+    out = all_to_all_single(send, [cap] * w, [cap] * w)
+    return out, out_counts.cpu()

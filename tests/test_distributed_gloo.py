@@ -402,3 +402,36 @@ def _body_elastic_resize(rank, world):
 
 def test_elastic_resize_world2():
     _run_dist(_body_elastic_resize, world_size=2, port=29547)
+
+
+def _body_padded_a2a(rank, world):
+    from deeprec_amd.parallel import comm
+
+    torch.manual_seed(20 + rank)
+    splits = [rank + 1, 3 - rank][:world]
+    rows = torch.randn(sum(splits), 4) + rank * 10
+    flat_counts = comm.exchange_counts(
+        torch.tensor(splits, dtype=torch.int64))
+    flat = comm.all_to_all_single(rows, splits, flat_counts.tolist())
+    cap = 8
+    padded, counts = comm.padded_all_to_all(rows, splits, cap)
+    assert padded.shape[0] == world * cap
+    assert counts.tolist() == flat_counts.tolist()
+    trimmed = torch.cat([padded[p * cap: p * cap + int(counts[p])]
+                         for p in range(world)])
+    torch.testing.assert_close(trimmed, flat)
+    # pad region is zeros on the wire
+    for p in range(world):
+        tail = padded[p * cap + int(counts[p]): (p + 1) * cap]
+        assert torch.equal(tail, torch.zeros_like(tail))
+    # over-cap raises
+    try:
+        comm.padded_all_to_all(rows, [cap + 1] + [0] * (world - 1), cap)
+        raise AssertionError("expected ValueError")
+    except ValueError:
+        pass
+    dist.barrier()
+
+
+def test_padded_a2a_world2():
+    _run_dist(_body_padded_a2a, world_size=2, port=29548)

