@@ -211,3 +211,21 @@ def test_end_to_end_training_matches_cpu():
     w_g = ev_g.gather(keys.to(DEV)).cpu()
     w_c = ev_c.gather(keys)
     torch.testing.assert_close(w_g, w_c, rtol=1e-4, atol=1e-5)
+
+
+@pytest.mark.gpu
+def test_memory_usage_gpu():
+    from deeprec_amd.embedding import EmbeddingVariable, embedding_lookup
+    from deeprec_amd.optimizers import AdamAsyncOptimizer
+
+    ev = EmbeddingVariable("mem_gpu", 16, device="cuda")
+    opt = AdamAsyncOptimizer(embedding_variables=[ev])
+    out = embedding_lookup(ev, torch.arange(500, device="cuda"),
+                           train=True)
+    out.sum().backward()
+    opt.step()
+    mu = ev.memory_usage()
+    assert mu["values_bytes"] >= 500 * 16 * 4
+    assert mu["slab_bytes"] >= 2 * mu["values_bytes"] // 2
+    assert mu["total_bytes"] == sum(
+        v for k, v in mu.items() if k != "total_bytes")
