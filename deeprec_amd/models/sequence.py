@@ -71,9 +71,13 @@ class DIN(_SeqBase):
         self.to(self.device_)
 
     def attend(self, seq, target, mask):
-        """seq [B,T,D], target [B,D] -> [B,D] attention-pooled."""
-        t = target.unsqueeze(1).expand_as(seq)
-        att_in = torch.cat([seq, t, seq - t, seq * t], dim=2)
+        """seq [B,T,D], target [B,D] -> [B,D] attention-pooled. The
+        attention features are built in the compute dtype (bf16 halves
+        the [B,T,4D] cat traffic, ~210 MB/step fp32 at batch 8192)."""
+        cd = self.compute_dtype if self.bf16 else seq.dtype
+        s16 = seq.to(cd)
+        t16 = target.to(cd).unsqueeze(1).expand_as(s16)
+        att_in = torch.cat([s16, t16, s16 - t16, s16 * t16], dim=2)
         scores = self.att(att_in).float().squeeze(2)  # [B, T]
         scores = scores.masked_fill(~mask, -1e9)
         w = torch.softmax(scores, dim=1)
@@ -132,8 +136,9 @@ class DIEN(_SeqBase):
         mask = (seq_ids > 0).float()
         h_seq = self.gru(seq)                            # [B,T,H]
         tgt_h = self.target_proj(target)                 # [B,H]
+        cd = self.compute_dtype if self.bf16 else h_seq.dtype
         att_in = torch.cat(
-            [h_seq, tgt_h.unsqueeze(1).expand_as(h_seq)], 2)
+            [h_seq.to(cd), tgt_h.to(cd).unsqueeze(1).expand_as(h_seq)], 2)
         scores = self.att(att_in).float().squeeze(2)
         scores = scores.masked_fill(mask == 0, -1e9)
         alpha = torch.softmax(scores, 1) * mask          # [B,T]
