@@ -32,3 +32,16 @@ from deeprec_amd.embedding.lookup import (  # noqa: F401
     group_embedding_lookup_sparse,
 )
 from deeprec_amd.embedding.ragged import RaggedIds  # noqa: F401
+from deeprec_amd.embedding.collection import (  # noqa: F401
+    EmbeddingCollection,
+)
+from deeprec_amd.embedding.hash_table import (  # noqa: F401
+    BloomFilterAdmitStrategy,
+    DistributedHashTable,
+    HashTable,
+)
+from deeprec_amd.embedding.options import CacheStrategy  # noqa: F401
+from deeprec_amd.ops.fused_norm import (  # noqa: F401
+    fused_l2_normalize,
+    fused_layer_norm,
+)
