@@ -111,6 +111,25 @@ class Estimator:
                 "accuracy": n_correct / max(n_total, 1),
                 "auc": auc.result()}
 
+    def train_and_evaluate(self, train_input_fn: Callable,
+                           eval_input_fn: Callable, train_steps: int,
+                           eval_steps: int = 10,
+                           eval_every: Optional[int] = None,
+                           hooks=None) -> list:
+        """Interleaved train/eval (tf.estimator.train_and_evaluate
+        surface): trains `train_steps` total, evaluating every
+        `eval_every` steps (default: once at the end). Returns the list
+        of eval metric dicts in order."""
+        results = []
+        every = eval_every or train_steps
+        done = 0
+        while done < train_steps:
+            chunk = min(every, train_steps - done)
+            self.train(train_input_fn, steps=chunk, hooks=hooks)
+            done += chunk
+            results.append(self.evaluate(eval_input_fn, steps=eval_steps))
+        return results
+
     @torch.no_grad()
     def predict(self, input_fn: Callable):
         for batch in input_fn():

@@ -362,3 +362,33 @@ def test_estimator_evaluate_reports_auc():
     metrics = est.evaluate(input_fn, steps=5)
     assert set(metrics) == {"loss", "accuracy", "auc"}
     assert 0.0 <= metrics["auc"] <= 1.0
+
+
+def test_estimator_train_and_evaluate():
+    import torch as t
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.training.estimator import Estimator
+
+    def model_fn(params):
+        m = DLRM(device="cpu", bf16=False, num_sparse=3,
+                 mlp_bot=(16,), mlp_top=(16,))
+        o = AdagradOptimizer(params=m.parameters(),
+                             embedding_variables=m.embedding_variables(),
+                             learning_rate=0.05)
+        return m, o
+
+    def input_fn():
+        g = t.Generator().manual_seed(1)
+        while True:
+            dense = t.randn(32, 13, generator=g)
+            ids = t.randint(0, 40, (32, 3), generator=g)
+            yield dense, ids, (dense[:, 0] > 0).float()
+
+    est = Estimator(model_fn)
+    results = est.train_and_evaluate(input_fn, input_fn, train_steps=12,
+                                     eval_steps=3, eval_every=6)
+    assert len(results) == 2
+    assert all({"loss", "accuracy", "auc"} <= set(r) for r in results)
+    # training on a learnable signal should not diverge
+    assert results[-1]["loss"] < results[0]["loss"] * 3
