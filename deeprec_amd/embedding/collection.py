@@ -200,7 +200,8 @@ class EmbeddingCollection:
         rank = getattr(self.storage, "_last_rank", None)
         self.storage._last_rank = None
         if rank is not None and rank.numel() == inverse.numel():
-            order = self.storage.ext.csr_scatter(inverse, rank, bounds)
+            order = self.storage.ext.csr_scatter(
+                inverse, rank, bounds, self.storage.error_flag)
         else:  # torch.unique (sort) path
             order = self.storage.ext.csr_order(inverse, bounds, m)
         return order, bounds, None, None

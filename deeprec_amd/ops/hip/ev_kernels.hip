@@ -419,14 +419,23 @@ __global__ void k_dedup_pass_c(
 }
 
 // order[bounds[c] + rank[j]] = j — direct scatter using pass C's ranks.
+// The destination is bounds-guarded: a corrupt inverse/rank pair (e.g. a
+// key pass C failed to find) must trip the error flag, not fault the GPU.
 __global__ void k_csr_scatter(const int32_t* __restrict__ inverse,
                               const int32_t* __restrict__ rank, int nnz,
                               const int32_t* __restrict__ bounds,
-                              int32_t* __restrict__ order) {
+                              int32_t* __restrict__ order,
+                              int32_t* __restrict__ error_flag) {
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
-  for (; j < nnz; j += stride)
-    order[bounds[inverse[j]] + rank[j]] = (int32_t)j;
+  for (; j < nnz; j += stride) {
+    int64_t o = (int64_t)bounds[inverse[j]] + rank[j];
+    if (o < 0 || o >= nnz) {
+      atomicExch(error_flag, 3);
+      continue;
+    }
+    order[o] = (int32_t)j;
+  }
 }
 
 // CSR order build: order[bounds[c] + pos++] = j (pos via per-key cursor).
@@ -1070,13 +1079,14 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ht_dedup_c(
 }
 
 torch::Tensor csr_scatter(torch::Tensor inverse, torch::Tensor rank,
-                          torch::Tensor bounds) {
+                          torch::Tensor bounds, torch::Tensor error_flag) {
   int64_t nnz = inverse.numel();
   auto order = torch::empty({nnz}, inverse.options());
   if (nnz == 0) return order;
   k_csr_scatter<<<n_blocks(nnz), kBlock, 0, current_stream()>>>(
       inverse.data_ptr<int32_t>(), rank.data_ptr<int32_t>(), (int)nnz,
-      bounds.data_ptr<int32_t>(), order.data_ptr<int32_t>());
+      bounds.data_ptr<int32_t>(), order.data_ptr<int32_t>(),
+      error_flag.data_ptr<int32_t>());
   return order;
 }
 

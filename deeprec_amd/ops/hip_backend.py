@@ -170,8 +170,9 @@ class HbmStorage:
         err = int(self.error_flag.cpu())
         if err != 0:
             raise RuntimeError(
-                f"HBM hash table error {err} (1=slab overflow, 2=table full)"
-                " — capacity management bug")
+                f"HBM hash table error {err} (1=slab overflow, "
+                "2=table full, 3=CSR scatter out of bounds)"
+                " — engine invariant violated")
 
     # ---------------- public interface ----------------
     def get_slab(self, name, width, init_value, dtype=torch.float32):
