@@ -1,0 +1,4 @@
+from deeprec_amd.checkpoint.saver import (  # noqa: F401
+    Saver,
+    latest_checkpoint,
+)
