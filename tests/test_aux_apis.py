@@ -392,3 +392,38 @@ def test_estimator_train_and_evaluate():
     assert all({"loss", "accuracy", "auc"} <= set(r) for r in results)
     # training on a learnable signal should not diverge
     assert results[-1]["loss"] < results[0]["loss"] * 3
+
+
+def test_async_embedding_stage_cpu():
+    from deeprec_amd.embedding.collection import EmbeddingCollection
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.training.async_stage import AsyncEmbeddingStage
+
+    torch.manual_seed(0)
+    g1 = torch.Generator().manual_seed(4)
+    g2 = torch.Generator().manual_seed(4)
+    names = [f"t{i}" for i in range(3)]
+    coll = EmbeddingCollection("async_c", names, 8, generator=g1)
+    ref = EmbeddingCollection("sync_c", names, 8, generator=g2)
+    opt = AdagradOptimizer(embedding_variables=[coll], learning_rate=0.1)
+    opt_r = AdagradOptimizer(embedding_variables=[ref], learning_rate=0.1)
+
+    batches = [torch.randint(0, 30, (8, 3)) for _ in range(5)]
+    stage = AsyncEmbeddingStage(coll)
+    stage.submit(batches[0])
+    for i in range(5):
+        emb = stage.take()
+        if i + 1 < len(batches):
+            stage.submit(batches[i + 1])
+        emb.sum().backward()
+        opt.step()
+        # reference synchronous path over the same batches
+        ref.lookup_matrix(batches[i]).sum().backward()
+        opt_r.step()
+    # on CPU the stage is synchronous -> trajectories identical
+    te, tr = coll.export_tables(), ref.export_tables()
+    for n in te:
+        ke, ve, _, _ = te[n]
+        kr, vr, _, _ = tr[list(tr)[list(te).index(n)]]
+        torch.testing.assert_close(ve[torch.argsort(ke)],
+                                   vr[torch.argsort(kr)])
