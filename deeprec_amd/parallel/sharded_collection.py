@@ -184,13 +184,17 @@ def _exchange_lookup(sev: ShardedEmbeddingCollection, uniq, counts, train):
     emb2 = coll.storage.gather(uniq2, slots2)        # [m2, D] fp32
     emb_out = emb2[inv2]                              # [n_recv, D]
     if sev.comm_dtype is not None:
-        # reduced-precision row exchange; int16 view keeps the transport
-        # dtype universally supported (gloo included)
-        payload = emb_out.to(sev.comm_dtype).view(torch.int16).contiguous()
-        emb_back = comm.all_to_all_single(payload, out_sp, in_sp)             .view(sev.comm_dtype).float()
+        # reduced-precision row exchange. RCCL ships bf16 natively
+        # (ncclBfloat16) but has NO int16 dtype; gloo is the opposite —
+        # the wire dtype is backend-dependent, payload bits identical.
+        lp = emb_out.to(sev.comm_dtype).contiguous()
+        if torch.distributed.get_backend() == "gloo":
+            emb_back = _a2a(sev, lp.view(torch.int16), out_sp,
+                            in_sp).view(sev.comm_dtype).float()
+        else:
+            emb_back = _a2a(sev, lp, out_sp, in_sp).float()
     else:
-        emb_back = comm.all_to_all_single(emb_out.contiguous(), out_sp,
-                                          in_sp)
+        emb_back = _a2a(sev, emb_out.contiguous(), out_sp, in_sp)
     emb = torch.empty_like(emb_back)
     emb[order_o] = emb_back                           # uniq order
     ctx = (order_o, inv2, slots2, uniq2, in_sp, out_sp)
