@@ -110,9 +110,10 @@ def padded_all_to_all(inp: torch.Tensor, in_splits, cap: int):
     for p, n in enumerate(in_splits):
         send[p * cap: p * cap + n] = inp[off: off + n]
         off += n
-    counts = torch.tensor(list(in_splits), dtype=torch.int64,
-                          device=inp.device if dist.get_backend() != "gloo"
-                          else "cpu")
-    out_counts = exchange_counts(counts.cpu()).to(counts.device)
+    # counts ride the collective too, so they must live where the
+    # backend wants its tensors (CUDA for nccl/RCCL, CPU for gloo)
+    cdev = inp.device if dist.get_backend() != "gloo" else "cpu"
+    counts = torch.tensor(list(in_splits), dtype=torch.int64, device=cdev)
+    out_counts = exchange_counts(counts)
     out = all_to_all_single(send, [cap] * w, [cap] * w)
     return out, out_counts.cpu()

@@ -132,10 +132,15 @@ def hierarchical_all_to_all(inp: torch.Tensor, splits_per_rank: List[int],
     sendbuf = inp[perm]
 
     # the full [w, w] splits matrix (one all_gather) provides every
-    # hop's exact counts: sender s holds all_splits[s][d] rows for rank d
-    all_splits = [torch.zeros(w, dtype=torch.int64) for _ in range(w)]
+    # hop's exact counts: sender s holds all_splits[s][d] rows for rank d.
+    # nccl/RCCL wants CUDA tensors in collectives; gloo wants CPU.
+    cdev = inp.device if dist.get_backend() != "gloo" else "cpu"
+    all_splits = [torch.zeros(w, dtype=torch.int64, device=cdev)
+                  for _ in range(w)]
     dist.all_gather(all_splits,
-                    torch.tensor(splits_per_rank, dtype=torch.int64))
+                    torch.tensor(splits_per_rank, dtype=torch.int64,
+                                 device=cdev))
+    all_splits = [t.cpu() for t in all_splits]
     me = comm.rank()
     my_node, my_local = me // node_size, me % node_size
     h1_out = [sum(int(all_splits[my_node * node_size + a][d])
