@@ -113,6 +113,9 @@ class EmbeddingVariableOption:
     init_option: InitializerOption = field(default_factory=InitializerOption)
     # initial hash-table capacity (entries); grows automatically
     init_capacity: int = 1 << 17
-    # record per-key frequency/version metadata even with no filter/evict
+    # API compatibility: this engine ALWAYS records per-key freq/version
+    # (SoA int32/int64 arrays are cheap in HBM and feed admission,
+    # eviction and LFU/LRU tier placement); the reference flags are
+    # accepted so option structs port unchanged
     record_freq: bool = False
     record_version: bool = False
