@@ -435,3 +435,8 @@ def _body_padded_a2a(rank, world):
 
 def test_padded_a2a_world2():
     _run_dist(_body_padded_a2a, world_size=2, port=29548)
+
+
+def test_hierarchical_a2a_world6():
+    """3 nodes x 2 GPUs — non-power-of-two node count."""
+    _run_dist(_body_hierarchical_a2a, world_size=6, port=29549)
