@@ -1,6 +1,7 @@
 from deeprec_amd.models.dlrm import DLRM  # noqa: F401
 from deeprec_amd.models.zoo import (  # noqa: F401
-    DBMTL, DCN, DSSM, ESMM, WDL, DeepFM, MMoE, MaskNet, PLE,
+    DBMTL, DCN, DSSM, ESMM, WDL, DeepFM, MMoE, MaskNet,
+    MLPerfDLRMDCN, PLE,
     SimpleMultiTask)
 from deeprec_amd.models.sequence import BST, DIEN, DIN  # noqa: F401
 
@@ -9,6 +10,7 @@ MODEL_REGISTRY = {
     "dcn": DCN, "dcnv2": lambda **kw: DCN(v2=True, **kw), "dssm": DSSM,
     "mmoe": MMoE, "esmm": ESMM, "dbmtl": DBMTL, "ple": PLE,
     "masknet": MaskNet, "simple_multitask": SimpleMultiTask,
+    "mlperf_dlrm_dcn": MLPerfDLRMDCN,
     "din": DIN, "dien": DIEN, "bst": BST,
 }
 SEQUENCE_MODELS = {"din", "dien", "bst"}

@@ -83,12 +83,19 @@ class DeepFM(RecModelBase):
 
 class _CrossNet(nn.Module):
     """DCN cross layers: x_{l+1} = x0 * (w^T x_l) + b + x_l
-    (reference: modelzoo/dcn). v2 uses a full matrix W per layer."""
+    (reference: modelzoo/dcn). v2 uses a full matrix W per layer;
+    low_rank > 0 factorizes it as U @ V^T (the MLPerf DLRM-DCN recipe,
+    reference: modelzoo/mlperf)."""
 
-    def __init__(self, dim, n_layers=3, v2=False):
+    def __init__(self, dim, n_layers=3, v2=False, low_rank=0):
         super().__init__()
         self.v2 = v2
-        if v2:
+        if v2 and low_rank > 0:
+            self.ws = nn.ModuleList([
+                nn.Sequential(nn.Linear(dim, low_rank, bias=False),
+                              nn.Linear(low_rank, dim))
+                for _ in range(n_layers)])
+        elif v2:
             self.ws = nn.ModuleList(
                 [nn.Linear(dim, dim) for _ in range(n_layers)])
         else:
@@ -130,6 +137,37 @@ class DCN(RecModelBase):
             deep_out = self.deep(x0.to(self.compute_dtype))
         return self.head(
             torch.cat([cross_out, deep_out.float()], 1)).squeeze(1)
+
+
+class MLPerfDLRMDCN(RecModelBase):
+    """MLPerf DLRM-DCNv2 (reference: modelzoo/mlperf): bottom MLP over
+    dense features, concat with the embeddings, low-rank DCNv2 cross
+    interaction, top MLP. Scaled to the synthetic harness shapes (the
+    MLPerf config uses dim 128 / rank 512 on real Criteo)."""
+
+    def __init__(self, embedding_dim=16, bot_sizes=(128, 64, 16),
+                 top_sizes=(512, 256), cross_layers=3, low_rank=32,
+                 device="cpu", bf16=True, **kw):
+        super().__init__(embedding_dim, device, bf16,
+                         name="mlperf_dlrm_dcn", **kw)
+        assert bot_sizes[-1] == embedding_dim
+        self.bot = make_mlp(bot_sizes, NUM_DENSE, device, self.bf16)
+        in_dim = (self.num_sparse + 1) * embedding_dim
+        self.cross = _CrossNet(in_dim, cross_layers, v2=True,
+                               low_rank=low_rank)
+        self.top = make_mlp(list(top_sizes) + [1], in_dim, device,
+                            self.bf16, final_activation=False)
+        self.to(self.device_)
+
+    def forward(self, dense, sparse_ids, train=True):
+        emb = self.sparse_feats(sparse_ids, train)
+        with self.amp():
+            dense_out = self.bot(dense.to(self.compute_dtype))
+        x0 = torch.cat([dense_out.float(), emb.flatten(1).float()], 1)
+        x = self.cross(x0)
+        with self.amp():
+            out = self.top(x.to(self.compute_dtype))
+        return out.float().squeeze(1)
 
 
 class DSSM(RecModelBase):
