@@ -121,6 +121,9 @@ def create_app(predictor: Predictor, batcher: Optional[DynamicBatcher] = None):
             return predictor.process(request)
         except KeyError as e:
             raise HTTPException(422, f"missing field {e}")
+        except (ValueError, TypeError, RuntimeError) as e:
+            # ragged lists, wrong dtypes, shape mismatches
+            raise HTTPException(422, f"malformed request: {e}")
 
     @app.post("/v1/predict_batch")
     def predict_batch(requests: list = Body(...)):

@@ -56,6 +56,10 @@ def test_http_predict_and_health(tmp_path):
     # missing field -> 422
     r = client.post("/v1/predict", json={"dense": [[0.0] * 13]})
     assert r.status_code == 422
+    # ragged / malformed payload -> 422, not 500
+    r = client.post("/v1/predict", json={"dense": [[0.0] * 13, [0.0]],
+                                         "sparse": [[1] * 4]})
+    assert r.status_code == 422
 
     r = client.post("/v1/reload")
     assert r.status_code == 200 and r.json()["reloaded"] is False
