@@ -15,6 +15,9 @@ import torch
 class PrefetchIterator:
     def __init__(self, dataset, depth: int = 2):
         self.ds = dataset
+        # datasets expose next_batch(); plain iterables are wrapped too
+        self._next = (dataset.next_batch if hasattr(dataset, "next_batch")
+                      else iter(dataset).__next__)
         self.q: queue.Queue = queue.Queue(maxsize=depth)
         self._stop = threading.Event()
         self.use_cuda = torch.cuda.is_available() and \
@@ -24,17 +27,23 @@ class PrefetchIterator:
         self.thread.start()
 
     def _produce(self):
-        return self.ds.next_batch()
+        return self._next()
 
     def _worker(self):
         while not self._stop.is_set():
-            if self.stream is not None:
-                with torch.cuda.stream(self.stream):
-                    batch = self._produce()
-                ev = torch.cuda.Event()
-                ev.record(self.stream)
-            else:
-                batch, ev = self._produce(), None
+            try:
+                if self.stream is not None:
+                    with torch.cuda.stream(self.stream):
+                        batch = self._produce()
+                    ev = torch.cuda.Event()
+                    ev.record(self.stream)
+                else:
+                    batch, ev = self._produce(), None
+            except BaseException as e:  # noqa: BLE001
+                # surface producer errors (incl. StopIteration) to the
+                # consumer instead of deadlocking its q.get()
+                self.q.put((e, "error"))
+                return
             try:
                 self.q.put((batch, ev), timeout=1.0)
             except queue.Full:
@@ -47,6 +56,8 @@ class PrefetchIterator:
 
     def __next__(self):
         batch, ev = self.q.get()
+        if ev == "error":
+            raise batch
         if ev is not None:
             torch.cuda.current_stream().wait_event(ev)
             for t in batch:

@@ -49,6 +49,15 @@ def build_argparser():
                    action="store_false")
     p.add_argument("--sharded", action="store_true",
                    help="embedding-parallel across ranks")
+    p.add_argument("--micro_batch", type=int, default=1,
+                   help="gradient-accumulation sub-batches per step")
+    p.add_argument("--storage", choices=["hbm", "hbm_dram"], default="hbm",
+                   help="EV storage tier (hbm_dram = host cold tier)")
+    p.add_argument("--hot_bytes", type=int, default=256 << 20,
+                   help="HBM hot-tier budget for --storage hbm_dram")
+    p.add_argument("--parquet", default=None,
+                   help="train from a parquet file (columns: label, "
+                        "dense_0..12, sparse_0..N) instead of synthetic")
     p.add_argument("--timeline", type=int, default=None)
     p.add_argument("--log_steps", type=int, default=50)
     p.add_argument("--seed", type=int, default=42)
@@ -57,6 +66,11 @@ def build_argparser():
 
 def make_ev_option(args) -> EmbeddingVariableOption:
     opt = EmbeddingVariableOption()
+    if args.storage == "hbm_dram":
+        from deeprec_amd.embedding.options import StorageOption, StorageType
+        opt.storage_option = StorageOption(
+            storage_type=StorageType.HBM_DRAM,
+            storage_size=[args.hot_bytes])
     if args.ev_filter == "counter":
         opt.filter_option = CounterFilter(filter_freq=args.filter_freq)
     elif args.ev_filter == "cbf":
@@ -67,6 +81,26 @@ def make_ev_option(args) -> EmbeddingVariableOption:
     elif args.ev_elimination == "l2":
         opt.evict_option = L2WeightEvict(l2_weight_threshold=0.01)
     return opt
+
+
+def _parquet_batches(path, batch_size, device, num_sparse):
+    """Adapt ParquetDataset rows to the (dense, ids, labels) batch shape."""
+    from deeprec_amd.data.parquet import ParquetDataset
+
+    class _Wrap:
+        def __iter__(self):
+            ds = ParquetDataset(path, batch_size, num_epochs=1 << 30)
+            for cols in ds:
+                dense = torch.stack(
+                    [cols[f"dense_{i}"].float() for i in range(13)],
+                    dim=1).to(device)
+                ids = torch.stack(
+                    [cols[f"sparse_{i}"].long()
+                     for i in range(num_sparse)], dim=1).to(device)
+                labels = cols["label"].float().to(device)
+                yield dense, ids, labels
+
+    return _Wrap()
 
 
 def main(argv=None):
@@ -104,9 +138,14 @@ def main(argv=None):
         broadcast_parameters(model.parameters())
         reducer = DenseGradAllreducer(model.parameters())
 
-    ds = CriteoSyntheticDataset(batch_size=args.batch_size, device=device,
-                                seed=args.seed, rank=rank,
-                                matrix_format=not is_seq)
+    if args.parquet and not is_seq:
+        from deeprec_amd.data.parquet import ParquetDataset
+        ds = _parquet_batches(args.parquet, args.batch_size, device,
+                              model.num_sparse)
+    else:
+        ds = CriteoSyntheticDataset(batch_size=args.batch_size,
+                                    device=device, seed=args.seed,
+                                    rank=rank, matrix_format=not is_seq)
     if args.smartstaged and not is_seq:
         from deeprec_amd.data.prefetch import PrefetchIterator
         batches = PrefetchIterator(ds, depth=2)
@@ -125,16 +164,25 @@ def main(argv=None):
     if args.timeline:
         hooks.append(ProfilerHook(args.timeline))
 
-    def step_fn():
+    def one_pass(scale=1.0):
         if is_seq:
             dense, ids, seq, target, labels = ds.next_seq_batch()
             logits = model(dense, ids[:, :model.num_sparse], seq, target)
         else:
             dense, ids, labels = next(it)
             logits = model(dense, ids)
-        loss = model.loss_fn(logits, labels)
-        opt.zero_grad()
+        loss = model.loss_fn(logits, labels) * scale
         loss.backward()
+        return loss
+
+    def step_fn():
+        opt.zero_grad()
+        if args.micro_batch > 1:
+            # reference MicroBatch: N accumulated sub-batches per apply
+            loss = sum(one_pass(1.0 / args.micro_batch)
+                       for _ in range(args.micro_batch))
+        else:
+            loss = one_pass()
         if reducer is not None:
             reducer.allreduce()
         opt.step()

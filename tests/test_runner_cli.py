@@ -15,3 +15,26 @@ def test_runner_with_filters_and_ckpt(tmp_path):
 def test_runner_sequence_model():
     main(["--model", "din", "--steps", "2", "--batch_size", "32",
           "--no_bf16", "--optimizer", "adagrad"])
+
+
+def test_runner_micro_batch_and_parquet(tmp_path):
+    import subprocess
+    import sys
+
+    import numpy as np
+    import pandas as pd
+
+    df = pd.DataFrame(
+        {"label": np.random.randint(0, 2, 128).astype("float32")})
+    for i in range(13):
+        df[f"dense_{i}"] = np.random.randn(128).astype("float32")
+    for i in range(26):
+        df[f"sparse_{i}"] = np.random.randint(0, 50, 128)
+    fn = str(tmp_path / "t.parquet")
+    df.to_parquet(fn)
+    r = subprocess.run(
+        [sys.executable, "-m", "deeprec_amd.models.runner", "--model",
+         "dlrm", "--steps", "2", "--batch_size", "32", "--no_bf16",
+         "--micro_batch", "2", "--parquet", fn],
+        capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0 and "RESULT" in r.stdout, r.stderr[-500:]
