@@ -427,3 +427,13 @@ def test_async_embedding_stage_cpu():
         kr, vr, _, _ = tr[list(tr)[list(te).index(n)]]
         torch.testing.assert_close(ve[torch.argsort(ke)],
                                    vr[torch.argsort(kr)])
+
+
+def test_fused_layer_norm_wrapper():
+    from deeprec_amd.ops.fused_norm import fused_layer_norm
+
+    x = torch.randn(6, 16)
+    w, b = torch.ones(16), torch.zeros(16)
+    torch.testing.assert_close(
+        fused_layer_norm(x, w, b),
+        torch.nn.functional.layer_norm(x, (16,), w, b))
