@@ -155,9 +155,17 @@ class Saver:
 
     def _owned_mask(self, ev, keys: torch.Tensor) -> torch.Tensor:
         if hasattr(ev, "world") and ev.world > 1:
-            from deeprec_amd.embedding.collection import KEY_BITS
-            raw = keys & ((1 << KEY_BITS) - 1)
-            return (raw % ev.world) == ev.rank
+            if _is_collection(ev):
+                # collections tag keys with table_id << KEY_BITS and route
+                # on the raw id — strip the tag before the mod
+                from deeprec_amd.embedding.collection import KEY_BITS
+                raw = keys & ((1 << KEY_BITS) - 1)
+                return (raw % ev.world) == ev.rank
+            # plain sharded EVs route lookups by the FULL key % world
+            # (sharded_embedding.py _ShardedPooledLookup); masking here
+            # would strand ids >= 2^48 or negative ids (64-bit feature
+            # hashing) on a rank that lookups never probe
+            return (keys % ev.world) == ev.rank
         return torch.ones(keys.numel(), dtype=torch.bool)
 
     def _restore_ev(self, ckpt_path: str, ev):

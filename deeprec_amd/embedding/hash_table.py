@@ -118,13 +118,15 @@ class DistributedHashTable:
         ]
 
     def lookup(self, keys: torch.Tensor, admit: bool = True):
-        out = torch.empty(keys.numel(), self.parts[0].shape[0],
-                          device=keys.device)
-        owner = keys % self.num_partitions
+        flat = keys.reshape(-1)
+        dev = self.parts[0].handle.device
+        out = torch.empty(flat.numel(), self.parts[0].shape[0], device=dev)
+        owner = flat % self.num_partitions
         for i, part in enumerate(self.parts):
             mask = owner == i
             if bool(mask.any()):
-                out[mask] = part.lookup(keys[mask], admit=admit)
+                rows = part.lookup(flat[mask], admit=admit)
+                out[mask] = rows.to(device=dev, dtype=out.dtype)
         return out.reshape(*keys.shape, -1)
 
     def size(self) -> int:

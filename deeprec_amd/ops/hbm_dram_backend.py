@@ -368,7 +368,18 @@ class HbmDramStorage(HbmStorage):
         m = keys.numel()
         if m == 0:
             return
-        counts = torch.ones(m, dtype=torch.int32, device=self.device)
+        # Admission must be unconditional for checkpointed rows: every
+        # (key, value) pair in a checkpoint was admitted at save time, so
+        # feed counts >= filter_freq (the true freqs when recorded) or the
+        # frequency filter would assign slot=-1 and the restored values
+        # would silently never be written (then re-initialized on next
+        # lookup — a full round trip losing trained rows).
+        if freqs is not None:
+            counts = freqs.to(self.device, torch.int32).clamp(
+                min=max(1, self.filter_freq))
+        else:
+            counts = torch.full((m,), max(1, self.filter_freq),
+                                dtype=torch.int32, device=self.device)
         slots = self.lookup_or_create(keys, counts, step=0, train=True)
         values = values.to(self.device, torch.float32)
         hot = (slots >= 0) & (slots < self.hot_rows)

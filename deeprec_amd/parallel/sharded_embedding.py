@@ -39,6 +39,12 @@ class ShardedEmbeddingVariable:
 
     # delegation so optimizers/savers treat this like an EV
     @property
+    def local(self):
+        """The rank-local EV (Saver and hooks expect `.local` on every
+        sharded wrapper, like ShardedEmbeddingCollection)."""
+        return self.local_ev
+
+    @property
     def device(self):
         return self.local_ev.device
 
@@ -142,12 +148,39 @@ class _ShardedPooledLookup(torch.autograd.Function):
         return (torch.zeros_like(sev.local_ev._anchor),) + (None,) * 9
 
 
+def _sharded_lookup_infer(sev: ShardedEmbeddingVariable, uniq, inverse,
+                          offsets, row_ids, combiner, weights, out_dtype):
+    """Inference path: no insert, no freq/version bump, no autograd —
+    mirrors ShardedEmbeddingCollection._lookup_cat's train=False branch."""
+    w = sev.world
+    owner = (uniq % w).to(torch.int64)
+    order = torch.argsort(owner, stable=True)
+    send_keys = uniq[order]
+    send_splits = torch.bincount(owner, minlength=w)
+    recv_splits = comm.exchange_counts(send_splits)
+    in_sp, out_sp = send_splits.tolist(), recv_splits.tolist()
+    recv_keys = comm.all_to_all_single(send_keys, in_sp, out_sp)
+    uniq2, inv2 = torch.unique(recv_keys, return_inverse=True)
+    slots2 = sev.local_ev.storage.lookup(uniq2)   # read-only probe
+    emb2 = sev.local_ev.storage.gather(uniq2, slots2)
+    emb_out = emb2[inv2]
+    emb_back = comm.all_to_all_single(emb_out.contiguous(), out_sp, in_sp)
+    emb_uniq = torch.empty_like(emb_back)
+    emb_uniq[order] = emb_back
+    return F.pooled_forward(emb_uniq, inverse, offsets, row_ids, combiner,
+                            weights, out_dtype)
+
+
 def sharded_embedding_lookup_sparse(sev: ShardedEmbeddingVariable,
                                     sp_ids: RaggedIds, combiner="mean",
                                     out_dtype=None, train=True):
     uniq, inverse, counts = torch.unique(
         sp_ids.values, return_inverse=True, return_counts=True)
     row_ids = sp_ids.row_ids()
+    if not (train and sev.trainable):
+        return _sharded_lookup_infer(
+            sev, uniq, inverse.to(torch.int32), sp_ids.offsets, row_ids,
+            combiner, sp_ids.weights, out_dtype)
     return _ShardedPooledLookup.apply(
         sev._anchor, sev, uniq, counts, inverse.to(torch.int32),
         sp_ids.offsets, row_ids, combiner, sp_ids.weights, out_dtype)

@@ -322,6 +322,10 @@ def test_hash_table_api():
     assert out.shape == (10, 4) and dht.size() == 10
     # routing is stable: same keys hit the same partitions
     torch.testing.assert_close(out, dht.lookup(dkeys, admit=False))
+    # multi-dimensional key tensors (regression: ADVICE r1 #5)
+    out2d = dht.lookup(dkeys.reshape(2, 5), admit=False)
+    assert out2d.shape == (2, 5, 4)
+    torch.testing.assert_close(out2d.reshape(10, 4), out)
 
 
 def test_rebalance_hook_noop_on_single_tier():

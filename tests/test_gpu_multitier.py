@@ -155,3 +155,46 @@ def test_lfu_rebalance_promotes_hot_keys():
     # frequencies preserved too
     f = st.frequencies(keys)
     assert int(f[64:].min()) > int(f[:64].max())
+
+
+def test_multitier_filter_restore_roundtrip():
+    """Checkpoint round trip with a frequency admission filter: restored
+    rows must come back trained, not re-initialized (regression: ADVICE
+    r1 high — import_ routed keys through admission with counts=1)."""
+    from deeprec_amd.embedding.options import CounterFilter
+    from deeprec_amd.ops.hbm_dram_backend import HbmDramStorage
+
+    dim, hot_rows = 8, 16
+    opt = EmbeddingVariableOption(
+        storage_option=StorageOption(storage_type=StorageType.HBM_DRAM,
+                                     storage_size=[hot_rows * dim * 4]),
+        filter_option=CounterFilter(filter_freq=3),
+        init_option=InitializerOption(initializer=1.0))
+    st = HbmDramStorage(dim, opt, device=DEV)
+    keys = torch.arange(64, dtype=torch.int64, device=DEV)
+    # admit everything (3 sightings), spilling past the 16-row hot tier
+    for _ in range(3):
+        st.lookup_or_create(keys, torch.ones(64, dtype=torch.int32,
+                                             device=DEV), step=0)
+    slots = st.lookup(keys)
+    assert bool((slots >= 0).all()) and bool((slots >= hot_rows).any())
+    # "train": write recognizable values into every row
+    trained = torch.arange(64, dtype=torch.float32,
+                           device=DEV)[:, None].repeat(1, dim) + 100.0
+    hot = slots < hot_rows
+    st.values[slots[hot].long()] = trained[hot]
+    cold_idx = (slots[~hot].cpu().long() - hot_rows)
+    st.values_cold[cold_idx] = trained[~hot].cpu()
+    k, v, f, ver = st.export()
+    order = torch.argsort(k)
+    # restore into a FRESH storage with the same filter
+    st2 = HbmDramStorage(dim, opt, device=DEV)
+    st2.import_(k, v, f.to(torch.int32), ver)
+    slots2 = st2.lookup(keys)
+    assert bool((slots2 >= 0).all()), \
+        "restored keys must be admitted regardless of the filter"
+    got = st2.materialize(keys, slots2)
+    torch.testing.assert_close(got, trained)
+    # frequencies survive the round trip
+    torch.testing.assert_close(st2.frequencies(keys).cpu(),
+                               st.frequencies(keys).cpu())
