@@ -30,14 +30,23 @@ def main():
     p.add_argument("--no-bf16", action="store_true")
     p.add_argument("--hip-graph", action="store_true", default=True,
                    help="capture the steady-state step in a hipGraph "
-                        "(single-GPU only; replay eliminates launch gaps)")
+                        "(replay eliminates launch gaps; the distributed "
+                        "step captures too via the padded all-to-all)")
     p.add_argument("--no-hip-graph", dest="hip_graph", action="store_false")
+    p.add_argument("--force-dist", action="store_true",
+                   help="run the distributed (sharded + RCCL) code path "
+                        "even at WORLD_SIZE=1 — exercises the exact "
+                        "nccl/RCCL call pattern of a full node on a "
+                        "single-GPU lease")
     args = p.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
-    distributed = world_size > 1
+    distributed = world_size > 1 or args.force_dist
+    if args.force_dist and world_size == 1:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29561")
 
     if args.device:
         device = torch.device(args.device)
@@ -60,14 +69,15 @@ def main():
 
     torch.manual_seed(42 + rank)
     bf16 = (not args.no_bf16) and device.type == "cuda"
-    use_graph = (args.hip_graph and not distributed
-                 and device.type == "cuda"
+    use_graph = (args.hip_graph and device.type == "cuda"
                  and args.optimizer == "adamasync")
     ev_option = None
     if use_graph:
         from deeprec_amd.embedding.options import EmbeddingVariableOption
         # no growth can happen inside a captured step: pre-size for the
-        # full synthetic id space
+        # full synthetic id space (distributed ranks also hold slot-less
+        # dedup entries for every key they SEE, so entries scale with the
+        # full space regardless of world size)
         ev_option = EmbeddingVariableOption(init_capacity=1 << 23)
     model = DLRM(device=device, bf16=bf16, sharded=distributed,
                  ev_option=ev_option)
@@ -113,19 +123,41 @@ def main():
     one_step = eager_step
     if use_graph:
       try:
-        st = model.collection.storage
-        st.enable_graph_mode(expected_entries=1 << 23,
-                             expected_slots=1 << 23)
-        # optimizer slabs must exist before capture
-        st.get_slab("adam_m", model.collection.dim, 0.0)
-        st.get_slab("adam_v", model.collection.dim, 0.0)
-        model.collection.graph_mode = True
+        coll = model.collection
+        st = coll.storage
         # capture needs warmed state (optimizer slabs, device beta powers,
-        # dense adam state) — always run at least 2 eager steps
-        for i in range(max(args.warmup, 2)):
-            eager_step(i)
-        torch.cuda.synchronize()
-        # static input buffers + whole-step capture
+        # dense adam state) — always run at least 2 eager steps. For the
+        # distributed path the eager steps also OBSERVE the per-peer
+        # split sizes that size the padded exchange.
+        if distributed:
+            for i in range(max(args.warmup, 2)):
+                eager_step(i)
+            torch.cuda.synchronize()
+            # pad cap: worst observed split + 50% headroom (overflow trips
+            # engine error 4 and fails loudly after the run)
+            cap = int(1.5 * max(coll.observed_max_split(), 1)) + 256
+            coll.enable_graph_mode(expected_entries=1 << 23,
+                                   expected_slots=1 << 23, pad_cap=cap)
+            st = coll.storage
+            st.get_slab("adam_m", coll.dim, 0.0)
+            st.get_slab("adam_v", coll.dim, 0.0)
+            # one eager padded-protocol step so RCCL communicators for the
+            # static splits exist before capture
+            eager_step(0)
+            torch.cuda.synchronize()
+        else:
+            st.enable_graph_mode(expected_entries=1 << 23,
+                                 expected_slots=1 << 23)
+            # optimizer slabs must exist before capture
+            st.get_slab("adam_m", coll.dim, 0.0)
+            st.get_slab("adam_v", coll.dim, 0.0)
+            coll.graph_mode = True
+            for i in range(max(args.warmup, 2)):
+                eager_step(i)
+            torch.cuda.synchronize()
+        # static input buffers + whole-step capture (the distributed step
+        # records its RCCL all-to-alls/all-reduce into the same graph —
+        # shapes are static via the padded exchange)
         sdense, sids, slabels = (t.clone() for t in batches[0])
         graph = torch.cuda.CUDAGraph()
         opt.zero_grad()
@@ -133,6 +165,8 @@ def main():
             logits = model(sdense, sids)
             loss = model.loss_fn(logits, slabels)
             loss.backward()
+            if reducer is not None:
+                reducer.allreduce(async_op=True)
             opt.step()
         torch.cuda.synchronize()
 
@@ -151,6 +185,9 @@ def main():
               flush=True)
         use_graph = False
         model.collection.graph_mode = False
+        if distributed:
+            model.collection._pad_cap = None  # back to the dynamic path
+            model.collection.local.graph_mode = False
         for i in range(args.warmup):
             eager_step(i)
     else:

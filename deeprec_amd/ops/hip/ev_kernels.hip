@@ -391,6 +391,92 @@ __global__ void k_zero_f32(float* __restrict__ p, int64_t n) {
     for (int64_t j = n & ~3LL; j < n; ++j) p[j] = 0.0f;
 }
 
+// ---------------------------------------------------------------------
+// padded peer routing — the captured distributed exchange
+// ---------------------------------------------------------------------
+// The sharded collection's all-to-all must have static shapes to live
+// inside a hipGraph (RCCL collectives capture, but only with fixed
+// splits). Every peer slot is padded to `cap` rows; pad keys are
+// PAD_KEY, which the owner deduplicates into one (ignored) entry, so no
+// count exchange is needed on the wire at all (≙ the reference's
+// two-phase count+payload protocol, all2all_input_dispatcher.cu:250-280,
+// re-designed shape-static for graph replay).
+
+static constexpr int64_t PAD_KEY = INT64_MAX;
+
+// Bump only the dedup epoch (NOT the step): the distributed step runs
+// TWO dedups per training step against the same table (requester-side
+// claim-only dedup, then owner-side dedup of the padded exchange), and
+// version/eviction semantics must advance once per step.
+__global__ void k_bump_epoch_only(int32_t* __restrict__ epoch_dev) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) epoch_dev[0] += 1;
+}
+
+// Reset the padded send layout: keys = PAD_KEY, counts = 0, cursors = 0.
+__global__ void k_route_fill(int64_t* __restrict__ send_keys,
+                             int32_t* __restrict__ send_cnt,
+                             int32_t* __restrict__ peer_cursor, int total,
+                             int world) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < total; i += stride) {
+    send_keys[i] = PAD_KEY;
+    send_cnt[i] = 0;
+    if (i < world) peer_cursor[i] = 0;
+  }
+}
+
+// Scatter the step's unique keys into per-peer padded blocks.
+// owner = (key & key_mask) % world (the raw id routes; composite table
+// tags are stripped). route_pos[u] remembers where unique u landed so
+// the returned embedding rows / outgoing grad rows can be addressed
+// without any host-side ordering. Overflowing a peer's cap sets
+// error_flag = 4 (host must re-capture with a larger cap).
+__global__ void k_route_pad(
+    const int64_t* __restrict__ uniq_keys, const int32_t* __restrict__ counts,
+    const int32_t* __restrict__ m_dev, int n_cap, int world, int cap,
+    int key_bits, int64_t* __restrict__ send_keys,
+    int32_t* __restrict__ send_cnt, int32_t* __restrict__ route_pos,
+    int32_t* __restrict__ peer_cursor, int32_t* __restrict__ error_flag) {
+  const int m = min(*m_dev, n_cap);
+  const int64_t key_mask =
+      key_bits > 0 ? (((int64_t)1 << key_bits) - 1) : ~(int64_t)0;
+  int64_t u = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; u < m; u += stride) {
+    const int64_t key = uniq_keys[u];
+    int owner = (int)((uint64_t)(key & key_mask) % (uint64_t)world);
+    int pos = atomicAdd(&peer_cursor[owner], 1);
+    if (pos >= cap) {
+      atomicExch(error_flag, 4);
+      route_pos[u] = owner * cap;  // in-bounds dummy; run is poisoned
+      continue;
+    }
+    int64_t o = (int64_t)owner * cap + pos;
+    send_keys[o] = key;
+    send_cnt[o] = counts[u];
+    route_pos[u] = (int32_t)o;
+  }
+}
+
+// dst[route_pos[u]] = src[u] for u < m (grad rows into the padded wire
+// layout; dst is pre-zeroed so pad rows carry zero gradient).
+__global__ void k_rows_to_padded(const float* __restrict__ src,
+                                 const int32_t* __restrict__ route_pos,
+                                 const int32_t* __restrict__ m_dev,
+                                 int n_cap, int dim,
+                                 float* __restrict__ dst) {
+  const int m = min(*m_dev, n_cap);
+  int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)m * dim;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; t < total; t += stride) {
+    int u = (int)(t / dim);
+    int d = (int)(t % dim);
+    dst[(int64_t)route_pos[u] * dim + d] = src[t];
+  }
+}
+
 // Pass C (per occurrence): inverse + per-batch counts.
 __global__ void k_dedup_pass_c(
     const int64_t* __restrict__ keys, int nnz,
@@ -1013,6 +1099,49 @@ void bump_epoch(torch::Tensor epoch_dev, torch::Tensor step_dev) {
       epoch_dev.data_ptr<int32_t>(), step_dev.data_ptr<int64_t>());
 }
 
+void bump_epoch_only(torch::Tensor epoch_dev) {
+  k_bump_epoch_only<<<1, 64, 0, current_stream()>>>(
+      epoch_dev.data_ptr<int32_t>());
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> route_pad(
+    torch::Tensor uniq_keys, torch::Tensor counts, torch::Tensor m_dev,
+    int64_t world, int64_t cap, int64_t key_bits,
+    torch::Tensor error_flag) {
+  int n_cap = uniq_keys.numel();
+  int total = (int)(world * cap);
+  auto opts_i32 = counts.options();
+  auto send_keys = torch::empty({(int64_t)total}, uniq_keys.options());
+  auto send_cnt = torch::empty({(int64_t)total}, opts_i32);
+  auto route_pos = torch::empty({(int64_t)n_cap}, opts_i32);
+  auto cursor = torch::empty({world}, opts_i32);
+  auto stream = current_stream();
+  k_route_fill<<<n_blocks(total), kBlock, 0, stream>>>(
+      send_keys.data_ptr<int64_t>(), send_cnt.data_ptr<int32_t>(),
+      cursor.data_ptr<int32_t>(), total, (int)world);
+  k_route_pad<<<n_blocks(n_cap), kBlock, 0, stream>>>(
+      uniq_keys.data_ptr<int64_t>(), counts.data_ptr<int32_t>(),
+      m_dev.data_ptr<int32_t>(), n_cap, (int)world, (int)cap, (int)key_bits,
+      send_keys.data_ptr<int64_t>(), send_cnt.data_ptr<int32_t>(),
+      route_pos.data_ptr<int32_t>(), cursor.data_ptr<int32_t>(),
+      error_flag.data_ptr<int32_t>());
+  return {send_keys, send_cnt, route_pos};
+}
+
+torch::Tensor rows_to_padded(torch::Tensor src, torch::Tensor route_pos,
+                             torch::Tensor m_dev, int64_t out_rows) {
+  int n_cap = route_pos.numel();
+  int dim = src.size(1);
+  auto dst = torch::empty({out_rows, (int64_t)dim}, src.options());
+  auto stream = current_stream();
+  k_zero_f32<<<n_blocks((out_rows * dim + 3) / 4), kBlock, 0, stream>>>(
+      dst.data_ptr<float>(), out_rows * dim);
+  k_rows_to_padded<<<n_blocks((int64_t)n_cap * dim), kBlock, 0, stream>>>(
+      src.data_ptr<float>(), route_pos.data_ptr<int32_t>(),
+      m_dev.data_ptr<int32_t>(), n_cap, dim, dst.data_ptr<float>());
+  return dst;
+}
+
 torch::Tensor ht_dedup_a_dev(
     torch::Tensor keys, torch::Tensor ht_keys, torch::Tensor ht_freq,
     torch::Tensor ht_version, torch::Tensor ht_epoch,
@@ -1536,6 +1665,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("ht_dedup_b_padded", &ht_dedup_b_padded);
   mod.def("ht_dedup_a_dev", &ht_dedup_a_dev);
   mod.def("bump_epoch", &bump_epoch);
+  mod.def("bump_epoch_only", &bump_epoch_only);
+  mod.def("route_pad", &route_pad);
+  mod.def("rows_to_padded", &rows_to_padded);
   mod.def("ht_dedup_c", &ht_dedup_c);
   mod.def("csr_order", &csr_order);
   mod.def("csr_scatter", &csr_scatter);
@@ -1558,4 +1690,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("apply_rmsprop", &apply_rmsprop);
   mod.def("apply_ftrl", &apply_ftrl);
   mod.attr("EMPTY_KEY") = EMPTY_KEY;
+  mod.attr("PAD_KEY") = INT64_MAX;
 }

@@ -163,16 +163,24 @@ class HbmStorage:
 
     def _export_entries(self):
         n = int(self.entry_counter.cpu())
-        return self.ext.ht_export(self.ht_keys, self.ht_slot, self.ht_freq,
-                                  self.ht_version, n)
+        keys, slots, freqs, versions = self.ext.ht_export(
+            self.ht_keys, self.ht_slot, self.ht_freq, self.ht_version, n)
+        # the padded sharded exchange admits one PAD_KEY sentinel entry
+        # (wire padding); it is engine-internal and never exported
+        if n and bool((keys == self.ext.PAD_KEY).any()):
+            keep = keys != self.ext.PAD_KEY
+            keys, slots = keys[keep], slots[keep]
+            freqs, versions = freqs[keep], versions[keep]
+        return keys, slots, freqs, versions
 
     def _check_error(self):
         err = int(self.error_flag.cpu())
         if err != 0:
             raise RuntimeError(
                 f"HBM hash table error {err} (1=slab overflow, "
-                "2=table full, 3=CSR scatter out of bounds)"
-                " — engine invariant violated")
+                "2=table full, 3=CSR scatter out of bounds, "
+                "4=padded all-to-all peer cap overflow — re-capture with "
+                "a larger pad_cap) — engine invariant violated")
 
     # ---------------- public interface ----------------
     def get_slab(self, name, width, init_value, dtype=torch.float32):
@@ -243,6 +251,66 @@ class HbmStorage:
             self.default_values, self.dvd_per_table, self.key_bits,
             self._init_limit(), self.filter_freq, self.error_flag)
         return uniq_buf, inverse, counts, slots
+
+    def dedup_only_capture(self, values_cat: torch.Tensor):
+        """Requester-side dedup for the padded sharded exchange: claim-only
+        pass A + pass C against the local table (NO admission, NO
+        freq/version writes — those happen once, on the owner). Non-owned
+        keys become slot-less entries here (the table doubles as the dedup
+        structure); they are cheap (~32 B) and never exported.
+
+        Returns (uniq_buf [nnz] padded, inverse i32 [nnz], counts i32
+        [nnz] padded, m_counter device scalar). Capture-safe: no host
+        syncs, all shapes static."""
+        nnz = values_cat.numel()
+        self.ext.bump_epoch(self._epoch_dev, self._step_dev)
+        uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
+        self._last_m_dev = m_counter
+        self.ext.ht_dedup_a_dev(values_cat, self.ht_keys, self.ht_freq,
+                                self.ht_version, self.ht_epoch,
+                                self.ht_compact, self._epoch_dev,
+                                self._step_dev, self.entry_counter,
+                                m_counter, uniq_buf, centry_buf,
+                                self.error_flag)
+        inverse, counts, rank = self.ext.ht_dedup_c(
+            values_cat, self.ht_keys, self.ht_compact, nnz)
+        self._last_rank = rank
+        return uniq_buf, inverse, counts, m_counter
+
+    def dedup_lookup_capture_owner(self, recv_keys: torch.Tensor,
+                                   recv_cnt: torch.Tensor):
+        """Owner-side dedup+admission for the padded sharded exchange:
+        full pipeline (A, C, B) over the w*cap received keys, except the
+        freq bump uses the SUMMED true per-batch counts carried on the
+        wire (a key arriving from several peers must count all of its
+        occurrences, not the number of peers). PAD_KEY pads dedup into
+        one engine-internal entry. Uses its own epoch bump (epoch-only:
+        the step already advanced in the requester pass)."""
+        nnz = recv_keys.numel()
+        self.ext.bump_epoch_only(self._epoch_dev)
+        uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
+        self.ext.ht_dedup_a_dev(recv_keys, self.ht_keys, self.ht_freq,
+                                self.ht_version, self.ht_epoch,
+                                self.ht_compact, self._epoch_dev,
+                                self._step_dev, self.entry_counter,
+                                m_counter, uniq_buf, centry_buf,
+                                self.error_flag)
+        inverse, _counts_c, _rank = self.ext.ht_dedup_c(
+            recv_keys, self.ht_keys, self.ht_compact, nnz)
+        inv_l = inverse.long()
+        cnt_sum = torch.zeros(nnz, dtype=torch.int32, device=self.device)
+        cnt_sum.index_add_(0, inv_l, recv_cnt)
+        slots = self.ext.ht_dedup_b_padded(
+            centry_buf, uniq_buf, m_counter, self.ht_slot, self.ht_freq,
+            self.ht_version, cnt_sum, self._step_dev,
+            self.slot_counter, self.max_slots, self.values,
+            self.default_values, self.dvd_per_table, self.key_bits,
+            self._init_limit(), self.filter_freq, self.error_flag)
+        return uniq_buf, inv_l, slots
 
     def prefers_dedup(self) -> bool:
         return self._uniq_ratio is None or self._uniq_ratio < 0.5

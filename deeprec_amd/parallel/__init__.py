@@ -56,8 +56,11 @@ class DenseGradAllreducer:
         launched without waiting — call wait() (or set it as the
         optimizer's pre_dense_step) before the dense optimizer step, so
         the all-reduce overlaps with the sparse applies."""
-        if not is_initialized() or world_size() == 1:
+        if not is_initialized():
             return
+        # world=1 still reduces: a single-GPU lease then exercises the
+        # exact RCCL call pattern (incl. under hipGraph capture) that a
+        # full node will run
         w = world_size()
         bucket, nbytes = [], 0
         for p in self.params:
@@ -97,7 +100,7 @@ class DenseGradAllreducer:
 def broadcast_parameters(params: Iterable[torch.Tensor], src: int = 0):
     """Sync initial dense weights (reference: hvd broadcast hook,
     hvd_strategy.py:472)."""
-    if not is_initialized() or world_size() == 1:
+    if not is_initialized():
         return
     for p in params:
         dist.broadcast(p.detach(), src=src)

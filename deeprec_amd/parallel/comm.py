@@ -38,8 +38,11 @@ def all_to_all_single(inp: torch.Tensor, in_splits, out_splits) -> torch.Tensor:
     """Variable-size all-to-all along dim 0. in_splits/out_splits: python
     lists of per-peer element counts (dim-0 rows)."""
     w = world_size()
-    if w == 1:
+    if w == 1 and not is_initialized():
         return inp.clone()
+    # NOTE: world=1 with an initialized group still goes through the
+    # backend (self-exchange) so the RCCL code path is exercised on a
+    # single-GPU lease exactly as it will run at world=8
     out_shape = (sum(out_splits),) + tuple(inp.shape[1:])
     out = torch.empty(out_shape, dtype=inp.dtype, device=inp.device)
     if _supports_all_to_all():
@@ -79,7 +82,7 @@ def exchange_counts(counts: torch.Tensor) -> torch.Tensor:
     arriving from each peer. The two-phase count-then-payload exchange of
     the reference's all-to-all dispatcher."""
     w = world_size()
-    if w == 1:
+    if w == 1 and not is_initialized():
         return counts.clone()
     return all_to_all_single(counts.contiguous(), [1] * w, [1] * w)
 
@@ -99,7 +102,7 @@ def padded_all_to_all(inp: torch.Tensor, in_splits, cap: int):
     w = world_size()
     if max(in_splits, default=0) > cap:
         raise ValueError(f"split {max(in_splits)} exceeds pad cap {cap}")
-    if w == 1:
+    if w == 1 and not is_initialized():
         out = torch.zeros((cap,) + tuple(inp.shape[1:]), dtype=inp.dtype,
                           device=inp.device)
         out[: inp.shape[0]] = inp

@@ -52,6 +52,9 @@ class ShardedEmbeddingCollection:
         self.name = name
         self.dim = embedding_dim
         self.n_tables = self.local.n_tables
+        # static-shape (padded) exchange mode — see static_mode()
+        self._pad_cap = None
+        self._observed_max_split = 0
 
     # ---- optimizer-facing delegation ----
     @property
@@ -134,10 +137,14 @@ class ShardedEmbeddingCollection:
     def _lookup_cat(self, values_cat, offsets_cat, row_ids_cat, row_coeff,
                     weights_cat, batch, out_dtype, train):
         coll = self.local
+        train = train and coll.trainable
+        if train and self._pad_cap is not None:
+            return _padded_sharded_lookup(
+                self, values_cat, offsets_cat, row_ids_cat, row_coeff,
+                weights_cat, batch, out_dtype)
         uniq, inverse, counts = torch.unique(
             values_cat, return_inverse=True, return_counts=True)
         inverse = inverse.to(torch.int32)
-        train = train and coll.trainable
         if not train:
             emb = _exchange_lookup(self, uniq, counts, train=False)[0]
             return coll._forward(uniq, None, inverse, offsets_cat,
@@ -149,6 +156,195 @@ class ShardedEmbeddingCollection:
             coll._anchor, self, uniq, counts, inverse, offsets_cat,
             row_ids_cat, order, bounds, chunk_u, chunk_k0, row_coeff,
             weights_cat, batch, out_dtype)
+
+    # ---- static-shape (padded) exchange — hipGraph-capturable ----
+    def static_mode(self, pad_cap: int):
+        """Switch the training exchange to the fixed-shape padded
+        all-to-all: every peer slot padded to pad_cap rows, pad keys =
+        PAD_KEY, no count exchange on the wire. All tensor shapes become
+        data-independent, which is what hipGraph capture (and RCCL
+        capture) requires. Works eagerly too (gloo/CPU included) — the
+        wire protocol is identical; only the local dedup engine differs.
+        pad_cap must exceed the per-peer unique-key count of every step
+        (observed_max_split() after eager warmup + slack is the intended
+        sizing); overflow raises via the engine error flag on GPU and
+        ValueError on CPU."""
+        self._pad_cap = int(pad_cap)
+
+    def observed_max_split(self) -> int:
+        """Largest per-peer split seen by the eager exchange (cap sizing
+        aid for static_mode)."""
+        return self._observed_max_split
+
+    def enable_graph_mode(self, expected_entries: int, expected_slots: int,
+                          pad_cap: int):
+        """Prepare the full distributed step for hipGraph capture:
+        pre-size the local table (requester-side dedup inserts slot-less
+        entries for EVERY key this rank sees, so size for the whole id
+        space, not the owned share), device-resident epoch/step, and the
+        padded exchange."""
+        self.local.storage.enable_graph_mode(expected_entries,
+                                             expected_slots)
+        self.local.graph_mode = True
+        self.static_mode(pad_cap)
+
+
+PAD_KEY = (1 << 63) - 1  # == ext.PAD_KEY (engine wire-padding sentinel)
+
+
+def _route_pad_cpu(uniq, counts, world, cap, key_bits):
+    """CPU/gloo reference of the k_route_pad kernel: scatter unique keys
+    into per-peer blocks of `cap` rows, pad with PAD_KEY. Returns
+    (send_keys [w*cap], send_cnt [w*cap] i32, route_pos [m] i64)."""
+    m = uniq.numel()
+    raw = uniq & ((1 << key_bits) - 1)
+    owner = (raw % world).to(torch.int64)
+    send_splits = torch.bincount(owner, minlength=world)
+    if m and int(send_splits.max()) > cap:
+        raise ValueError(
+            f"padded exchange overflow: peer split {int(send_splits.max())}"
+            f" exceeds pad cap {cap}")
+    order = torch.argsort(owner, stable=True)
+    group_start = torch.zeros(world, dtype=torch.int64)
+    group_start[1:] = send_splits.cumsum(0)[:-1]
+    sorted_owner = owner[order]
+    rank_in_group = torch.arange(m, dtype=torch.int64) \
+        - group_start[sorted_owner]
+    dest = sorted_owner * cap + rank_in_group
+    send_keys = torch.full((world * cap,), PAD_KEY, dtype=torch.int64)
+    send_cnt = torch.zeros(world * cap, dtype=torch.int32)
+    send_keys[dest] = uniq[order]
+    send_cnt[dest] = counts[order].to(torch.int32)
+    route_pos = torch.empty(m, dtype=torch.int64)
+    route_pos[order] = dest
+    return send_keys, send_cnt, route_pos
+
+
+def _padded_sharded_lookup(sev, values_cat, offsets_cat, row_ids_cat,
+                           row_coeff, weights_cat, batch, out_dtype):
+    """Fixed-shape training lookup: requester dedup -> padded all-to-all
+    of (keys, counts) -> owner dedup+admission -> padded all-to-all of
+    rows back -> fused pooling. Every tensor shape is independent of the
+    data, so the WHOLE sequence (RCCL collectives included) records into
+    a hipGraph and replays. Eager execution uses the identical wire
+    protocol (gloo/CPU runs it in tests)."""
+    coll = sev.local
+    st = coll.storage
+    w, cap = sev.world, sev._pad_cap
+    dev = coll.device
+    is_gpu = dev.type == "cuda"
+    splits = [cap] * w
+    if is_gpu:
+        uniq_buf, inverse, counts, m_dev = st.dedup_only_capture(values_cat)
+        send_keys, send_cnt, route_pos = st.ext.route_pad(
+            uniq_buf, counts, m_dev, w, cap, KEY_BITS, st.error_flag)
+        # CSR prep consumes pass C's rank before anything else overwrites
+        order, bounds, _, _ = coll._prep_backward(inverse, counts)
+        m_req = values_cat.numel()
+    else:
+        uniq, inv_t, cnt_t = torch.unique(
+            values_cat, return_inverse=True, return_counts=True)
+        inverse = inv_t.to(torch.int32)
+        st._last_rank = None
+        order, bounds, _, _ = coll._prep_backward(inverse, cnt_t)
+        send_keys, send_cnt, route_pos = _route_pad_cpu(
+            uniq, cnt_t, w, cap, KEY_BITS)
+        m_dev = None
+        m_req = uniq.numel()
+    recv_keys = comm.all_to_all_single(send_keys, splits, splits)
+    recv_cnt = comm.all_to_all_single(send_cnt, splits, splits)
+    if is_gpu:
+        uniq2, inv2, slots2 = st.dedup_lookup_capture_owner(recv_keys,
+                                                            recv_cnt)
+        slots_elem = slots2.index_select(0, inv2)
+        wire = sev.comm_dtype or torch.float32
+        reply = st.ext.ev_gather(
+            st.values, st.default_values, recv_keys, slots_elem,
+            st._no_permission_value(), st._use_no_permission(), wire)
+        own_valid = None
+        emb_rows = comm.all_to_all_single(reply, splits, splits).float()
+        inverse_final = route_pos.index_select(0, inverse.long())
+        keys_for_kernel = recv_keys
+    else:
+        uniq2, inv2 = torch.unique(recv_keys, return_inverse=True)
+        cnt2 = torch.zeros(uniq2.numel(), dtype=torch.int64)
+        cnt2.index_add_(0, inv2, recv_cnt.long())
+        own_valid = uniq2 != PAD_KEY
+        slots_v = st.lookup_or_create(uniq2[own_valid], cnt2[own_valid],
+                                      get_global_step(), train=True)
+        emb_v = st.gather(uniq2[own_valid], slots_v)
+        reply = torch.zeros(uniq2.numel(), coll.dim)
+        reply[own_valid] = emb_v
+        reply = reply[inv2]
+        slots2 = slots_v
+        if sev.comm_dtype is not None:
+            lp = reply.to(sev.comm_dtype).contiguous()
+            emb_rows = comm.all_to_all_single(
+                lp.view(torch.int16), splits, splits) \
+                .view(sev.comm_dtype).float()
+        else:
+            emb_rows = comm.all_to_all_single(reply.contiguous(), splits,
+                                              splits)
+        inverse_final = route_pos[inverse.long()].to(torch.int32)
+        keys_for_kernel = uniq2
+    if is_gpu:
+        inverse_final = inverse_final.to(torch.int32)
+    return _PaddedShardedLookup.apply(
+        coll._anchor, sev, emb_rows, keys_for_kernel, inverse_final,
+        offsets_cat, row_ids_cat, order, bounds, row_coeff, weights_cat,
+        batch, out_dtype, m_req, m_dev, route_pos, inv2, slots2, uniq2,
+        own_valid)
+
+
+class _PaddedShardedLookup(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, anchor, sev, emb_rows, keys_for_kernel, inverse_final,
+                offsets_cat, row_ids_cat, order, bounds, row_coeff,
+                weights_cat, batch, out_dtype, m_req, m_dev, route_pos,
+                inv2, slots2, uniq2, own_valid):
+        out = sev.local._forward(keys_for_kernel, None, inverse_final,
+                                 offsets_cat, weights_cat, batch, out_dtype,
+                                 emb_override=emb_rows)
+        ctx.sev = sev
+        ctx.batch = batch
+        ctx.weights_cat = weights_cat
+        ctx.m_req = m_req
+        ctx.m_dev = m_dev
+        ctx.own_valid = own_valid
+        ctx.save_for_backward(order, bounds, row_ids_cat, row_coeff,
+                              route_pos, inv2, slots2, uniq2)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        sev = ctx.sev
+        coll = sev.local
+        st = coll.storage
+        w, cap = sev.world, sev._pad_cap
+        splits = [cap] * w
+        order, bounds, row_ids_cat, row_coeff, route_pos, inv2, slots2, \
+            uniq2 = ctx.saved_tensors
+        grad_unique = coll._backward(grad_out, order, bounds, None, None,
+                                     row_ids_cat, ctx.weights_cat,
+                                     row_coeff, ctx.m_req, ctx.batch)
+        if coll.device.type == "cuda":
+            grad_send = st.ext.rows_to_padded(grad_unique, route_pos,
+                                              ctx.m_dev, w * cap)
+            grad_recv = comm.all_to_all_single(grad_send, splits, splits)
+            grad2 = torch.zeros(w * cap, grad_recv.shape[1],
+                                device=grad_recv.device,
+                                dtype=grad_recv.dtype)
+            grad2.index_add_(0, inv2, grad_recv)
+            coll.accumulate_grad(slots2, uniq2, grad2)
+        else:
+            grad_send = torch.zeros(w * cap, grad_unique.shape[1])
+            grad_send[route_pos] = grad_unique
+            grad_recv = comm.all_to_all_single(grad_send, splits, splits)
+            grad2 = torch.zeros(uniq2.numel(), grad_recv.shape[1])
+            grad2.index_add_(0, inv2, grad_recv)
+            valid = ctx.own_valid
+            coll.accumulate_grad(slots2, uniq2[valid], grad2[valid])
+        return (torch.zeros_like(coll._anchor),) + (None,) * 19
 
 
 def _a2a(sev: ShardedEmbeddingCollection, tensor, in_sp, out_sp):
@@ -173,6 +369,9 @@ def _exchange_lookup(sev: ShardedEmbeddingCollection, uniq, counts, train):
     send_splits = torch.bincount(owner, minlength=w)
     recv_splits = comm.exchange_counts(send_splits)
     in_sp, out_sp = send_splits.tolist(), recv_splits.tolist()
+    # cap-sizing aid for static_mode: track the largest split observed
+    sev._observed_max_split = max(sev._observed_max_split,
+                                  max(in_sp), max(out_sp))
     recv_keys = _a2a(sev, send_keys, in_sp, out_sp)
     recv_counts = _a2a(sev, send_counts, in_sp, out_sp)
     uniq2, inv2 = torch.unique(recv_keys, return_inverse=True)

@@ -442,6 +442,108 @@ def test_hierarchical_a2a_world6():
     _run_dist(_body_hierarchical_a2a, world_size=6, port=29549)
 
 
+def _body_padded_collection(rank, world):
+    """static_mode (padded fixed-shape exchange) must train identically to
+    the single-process reference collection — the wire protocol that the
+    captured distributed step replays on hardware."""
+    from deeprec_amd import EmbeddingVariableOption
+    from deeprec_amd.embedding.collection import EmbeddingCollection
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.parallel.sharded_collection import (
+        ShardedEmbeddingCollection)
+
+    def init(t):
+        gen = torch.Generator().manual_seed(21)
+        t.normal_(0, 1, generator=gen)
+
+    opt_ev = EmbeddingVariableOption(
+        init_option=InitializerOption(initializer=init, default_value_dim=4))
+    sc = ShardedEmbeddingCollection("pc", ["a", "b"], 8, ev_option=opt_ev)
+    sc.static_mode(pad_cap=64)
+    ref = EmbeddingCollection("pref", ["a", "b"], 8, ev_option=opt_ev)
+    o_s = AdagradOptimizer(embedding_variables=[sc], learning_rate=0.1)
+    o_r = AdagradOptimizer(embedding_variables=[ref], learning_rate=0.1)
+    for step in range(3):
+        g = torch.Generator().manual_seed(900 + step)
+        ids_all = torch.randint(0, 40, (8 * world, 2), generator=g)
+        out = sc.lookup_matrix(ids_all[rank * 8:(rank + 1) * 8])
+        (out ** 2).sum().backward()
+        o_s.step()
+        (ref.lookup_matrix(ids_all) ** 2).sum().backward()
+        o_r.step()
+    tabs_s, tabs_r = sc.export_tables(), ref.export_tables()
+    for name in ("a", "b"):
+        ks, vs, fs, _ = tabs_s[name]
+        kr, vr, fr, _ = tabs_r[name]
+        gk = [None] * world
+        gv = [None] * world
+        gf = [None] * world
+        dist.all_gather_object(gk, ks)
+        dist.all_gather_object(gv, vs)
+        dist.all_gather_object(gf, fs)
+        ka, va, fa = torch.cat(gk), torch.cat(gv), torch.cat(gf)
+        oi, ri = torch.argsort(ka), torch.argsort(kr)
+        torch.testing.assert_close(ka[oi], kr[ri])
+        torch.testing.assert_close(va[oi], vr[ri], rtol=1e-4, atol=1e-5)
+        # frequency counters ride the wire (summed true counts)
+        torch.testing.assert_close(fa[oi], fr[ri])
+
+
+def test_padded_collection_world2():
+    _run_dist(_body_padded_collection, world_size=2, port=29552)
+
+
+def test_padded_collection_world4():
+    _run_dist(_body_padded_collection, world_size=4, port=29553)
+
+
+def _body_padded_overflow(rank, world):
+    from deeprec_amd.parallel.sharded_collection import (
+        ShardedEmbeddingCollection)
+
+    sc = ShardedEmbeddingCollection("povf", ["a"], 4)
+    sc.static_mode(pad_cap=2)  # far below the per-peer unique count
+    ids = torch.arange(32).reshape(32, 1)
+    try:
+        out = sc.lookup_matrix(ids)
+        out.sum().backward()
+        raise AssertionError("expected pad-cap overflow ValueError")
+    except ValueError as e:
+        assert "overflow" in str(e)
+    dist.barrier()
+
+
+def test_padded_overflow_world2():
+    _run_dist(_body_padded_overflow, world_size=2, port=29554)
+
+
+def _body_padded_bf16_wire(rank, world):
+    """Padded exchange with bf16 row transport matches fp32 within bf16
+    quantization (the gloo int16-view wire branch)."""
+    from deeprec_amd.parallel.sharded_collection import (
+        ShardedEmbeddingCollection)
+
+    names = [f"t{i}" for i in range(3)]
+    ids = torch.randint(0, 50, (16, 3),
+                        generator=torch.Generator().manual_seed(4))
+    g1 = torch.Generator().manual_seed(6)
+    g2 = torch.Generator().manual_seed(6)
+    s32 = ShardedEmbeddingCollection("pw32", names, 8, generator=g1)
+    s16 = ShardedEmbeddingCollection("pw16", names, 8, generator=g2,
+                                     comm_dtype=torch.bfloat16)
+    s32.static_mode(pad_cap=128)
+    s16.static_mode(pad_cap=128)
+    o32 = s32.lookup_matrix(ids, train=True)
+    o16 = s16.lookup_matrix(ids, train=True)
+    torch.testing.assert_close(o16, o32, rtol=1e-2, atol=1e-2)
+    (o16.sum() + o32.sum()).backward()
+    dist.barrier()
+
+
+def test_padded_bf16_wire_world2():
+    _run_dist(_body_padded_bf16_wire, world_size=2, port=29555)
+
+
 def _body_sharded_ev_checkpoint(rank, world):
     """Sharded plain-EV save/restore, including 64-bit-hash ids (>= 2^48
     and negative): restore ownership must match the full-key % world
