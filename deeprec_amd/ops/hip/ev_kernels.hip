@@ -40,6 +40,7 @@ static inline hipStream_t current_stream() {
 #define DEV_INLINE __device__ __forceinline__
 
 static constexpr int64_t EMPTY_KEY = INT64_MIN;  // reserved sentinel
+static constexpr int64_t PAD_KEY = INT64_MAX;    // wire-padding sentinel
 static constexpr int kBlock = 256;
 
 namespace {
@@ -344,7 +345,10 @@ __global__ void k_dedup_pass_b_padded(
     int32_t* __restrict__ out_slots, int32_t* __restrict__ error_flag) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= n_cap) return;
-  if (c >= *m_dev) {
+  if (c >= *m_dev || uniq_keys[c] == PAD_KEY) {
+    // tail slot, or the wire-padding sentinel of the sharded exchange:
+    // no admission, no metadata (PAD_KEY's composite-decomposed default
+    // row would read far out of bounds of default_values)
     out_slots[c] = -1;
     return;
   }
@@ -401,8 +405,6 @@ __global__ void k_zero_f32(float* __restrict__ p, int64_t n) {
 // count exchange is needed on the wire at all (≙ the reference's
 // two-phase count+payload protocol, all2all_input_dispatcher.cu:250-280,
 // re-designed shape-static for graph replay).
-
-static constexpr int64_t PAD_KEY = INT64_MAX;
 
 // Bump only the dedup epoch (NOT the step): the distributed step runs
 // TWO dedups per training step against the same table (requester-side
