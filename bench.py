@@ -169,6 +169,8 @@ def main():
                 reducer.allreduce(async_op=True)
             opt.step()
         torch.cuda.synchronize()
+        if rank == 0:
+            print("bench: hipGraph capture ok", flush=True)
 
         def graph_step(i):
             dense, sparse, labels = batches[i % len(batches)]
@@ -194,16 +196,23 @@ def main():
         for i in range(args.warmup):
             one_step(i)
 
+    if rank == 0:
+        print(f"bench: warmup done (graph={'on' if use_graph else 'off'})",
+              flush=True)
     if distributed:
         import torch.distributed as dist
         dist.barrier()
     if device.type == "cuda":
         torch.cuda.synchronize()
+    if rank == 0:
+        print("bench: timed region start", flush=True)
     t0 = time.perf_counter()
     for i in range(args.steps):
         one_step(args.warmup + i)
     if device.type == "cuda":
         torch.cuda.synchronize()
+    if rank == 0:
+        print("bench: timed region end", flush=True)
     if distributed:
         import torch.distributed as dist
         dist.barrier()
@@ -223,8 +232,9 @@ def main():
     samples_per_sec = global_batch * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    out_json = None
     if rank == 0:
-        print(json.dumps({
+        out_json = json.dumps({
             "metric": "samples/sec (whole node) DLRM Criteo-TB-shaped synthetic",
             "value": samples_per_sec,
             "unit": "samples/sec",
@@ -244,11 +254,32 @@ def main():
                        "hip_graph": use_graph,
                        "optimizer": args.optimizer,
                        "embedding_dim": 16, "num_tables": 26},
-        }))
+        })
+        print(out_json, flush=True)
 
     if distributed:
+        import sys
+        import threading
         import torch.distributed as dist
-        dist.destroy_process_group()
+        sys.stdout.flush()
+        if use_graph:
+            # release the captured graph BEFORE comm teardown: RCCL work
+            # recorded during capture never completes host-side
+            # bookkeeping, so the watchdog would wait on it forever
+            one_step = None  # noqa: F841 (drops the closure's graph ref)
+            graph = None  # noqa: F841
+            import gc
+            gc.collect()
+            torch.cuda.synchronize()
+        # teardown of a comm that carried captured collectives can still
+        # wedge in the driver; bound it — the job is done and the output
+        # is flushed, so a stuck teardown must not fail the bench
+        th = threading.Thread(target=dist.destroy_process_group,
+                              daemon=True)
+        th.start()
+        th.join(timeout=30)
+        if th.is_alive():
+            os._exit(0)
 
 
 if __name__ == "__main__":
