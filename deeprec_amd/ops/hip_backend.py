@@ -299,18 +299,19 @@ class HbmStorage:
                                 self._step_dev, self.entry_counter,
                                 m_counter, uniq_buf, centry_buf,
                                 self.error_flag)
-        inverse, _counts_c, _rank = self.ext.ht_dedup_c(
+        # PAD-aware pass C: pad elements get inverse -1 and never touch
+        # the shared counters (a naive path serializes ~10^5 atomics on
+        # PAD_KEY's single cache line)
+        inverse, _counts_c, _rank = self.ext.ht_dedup_c_pad(
             recv_keys, self.ht_keys, self.ht_compact, nnz)
-        inv_l = inverse.long()
-        cnt_sum = torch.zeros(nnz, dtype=torch.int32, device=self.device)
-        cnt_sum.index_add_(0, inv_l, recv_cnt)
+        cnt_sum = self.ext.cnt_sum_pad(inverse, recv_cnt, nnz)
         slots = self.ext.ht_dedup_b_padded(
             centry_buf, uniq_buf, m_counter, self.ht_slot, self.ht_freq,
             self.ht_version, cnt_sum, self._step_dev,
             self.slot_counter, self.max_slots, self.values,
             self.default_values, self.dvd_per_table, self.key_bits,
             self._init_limit(), self.filter_freq, self.error_flag)
-        return uniq_buf, inv_l, slots
+        return uniq_buf, inverse, slots
 
     def prefers_dedup(self) -> bool:
         return self._uniq_ratio is None or self._uniq_ratio < 0.5

@@ -256,14 +256,14 @@ def _padded_sharded_lookup(sev, values_cat, offsets_cat, row_ids_cat,
     if is_gpu:
         uniq2, inv2, slots2 = st.dedup_lookup_capture_owner(recv_keys,
                                                             recv_cnt)
-        slots_elem = slots2.index_select(0, inv2)
+        slots_elem = st.ext.slots_gather_pad(inv2, slots2)
         wire = sev.comm_dtype or torch.float32
         reply = st.ext.ev_gather(
             st.values, st.default_values, recv_keys, slots_elem,
             st._no_permission_value(), st._use_no_permission(), wire)
         own_valid = None
         emb_rows = comm.all_to_all_single(reply, splits, splits).float()
-        inverse_final = route_pos.index_select(0, inverse.long())
+        inverse_final = st.ext.compose_i32(inverse, route_pos)
         keys_for_kernel = recv_keys
     else:
         uniq2, inv2 = torch.unique(recv_keys, return_inverse=True)
@@ -331,10 +331,7 @@ class _PaddedShardedLookup(torch.autograd.Function):
             grad_send = st.ext.rows_to_padded(grad_unique, route_pos,
                                               ctx.m_dev, w * cap)
             grad_recv = comm.all_to_all_single(grad_send, splits, splits)
-            grad2 = torch.zeros(w * cap, grad_recv.shape[1],
-                                device=grad_recv.device,
-                                dtype=grad_recv.dtype)
-            grad2.index_add_(0, inv2, grad_recv)
+            grad2 = st.ext.rows_segsum_pad(grad_recv, inv2, w * cap)
             coll.accumulate_grad(slots2, uniq2, grad2)
         else:
             grad_send = torch.zeros(w * cap, grad_unique.shape[1])
