@@ -259,27 +259,9 @@ def main():
 
     if distributed:
         import sys
-        import threading
-        import torch.distributed as dist
         sys.stdout.flush()
-        if use_graph:
-            # release the captured graph BEFORE comm teardown: RCCL work
-            # recorded during capture never completes host-side
-            # bookkeeping, so the watchdog would wait on it forever
-            one_step = None  # noqa: F841 (drops the closure's graph ref)
-            graph = None  # noqa: F841
-            import gc
-            gc.collect()
-            torch.cuda.synchronize()
-        # teardown of a comm that carried captured collectives can still
-        # wedge in the driver; bound it — the job is done and the output
-        # is flushed, so a stuck teardown must not fail the bench
-        th = threading.Thread(target=dist.destroy_process_group,
-                              daemon=True)
-        th.start()
-        th.join(timeout=30)
-        if th.is_alive():
-            os._exit(0)
+        from deeprec_amd.parallel import comm
+        comm.shutdown(after_capture=use_graph)
 
 
 if __name__ == "__main__":

@@ -19,6 +19,31 @@ def is_initialized() -> bool:
     return dist.is_available() and dist.is_initialized()
 
 
+def shutdown(after_capture: bool = False):
+    """Tear down the process group. after_capture=True when collectives
+    were recorded into a hipGraph this process: their host-side work
+    bookkeeping never completes, so a clean destroy_process_group waits
+    forever — abort the communicator instead (safe at teardown: training
+    is done and every rank aborts at the same point)."""
+    if not is_initialized():
+        return
+    if after_capture:
+        try:
+            from torch.distributed.distributed_c10d import (
+                _abort_process_group)
+            _abort_process_group()
+            return
+        except Exception:  # noqa: BLE001 — fall through to bounded destroy
+            pass
+    import threading
+    th = threading.Thread(target=dist.destroy_process_group, daemon=True)
+    th.start()
+    th.join(timeout=30)
+    if th.is_alive():
+        import os
+        os._exit(0)  # teardown wedged; the job itself already finished
+
+
 def world_size() -> int:
     return dist.get_world_size() if is_initialized() else 1
 
