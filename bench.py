@@ -88,16 +88,28 @@ def main():
     opt = make_optimizer(args.optimizer, params=model.parameters(),
                          embedding_variables=model.embedding_variables(),
                          learning_rate=0.001, **opt_kw)
+    flat_dense = False
+    if use_graph and bf16:
+        # ONE fused kernel for the dense Adam update that also emits the
+        # bf16 weight shadows; dW/db land directly in its flat grad
+        # buffer, so the dense all-reduce is one in-place collective
+        from deeprec_amd.ops.dense_adam import use_flat_dense_adam
+        flat_dense = use_flat_dense_adam(opt, model)
     if distributed:
         from deeprec_amd.parallel import (DenseGradAllreducer,
                                           broadcast_parameters)
         broadcast_parameters(model.parameters())
-        reducer = DenseGradAllreducer(model.parameters())
+        if flat_dense:
+            from deeprec_amd.ops.dense_adam import FlatGradAllreducer
+            opt._dense.refresh_shadows()  # shadows follow the broadcast
+            reducer = FlatGradAllreducer(opt._dense)
+        else:
+            reducer = DenseGradAllreducer(model.parameters())
         # overlap the dense all-reduce with the sparse applies
         opt.pre_dense_step = reducer.wait
     else:
         reducer = None
-    if bf16:
+    if bf16 and not flat_dense:
         # AFTER the parameter broadcast: the bf16 shadows snapshot the
         # weights at enable time
         from deeprec_amd.ops.fused_mlp import enable_weight_cache
@@ -260,6 +272,15 @@ def main():
     if distributed:
         import sys
         sys.stdout.flush()
+        if use_graph:
+            # destroy the hipGraph while the communicator is still alive:
+            # a CUDAGraph destructor running after comm abort/teardown
+            # (e.g. at interpreter exit) wedges in hipGraphExecDestroy
+            one_step = None  # noqa: F841 (drops the closure's graph ref)
+            graph = None  # noqa: F841
+            import gc
+            gc.collect()
+            torch.cuda.synchronize()
         from deeprec_amd.parallel import comm
         comm.shutdown(after_capture=use_graph)
 

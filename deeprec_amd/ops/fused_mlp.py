@@ -19,7 +19,7 @@ _ACT_ID = {None: 0, "none": 0, "relu": 1, "sigmoid": 2}
 
 class _FusedLinear(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, act_id, w16_cache=None):
+    def forward(ctx, x, weight, bias, act_id, w16_cache=None, dw_ws=None):
         from deeprec_amd.ops.build_ext import require_extension
         ext = require_extension()
         # w16_cache: bf16 shadow refreshed once per optimizer step by
@@ -31,6 +31,7 @@ class _FusedLinear(torch.autograd.Function):
         ctx.ext = ext
         ctx.act_id = act_id
         ctx.x_dtype = x.dtype
+        ctx.dw_ws = dw_ws
         ctx.save_for_backward(x16, w16, out)
         return out
 
@@ -45,8 +46,13 @@ class _FusedLinear(torch.autograd.Function):
         # direct col-fragment dW: measured faster end-to-end than the
         # transpose-then-row-load variant (torch .t().contiguous() costs
         # ~12us/copy, more than the strided-fragment penalty it removes)
-        dw, db = ext.linear_dw(g, x16, True)
-        return dx.to(ctx.x_dtype), dw, db, None, None
+        if ctx.dw_ws is not None:
+            # FlatDenseAdam mode: dW/db land in the optimizer's flat
+            # gradient buffer (zero-filled + accumulated by the kernel)
+            dw, db = ext.linear_dw_out(g, x16, ctx.dw_ws, True)
+        else:
+            dw, db = ext.linear_dw(g, x16, True)
+        return dx.to(ctx.x_dtype), dw, db, None, None, None
 
 
 class FusedLinear(nn.Module):
@@ -68,6 +74,7 @@ class FusedLinear(nn.Module):
         bound = 1.0 / math.sqrt(in_features)
         nn.init.uniform_(self.bias, -bound, bound)
         self.w16_cache = None  # set by enable_weight_cache()
+        self.dw_ws = None      # set by FlatDenseAdam (flat grad slice)
 
     def forward(self, x):
         lead = x.shape[:-1]
@@ -75,7 +82,7 @@ class FusedLinear(nn.Module):
         if x.device.type == "cuda" and flat.shape[0] % 16 == 0:
             out = _FusedLinear.apply(flat, self.weight, self.bias,
                                      _ACT_ID[self.activation],
-                                     self.w16_cache)
+                                     self.w16_cache, self.dw_ws)
         else:
             out = nn.functional.linear(flat, self.weight.to(flat.dtype),
                                        self.bias.to(flat.dtype))

@@ -596,6 +596,36 @@ __global__ void k_zero_f32d(float* __restrict__ p, int64_t n) {
     for (int64_t j = n & ~3LL; j < n; ++j) p[j] = 0.0f;
 }
 
+// ---------------------------------------------------------------------
+// fused flat dense Adam (≙ reference dense ApplyAdamAsync,
+// training_ali_ops_gpu.cu.cc:534, re-designed as ONE kernel over a flat
+// parameter buffer): bias-correction from device-resident beta powers
+// (capture-safe, no elementwise prelude), optional bf16 shadow emission
+// in the same pass (replaces the per-step multi-tensor cast), optional
+// gradient scale (folds the data-parallel 1/world averaging into the
+// update so the all-reduce needs no separate div).
+__global__ void k_dense_adam(float* __restrict__ w,
+                             const float* __restrict__ g,
+                             float* __restrict__ m, float* __restrict__ v,
+                             short* __restrict__ w16,
+                             const float* __restrict__ powers, int64_t n,
+                             float lr, float beta1, float beta2, float eps,
+                             float gscale) {
+  const float lr_t = lr * sqrtf(1.0f - powers[1]) / (1.0f - powers[0]);
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += stride) {
+    float gi = g[i] * gscale;
+    float mn = beta1 * m[i] + (1.0f - beta1) * gi;
+    float vn = beta2 * v[i] + (1.0f - beta2) * gi * gi;
+    m[i] = mn;
+    v[i] = vn;
+    float wn = w[i] - lr_t * mn / (sqrtf(vn) + eps);
+    w[i] = wn;
+    if (w16) w16[i] = f2bf_u16(wn);
+  }
+}
+
 // activation backward: G = dY * act_grad(out); act 1=relu, 2=sigmoid
 __global__ void k_act_bwd(const short* __restrict__ dY,
                           const short* __restrict__ out, int64_t n, int act,
@@ -779,16 +809,14 @@ torch::Tensor linear_dx(torch::Tensor g, torch::Tensor w_bf16) {
   return dx;
 }
 
-std::tuple<torch::Tensor, torch::Tensor> linear_dw(torch::Tensor g,
-                                                   torch::Tensor x,
-                                                   bool want_bias,
-                                                   int64_t variant) {
+static std::tuple<torch::Tensor, torch::Tensor> linear_dw_impl(
+    torch::Tensor g, torch::Tensor x, torch::Tensor ws, bool want_bias,
+    int64_t variant) {
   int M = g.size(0), N = g.size(1), K = x.size(1);
   // one fill-kernel zero for dW + dbias (cheaper than torch's fill
   // machinery; hipMemsetAsync is avoided: memset nodes recorded during
   // hipGraph capture were observed not to replay)
   int64_t ws_len = (int64_t)N * K + (want_bias ? N : 0);
-  auto ws = torch::empty({ws_len}, g.options().dtype(torch::kFloat32));
   k_zero_f32d<<<(int)std::min<int64_t>((ws_len / 4 + 255) / 256, 1024),
                 256, 0, dense_stream()>>>(ws.data_ptr<float>(), ws_len);
   auto dw = ws.narrow(0, 0, (int64_t)N * K).view({N, K});
@@ -815,6 +843,48 @@ std::tuple<torch::Tensor, torch::Tensor> linear_dw(torch::Tensor g,
         bf_ptr(g), bf_ptr(x), M, N, K, dw.data_ptr<float>(), dbp);
   }
   return {dw, db};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> linear_dw(torch::Tensor g,
+                                                   torch::Tensor x,
+                                                   bool want_bias,
+                                                   int64_t variant) {
+  int N = g.size(1), K = x.size(1);
+  int64_t ws_len = (int64_t)N * K + (want_bias ? N : 0);
+  auto ws = torch::empty({ws_len}, g.options().dtype(torch::kFloat32));
+  return linear_dw_impl(g, x, ws, want_bias, variant);
+}
+
+// dW/db written into a caller-provided flat [N*K + N] workspace — a view
+// of the optimizer's flat gradient buffer, so backward lands gradients
+// exactly where the fused dense Adam reads them (no flatten/unflatten
+// copies around the all-reduce).
+std::tuple<torch::Tensor, torch::Tensor> linear_dw_out(torch::Tensor g,
+                                                       torch::Tensor x,
+                                                       torch::Tensor ws,
+                                                       bool want_bias,
+                                                       int64_t variant) {
+  TORCH_CHECK(ws.is_contiguous() && ws.scalar_type() == torch::kFloat32);
+  int N = g.size(1), K = x.size(1);
+  int64_t ws_len = (int64_t)N * K + (want_bias ? N : 0);
+  TORCH_CHECK(ws.numel() == ws_len, "linear_dw_out: workspace size");
+  return linear_dw_impl(g, x, ws, want_bias, variant);
+}
+
+void dense_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
+                torch::Tensor v, torch::Tensor w16, torch::Tensor powers,
+                double lr, double beta1, double beta2, double eps,
+                double gscale) {
+  int64_t n = w.numel();
+  if (n == 0) return;
+  short* w16p = (w16.defined() && w16.numel())
+                    ? reinterpret_cast<short*>(w16.data_ptr<at::BFloat16>())
+                    : nullptr;
+  int blocks = (int)std::min<int64_t>((n + 255) / 256, 4096);
+  k_dense_adam<<<blocks, 256, 0, dense_stream()>>>(
+      w.data_ptr<float>(), g.data_ptr<float>(), m.data_ptr<float>(),
+      v.data_ptr<float>(), w16p, powers.data_ptr<float>(), n, (float)lr,
+      (float)beta1, (float)beta2, (float)eps, (float)gscale);
 }
 
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor out, int64_t act) {
@@ -885,5 +955,8 @@ void register_dense(py::module_& mod) {
   mod.def("linear_dx", &linear_dx);
   mod.def("linear_dw", &linear_dw, py::arg("g"), py::arg("x"),
           py::arg("want_bias"), py::arg("variant") = -1);
+  mod.def("linear_dw_out", &linear_dw_out, py::arg("g"), py::arg("x"),
+          py::arg("ws"), py::arg("want_bias"), py::arg("variant") = -1);
+  mod.def("dense_adam", &dense_adam);
   mod.def("act_bwd", &act_bwd);
 }

@@ -196,3 +196,88 @@ def test_fused_l2_normalize_matches_torch():
         torch.nn.functional.normalize(xr, dim=-1).backward(g.double())
         torch.testing.assert_close(x.grad.float(), xr.grad.float(),
                                    rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.gpu
+def test_flat_dense_adam_matches_torch_adam():
+    """FlatDenseAdam (flat buffer, fused update + bf16 shadow emission,
+    dW landed in the flat grad buffer) must track a torch.optim.Adam
+    trajectory over the same FusedLinear stack."""
+    from deeprec_amd.ops.dense_adam import use_flat_dense_adam
+    from deeprec_amd.ops.fused_mlp import fused_mlp
+    from deeprec_amd.optimizers import AdamAsyncOptimizer
+
+    torch.manual_seed(3)
+    ma = fused_mlp([64, 32], 16).to("cuda")
+    torch.manual_seed(3)
+    mb = fused_mlp([64, 32], 16).to("cuda")
+    for pa, pb in zip(ma.parameters(), mb.parameters()):
+        torch.testing.assert_close(pa, pb)
+    oa = AdamAsyncOptimizer(params=ma.parameters(), learning_rate=0.01,
+                            graph_safe=True)
+    assert use_flat_dense_adam(oa, ma)
+    ob = AdamAsyncOptimizer(params=mb.parameters(), learning_rate=0.01)
+    for step in range(5):
+        x = torch.randn(128, 16, device="cuda",
+                        generator=torch.Generator("cuda").manual_seed(step))
+        for m, o in ((ma, oa), (mb, ob)):
+            o.zero_grad()
+            (m(x) ** 2).mean().backward()
+            o.step()
+    # same trajectory up to Adam epsilon placement (TF-style vs torch)
+    for pa, pb in zip(ma.parameters(), mb.parameters()):
+        torch.testing.assert_close(pa, pb, rtol=2e-3, atol=2e-4)
+    # shadows track the master weights
+    for mod in ma.modules():
+        if hasattr(mod, "w16_cache") and mod.w16_cache is not None:
+            torch.testing.assert_close(mod.w16_cache.float(),
+                                       mod.weight.detach().float(),
+                                       rtol=1e-2, atol=1e-2)
+
+
+@pytest.mark.gpu
+def test_flat_dense_adam_captured_replay():
+    """The fused dense update + shadow emission replays correctly in a
+    hipGraph (fresh inputs per replay)."""
+    from deeprec_amd.ops.dense_adam import use_flat_dense_adam
+    from deeprec_amd.ops.fused_mlp import fused_mlp
+    from deeprec_amd.optimizers import AdamAsyncOptimizer
+
+    torch.manual_seed(9)
+    m = fused_mlp([64, 32], 16).to("cuda")
+    o = AdamAsyncOptimizer(params=m.parameters(), learning_rate=0.01,
+                           graph_safe=True)
+    assert use_flat_dense_adam(o, m)
+    xs = [torch.randn(128, 16, device="cuda") for _ in range(6)]
+
+    def step(x):
+        o.zero_grad()
+        (m(x) ** 2).mean().backward()
+        o.step()
+
+    step(xs[0])
+    step(xs[1])
+    torch.cuda.synchronize()
+    sx = xs[2].clone()
+    g = torch.cuda.CUDAGraph()
+    o.zero_grad()
+    with torch.cuda.graph(g):
+        (m(sx) ** 2).mean().backward()
+        o.step()
+    torch.cuda.synchronize()
+    w_after_capture = o._dense.w.clone()
+    for x in xs[3:6]:
+        sx.copy_(x)
+        g.replay()
+    torch.cuda.synchronize()
+    # weights moved across replays and powers advanced 3 more times
+    assert not torch.equal(w_after_capture, o._dense.w)
+    expected_p0 = 0.9 ** 6  # 2 eager + capture-recorded? capture does not
+    # execute, so: 2 eager + 3 replays + the step() recorded during
+    # capture runs 0 times at record = 5 executions... powers advance
+    # once per executed step + once at capture? capture records the
+    # update_powers kernel but does not run it
+    del expected_p0
+    p0 = float(o._dense.powers[0].cpu())
+    assert abs(p0 - 0.9 ** 5) < 1e-6, p0
+    assert torch.isfinite(o._dense.w).all()
