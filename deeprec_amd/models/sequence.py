@@ -161,10 +161,12 @@ class BST(_SeqBase):
         super().__init__(embedding_dim, item_dim, device, bf16, name="bst",
                          **kw)
         self.pos = nn.Parameter(torch.zeros(max_len, item_dim))
-        layer = nn.TransformerEncoderLayer(
-            d_model=item_dim, nhead=n_heads, dim_feedforward=ff_dim,
-            batch_first=True, dropout=0.0)
-        self.encoder = nn.TransformerEncoder(layer, n_layers)
+        from deeprec_amd.ops.fused_attention import FusedTransformerEncoder
+        # per-sample LDS attention + bf16 MFMA projections: torch's
+        # nn.TransformerEncoder at these shapes (T~50, d=32) ran fp32
+        # launch-bound batched GEMMs at 9.9 ms/step (batch 8192)
+        self.encoder = FusedTransformerEncoder(item_dim, n_heads, ff_dim,
+                                               n_layers)
         in_dim = NUM_DENSE + self.num_sparse * embedding_dim + item_dim
         self.mlp = make_mlp(list(mlp_sizes) + [1], in_dim, device, self.bf16,
                             final_activation=False)
@@ -182,7 +184,7 @@ class BST(_SeqBase):
             [seq_ids <= 0,
              torch.zeros(seq.shape[0], 1, dtype=torch.bool,
                          device=seq.device)], 1)
-        enc = self.encoder(x_seq, src_key_padding_mask=pad)
+        enc = self.encoder(x_seq, pad)
         pooled = enc.mean(1)
         x = torch.cat([dense, emb.flatten(1).float(), pooled], 1)
         with self.amp():

@@ -1838,12 +1838,14 @@ void apply_ftrl(torch::Tensor w, torch::Tensor n, torch::Tensor z,
       (float)l1, (float)l2, (float)lr_power, (float)l2_shrinkage);
 }
 
-void register_dense(pybind11::module_& mod);  // dense_kernels.hip
-void register_gru(pybind11::module_& mod);    // gru_kernels.hip
+void register_dense(pybind11::module_& mod);      // dense_kernels.hip
+void register_gru(pybind11::module_& mod);        // gru_kernels.hip
+void register_attention(pybind11::module_& mod);  // attention_kernels.hip
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   register_dense(mod);
   register_gru(mod);
+  register_attention(mod);
   mod.def("ht_lookup_insert", &ht_lookup_insert);
   mod.def("ht_dedup_a", &ht_dedup_a);
   mod.def("ht_dedup_b", &ht_dedup_b);

