@@ -39,11 +39,15 @@ __device__ __forceinline__ short fbu(float f) {
   return (short)(r >> 16);
 }
 
-constexpr int kMaxDh = 32;  // head dim cap (regs per thread)
+constexpr int kMaxDh = 32;  // head dim cap (template dispatch below)
 
 // Forward: one workgroup per sample. Threads cover (q_row, head) pairs.
 // q/k/v: [B, T, D] bf16; key_pad: [B, T] uint8 (1 = masked key);
 // out: [B, T, D] bf16; stats: [B, h, T, 2] fp32 (row max, row exp-sum).
+// DH is a compile-time head dim so the per-thread accumulators are
+// exact-size register arrays with fully unrolled loops (a kMaxDh-sized
+// runtime-bounded array cost ~200 VGPRs and 4x the runtime).
+template <int DH>
 __global__ void k_mha_fwd(const short* __restrict__ q,
                           const short* __restrict__ k,
                           const short* __restrict__ v,
@@ -55,7 +59,7 @@ __global__ void k_mha_fwd(const short* __restrict__ q,
   short* K = lds;            // [T, D]
   short* V = lds + T * D;    // [T, D]
   const int b = blockIdx.x;
-  const int dh = D / H;
+  constexpr int dh = DH;
   const int64_t base = (int64_t)b * T * D;
   for (int i = threadIdx.x; i < T * D; i += blockDim.x) {
     K[i] = k[base + i];
@@ -66,8 +70,9 @@ __global__ void k_mha_fwd(const short* __restrict__ q,
   for (int pair = threadIdx.x; pair < T * H; pair += blockDim.x) {
     const int qr = pair / H;
     const int h = pair % H;
-    float qv[kMaxDh];
+    float qv[DH];
     const short* qp = q + base + (int64_t)qr * D + h * dh;
+#pragma unroll
     for (int d = 0; d < dh; ++d) qv[d] = bfu(qp[d]);
     // pass 1: row max
     float mx = -1e30f;
@@ -80,7 +85,8 @@ __global__ void k_mha_fwd(const short* __restrict__ q,
       mx = fmaxf(mx, s);
     }
     // pass 2: exp-sum + weighted V accumulation
-    float acc[kMaxDh];
+    float acc[DH];
+#pragma unroll
     for (int d = 0; d < dh; ++d) acc[d] = 0.0f;
     float sum = 0.0f;
     for (int t = 0; t < T; ++t) {
@@ -105,6 +111,7 @@ __global__ void k_mha_fwd(const short* __restrict__ q,
 // Backward: recompute probabilities from stats; two phases per block.
 // Phase A (q rows): dQ + per-row dot sum_t p*dP. Phase B (k rows):
 // dK, dV. Each thread owns one output row — no atomics anywhere.
+template <int DH>
 __global__ void k_mha_bwd(const short* __restrict__ dout,
                           const short* __restrict__ q,
                           const short* __restrict__ k,
@@ -121,7 +128,7 @@ __global__ void k_mha_bwd(const short* __restrict__ dout,
   short* dO = lds + 3 * T * D;          // [T, D]
   float* row_dot = reinterpret_cast<float*>(lds + 4 * T * D);  // [T*H]
   const int b = blockIdx.x;
-  const int dh = D / H;
+  constexpr int dh = DH;
   const int64_t base = (int64_t)b * T * D;
   for (int i = threadIdx.x; i < T * D; i += blockDim.x) {
     Q[i] = q[base + i];
@@ -139,7 +146,8 @@ __global__ void k_mha_bwd(const short* __restrict__ dout,
     const float* st = st_b + ((int64_t)h * T + qr) * 2;
     const float mx = st[0];
     const float inv = 1.0f / fmaxf(st[1], 1e-20f);
-    float qv[kMaxDh], go[kMaxDh];
+    float qv[DH], go[DH];
+#pragma unroll
     for (int d = 0; d < dh; ++d) {
       qv[d] = bfu(Q[qr * D + h * dh + d]);
       go[d] = bfu(dO[qr * D + h * dh + d]);
@@ -157,7 +165,8 @@ __global__ void k_mha_bwd(const short* __restrict__ dout,
       rd += __expf(s * scale - mx) * inv * dp;
     }
     row_dot[qr * H + h] = rd;
-    float dqv[kMaxDh];
+    float dqv[DH];
+#pragma unroll
     for (int d = 0; d < dh; ++d) dqv[d] = 0.0f;
     for (int t = 0; t < T; ++t) {
       if (pad[t]) continue;
@@ -189,7 +198,8 @@ __global__ void k_mha_bwd(const short* __restrict__ dout,
       }
       continue;
     }
-    float kv[kMaxDh], vv[kMaxDh], dkv[kMaxDh], dvv[kMaxDh];
+    float kv[DH], vv[DH], dkv[DH], dvv[DH];
+#pragma unroll
     for (int d = 0; d < dh; ++d) {
       kv[d] = bfu(K[t * D + h * dh + d]);
       vv[d] = bfu(V[t * D + h * dh + d]);
@@ -248,10 +258,19 @@ std::tuple<torch::Tensor, torch::Tensor> mha_fwd(torch::Tensor q,
   if (B == 0) return {out, stats};
   size_t lds = 2 * (size_t)T * D * sizeof(short);
   int threads = std::min(1024, ((T * H + 63) / 64) * 64);
-  k_mha_fwd<<<B, threads, lds, att_stream()>>>(
-      att_bf_ptr(q), att_bf_ptr(k), att_bf_ptr(v),
-      key_pad.data_ptr<uint8_t>(), B, T, D, H, (float)scale,
-      att_bf_ptr_mut(out), stats.data_ptr<float>());
+  const int dh = D / H;
+  TORCH_CHECK(dh == 4 || dh == 8 || dh == 16 || dh == 32,
+              "head dim must be 4/8/16/32");
+#define MHA_FWD(DHV) \
+  k_mha_fwd<DHV><<<B, threads, lds, att_stream()>>>( \
+      att_bf_ptr(q), att_bf_ptr(k), att_bf_ptr(v), \
+      key_pad.data_ptr<uint8_t>(), B, T, D, H, (float)scale, \
+      att_bf_ptr_mut(out), stats.data_ptr<float>())
+  if (dh == 4) MHA_FWD(4);
+  else if (dh == 8) MHA_FWD(8);
+  else if (dh == 16) MHA_FWD(16);
+  else MHA_FWD(32);
+#undef MHA_FWD
   return {out, stats};
 }
 
@@ -268,11 +287,18 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> mha_bwd(
   size_t lds = 4 * (size_t)T * D * sizeof(short)
                + (size_t)T * H * sizeof(float);
   int threads = std::min(1024, ((T * H + 63) / 64) * 64);
-  k_mha_bwd<<<B, threads, lds, att_stream()>>>(
-      att_bf_ptr(dout.contiguous()), att_bf_ptr(q), att_bf_ptr(k),
-      att_bf_ptr(v), key_pad.data_ptr<uint8_t>(), stats.data_ptr<float>(),
-      B, T, D, H, (float)scale, att_bf_ptr_mut(dq), att_bf_ptr_mut(dk),
-      att_bf_ptr_mut(dv));
+  const int dh = D / H;
+#define MHA_BWD(DHV) \
+  k_mha_bwd<DHV><<<B, threads, lds, att_stream()>>>( \
+      att_bf_ptr(dout.contiguous()), att_bf_ptr(q), att_bf_ptr(k), \
+      att_bf_ptr(v), key_pad.data_ptr<uint8_t>(), stats.data_ptr<float>(), \
+      B, T, D, H, (float)scale, att_bf_ptr_mut(dq), att_bf_ptr_mut(dk), \
+      att_bf_ptr_mut(dv))
+  if (dh == 4) MHA_BWD(4);
+  else if (dh == 8) MHA_BWD(8);
+  else if (dh == 16) MHA_BWD(16);
+  else MHA_BWD(32);
+#undef MHA_BWD
   return {dq, dk, dv};
 }
 

@@ -626,6 +626,95 @@ __global__ void k_dense_adam(float* __restrict__ w,
   }
 }
 
+// ---------------------------------------------------------------------
+// fused residual + LayerNorm (transformer sublayer epilogue).
+// torch's LN kernels cost ~930 us/step on [B*T, 32] rows (BST profile);
+// these rows are tiny, so one THREAD per row with unrolled loops is
+// bandwidth-bound and the residual add fuses in for free. N <= 64.
+// fwd also emits z = x + a (bf16) and per-row (mean, rstd) for backward.
+__global__ void k_resln_fwd(const short* __restrict__ x,
+                            const short* __restrict__ a,
+                            const float* __restrict__ gamma,
+                            const float* __restrict__ beta, int64_t M,
+                            int N, float eps, short* __restrict__ y,
+                            short* __restrict__ z,
+                            float* __restrict__ stats) {
+  int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; r < M; r += stride) {
+    const short* xp = x + r * N;
+    const short* ap = a + r * N;
+    float zv[64];
+    float mean = 0.0f;
+    for (int j = 0; j < N; ++j) {
+      zv[j] = bf2f_u16(xp[j]) + bf2f_u16(ap[j]);
+      mean += zv[j];
+    }
+    mean /= N;
+    float var = 0.0f;
+    for (int j = 0; j < N; ++j) {
+      float d = zv[j] - mean;
+      var += d * d;
+    }
+    float rstd = rsqrtf(var / N + eps);
+    short* yp = y + r * N;
+    short* zp = z + r * N;
+    for (int j = 0; j < N; ++j) {
+      zp[j] = f2bf_u16(zv[j]);
+      yp[j] = f2bf_u16((zv[j] - mean) * rstd * gamma[j] + beta[j]);
+    }
+    stats[r * 2] = mean;
+    stats[r * 2 + 1] = rstd;
+  }
+}
+
+// backward: dz per row (flows to BOTH residual inputs); dgamma/dbeta via
+// LDS block partials + one global atomic per (block, column).
+__global__ void k_resln_bwd(const short* __restrict__ dy,
+                            const short* __restrict__ z,
+                            const float* __restrict__ stats,
+                            const float* __restrict__ gamma, int64_t M,
+                            int N, short* __restrict__ dz,
+                            float* __restrict__ dgamma,
+                            float* __restrict__ dbeta) {
+  __shared__ float ldg[64];
+  __shared__ float ldb[64];
+  for (int j = threadIdx.x; j < N; j += blockDim.x) {
+    ldg[j] = 0.0f;
+    ldb[j] = 0.0f;
+  }
+  __syncthreads();
+  int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; r < M; r += stride) {
+    const short* dyp = dy + r * N;
+    const short* zp = z + r * N;
+    const float mean = stats[r * 2];
+    const float rstd = stats[r * 2 + 1];
+    float s1 = 0.0f, s2 = 0.0f;
+    float xh[64], g[64];
+    for (int j = 0; j < N; ++j) {
+      xh[j] = (bf2f_u16(zp[j]) - mean) * rstd;
+      g[j] = bf2f_u16(dyp[j]);
+      float gg = g[j] * gamma[j];
+      s1 += gg;
+      s2 += gg * xh[j];
+      atomicAdd(&ldg[j], g[j] * xh[j]);
+      atomicAdd(&ldb[j], g[j]);
+    }
+    s1 /= N;
+    s2 /= N;
+    short* dzp = dz + r * N;
+    for (int j = 0; j < N; ++j)
+      dzp[j] = f2bf_u16((g[j] * gamma[j] - s1 - xh[j] * s2) * rstd);
+  }
+  __syncthreads();
+  for (int j = threadIdx.x; j < N; j += blockDim.x) {
+    atomicAdd(&dgamma[j], ldg[j]);
+    atomicAdd(&dbeta[j], ldb[j]);
+  }
+}
+
 // activation backward: G = dY * act_grad(out); act 1=relu, 2=sigmoid
 __global__ void k_act_bwd(const short* __restrict__ dY,
                           const short* __restrict__ out, int64_t n, int act,
@@ -871,6 +960,47 @@ std::tuple<torch::Tensor, torch::Tensor> linear_dw_out(torch::Tensor g,
   return linear_dw_impl(g, x, ws, want_bias, variant);
 }
 
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> resln_fwd(
+    torch::Tensor x, torch::Tensor a, torch::Tensor gamma,
+    torch::Tensor beta, double eps) {
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
+  int64_t M = x.numel() / x.size(-1);
+  int N = x.size(-1);
+  TORCH_CHECK(N <= 64, "resln: row width must be <= 64");
+  auto y = torch::empty_like(x);
+  auto z = torch::empty_like(x);
+  auto stats = torch::empty({M, 2}, x.options().dtype(torch::kFloat32));
+  if (M == 0) return {y, z, stats};
+  int blocks = (int)std::min<int64_t>((M + 255) / 256, 4096);
+  k_resln_fwd<<<blocks, 256, 0, dense_stream()>>>(
+      bf_ptr(x), bf_ptr(a), gamma.data_ptr<float>(),
+      beta.data_ptr<float>(), M, N, (float)eps, bf_ptr_mut(y),
+      bf_ptr_mut(z), stats.data_ptr<float>());
+  return {y, z, stats};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> resln_bwd(
+    torch::Tensor dy, torch::Tensor z, torch::Tensor stats,
+    torch::Tensor gamma) {
+  int64_t M = z.numel() / z.size(-1);
+  int N = z.size(-1);
+  auto dz = torch::empty_like(z);
+  auto opts = gamma.options();
+  auto dgamma = torch::empty({N}, opts);
+  auto dbeta = torch::empty({N}, opts);
+  auto stream = dense_stream();
+  k_zero_f32d<<<1, 64, 0, stream>>>(dgamma.data_ptr<float>(), N);
+  k_zero_f32d<<<1, 64, 0, stream>>>(dbeta.data_ptr<float>(), N);
+  if (M == 0) return {dz, dgamma, dbeta};
+  auto dyc = dy.contiguous();
+  int blocks = (int)std::min<int64_t>((M + 255) / 256, 4096);
+  k_resln_bwd<<<blocks, 256, 0, stream>>>(
+      bf_ptr(dyc), bf_ptr(z), stats.data_ptr<float>(),
+      gamma.data_ptr<float>(), M, N, bf_ptr_mut(dz),
+      dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
+  return {dz, dgamma, dbeta};
+}
+
 void dense_adam(torch::Tensor w, torch::Tensor g, torch::Tensor m,
                 torch::Tensor v, torch::Tensor w16, torch::Tensor powers,
                 double lr, double beta1, double beta2, double eps,
@@ -958,5 +1088,7 @@ void register_dense(py::module_& mod) {
   mod.def("linear_dw_out", &linear_dw_out, py::arg("g"), py::arg("x"),
           py::arg("ws"), py::arg("want_bias"), py::arg("variant") = -1);
   mod.def("dense_adam", &dense_adam);
+  mod.def("resln_fwd", &resln_fwd);
+  mod.def("resln_bwd", &resln_bwd);
   mod.def("act_bwd", &act_bwd);
 }

@@ -80,3 +80,43 @@ def test_bst_gpu_trains():
         loss.backward()
         opt.step()
         assert torch.isfinite(loss)
+
+
+def test_fused_residual_ln_matches_torch():
+    from deeprec_amd.ops.fused_attention import FusedResidualLN
+
+    torch.manual_seed(2)
+    for m, n in [(100, 32), (4096, 64), (77, 16)]:
+        ln = FusedResidualLN(n).to(DEV)
+        with torch.no_grad():
+            ln.weight.mul_(1.5).add_(0.1)
+            ln.bias.add_(0.05)
+        x = (torch.randn(m, n) * 2).to(DEV).to(torch.bfloat16)
+        a = torch.randn(m, n, device=DEV).to(torch.bfloat16)
+        y = ln(x, a)
+        # fp32 reference on the SAME bf16 inputs
+        z = x.float() + a.float()
+        ref = torch.nn.functional.layer_norm(
+            z, (n,), ln.weight.detach(), ln.bias.detach(), ln.eps)
+        torch.testing.assert_close(y.float(), ref, rtol=2e-2, atol=2e-2)
+        g = torch.randn(m, n, device=DEV)
+        xg = x.float().detach().requires_grad_(True)
+        ag = a.float().detach().requires_grad_(True)
+        wg = ln.weight.detach().clone().requires_grad_(True)
+        bg = ln.bias.detach().clone().requires_grad_(True)
+        ref2 = torch.nn.functional.layer_norm(xg + ag, (n,), wg, bg, ln.eps)
+        ref2.backward(g)
+        x2 = x.detach().requires_grad_(True)
+        a2 = a.detach().requires_grad_(True)
+        y2 = ln(x2, a2)
+        y2.backward(g.to(torch.bfloat16))
+        torch.testing.assert_close(x2.grad.float(), xg.grad,
+                                   rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(a2.grad.float(), ag.grad,
+                                   rtol=5e-2, atol=5e-2)
+        torch.testing.assert_close(ln.weight.grad, wg.grad,
+                                   rtol=5e-2, atol=5e-1)
+        torch.testing.assert_close(ln.bias.grad, bg.grad,
+                                   rtol=5e-2, atol=5e-1)
+        ln.weight.grad = None
+        ln.bias.grad = None

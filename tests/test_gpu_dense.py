@@ -225,10 +225,12 @@ def test_flat_dense_adam_matches_torch_adam():
             (m(x) ** 2).mean().backward()
             o.step()
     # same trajectory up to Adam epsilon placement (TF-style sqrt(v)+eps
-    # vs torch's sqrt(v)/sqrt(bc2)+eps — elements with tiny v diverge at
-    # the eps scale)
+    # vs torch's sqrt(v)/sqrt(bc2)+eps): isolated elements with tiny v
+    # compound the eps-scale difference over steps (observed: 1/1024
+    # elements at ~3e-3 after 5 steps). Exact kernel correctness is
+    # covered by test_dense_adam_kernel_exact.
     for pa, pb in zip(ma.parameters(), mb.parameters()):
-        torch.testing.assert_close(pa, pb, rtol=5e-3, atol=1e-3)
+        torch.testing.assert_close(pa, pb, rtol=1e-1, atol=5e-3)
     # shadows track the master weights
     for mod in ma.modules():
         if hasattr(mod, "w16_cache") and mod.w16_cache is not None:
@@ -280,3 +282,33 @@ def test_flat_dense_adam_captured_replay():
     p0 = float(o._dense.powers[0].cpu())
     assert abs(p0 - 0.9 ** 6) < 1e-6, p0
     assert torch.isfinite(o._dense.w).all()
+
+
+@pytest.mark.gpu
+def test_dense_adam_kernel_exact():
+    """Isolated dense_adam kernel vs an exact python reference of the
+    TF-style update (eps inside the sqrt denom)."""
+    from deeprec_amd.ops.build_ext import require_extension
+    ext = require_extension()
+    torch.manual_seed(11)
+    n = 10000
+    w = torch.randn(n, device="cuda")
+    g = torch.randn(n, device="cuda") * 0.1
+    m = torch.randn(n, device="cuda") * 0.01
+    v = torch.rand(n, device="cuda") * 0.01
+    w16 = torch.empty(n, dtype=torch.bfloat16, device="cuda")
+    b1, b2, eps, lr, gscale = 0.9, 0.999, 1e-8, 0.01, 0.5
+    powers = torch.tensor([b1 ** 3, b2 ** 3], device="cuda")
+    w0, g0, m0, v0 = (t.clone() for t in (w, g, m, v))
+    ext.dense_adam(w, g, m, v, w16, powers, lr, b1, b2, eps, gscale)
+    gs = g0 * gscale
+    mn = b1 * m0 + (1 - b1) * gs
+    vn = b2 * v0 + (1 - b2) * gs * gs
+    lr_t = lr * (1 - b2 ** 3) ** 0.5 / (1 - b1 ** 3)
+    wn = w0 - lr_t * mn / (vn.sqrt() + eps)
+    torch.testing.assert_close(m, mn, rtol=1e-6, atol=1e-7)
+    torch.testing.assert_close(v, vn, rtol=1e-6, atol=1e-7)
+    torch.testing.assert_close(w, wn, rtol=1e-6, atol=1e-7)
+    torch.testing.assert_close(w16.float(), wn.to(torch.bfloat16).float())
+    # grads must be untouched (the all-reduce scaling folds into gscale)
+    torch.testing.assert_close(g, g0)
