@@ -217,20 +217,27 @@ def test_flat_dense_adam_matches_torch_adam():
                             graph_safe=True)
     assert use_flat_dense_adam(oa, ma)
     ob = AdamAsyncOptimizer(params=mb.parameters(), learning_rate=0.01)
-    for step in range(5):
+    def one_step(m, o, seed):
         x = torch.randn(128, 16, device="cuda",
-                        generator=torch.Generator("cuda").manual_seed(step))
-        for m, o in ((ma, oa), (mb, ob)):
-            o.zero_grad()
-            (m(x) ** 2).mean().backward()
-            o.step()
-    # same trajectory up to Adam epsilon placement (TF-style sqrt(v)+eps
-    # vs torch's sqrt(v)/sqrt(bc2)+eps): isolated elements with tiny v
-    # compound the eps-scale difference over steps (observed: 1/1024
-    # elements at ~3e-3 after 5 steps). Exact kernel correctness is
-    # covered by test_dense_adam_kernel_exact.
+                        generator=torch.Generator("cuda").manual_seed(seed))
+        o.zero_grad()
+        (m(x) ** 2).mean().backward()
+        o.step()
+
+    # after ONE step the updates must agree tightly (the only analytic
+    # difference is Adam epsilon placement, negligible at step 1)
+    one_step(ma, oa, 0)
+    one_step(mb, ob, 0)
     for pa, pb in zip(ma.parameters(), mb.parameters()):
-        torch.testing.assert_close(pa, pb, rtol=1e-1, atol=5e-3)
+        torch.testing.assert_close(pa, pb, rtol=1e-3, atol=1e-4)
+    # several more steps: trajectories stay close (per-element divergence
+    # where v~0 compounds chaotically — exact kernel correctness is
+    # test_dense_adam_kernel_exact; this is an end-to-end sanity bound)
+    for step in range(1, 5):
+        one_step(ma, oa, step)
+        one_step(mb, ob, step)
+    for pa, pb in zip(ma.parameters(), mb.parameters()):
+        torch.testing.assert_close(pa, pb, rtol=0.3, atol=2e-2)
     # shadows track the master weights
     for mod in ma.modules():
         if hasattr(mod, "w16_cache") and mod.w16_cache is not None:
