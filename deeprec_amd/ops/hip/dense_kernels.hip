@@ -629,29 +629,39 @@ __global__ void k_dense_adam(float* __restrict__ w,
 // ---------------------------------------------------------------------
 // fused residual + LayerNorm (transformer sublayer epilogue).
 // torch's LN kernels cost ~930 us/step on [B*T, 32] rows (BST profile);
-// these rows are tiny, so one THREAD per row with unrolled loops is
-// bandwidth-bound and the residual add fuses in for free. N <= 64.
+// one THREAD per row with N as a template parameter keeps every row
+// array in registers (a runtime-bounded local array spills to scratch —
+// measured 175/348 us per call; this form is bandwidth-bound).
 // fwd also emits z = x + a (bf16) and per-row (mean, rstd) for backward.
+template <int N>
 __global__ void k_resln_fwd(const short* __restrict__ x,
                             const short* __restrict__ a,
                             const float* __restrict__ gamma,
                             const float* __restrict__ beta, int64_t M,
-                            int N, float eps, short* __restrict__ y,
+                            float eps, short* __restrict__ y,
                             short* __restrict__ z,
                             float* __restrict__ stats) {
+  float gm[N], bt[N];
+#pragma unroll
+  for (int j = 0; j < N; ++j) {
+    gm[j] = gamma[j];
+    bt[j] = beta[j];
+  }
   int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; r < M; r += stride) {
     const short* xp = x + r * N;
     const short* ap = a + r * N;
-    float zv[64];
+    float zv[N];
     float mean = 0.0f;
+#pragma unroll
     for (int j = 0; j < N; ++j) {
       zv[j] = bf2f_u16(xp[j]) + bf2f_u16(ap[j]);
       mean += zv[j];
     }
     mean /= N;
     float var = 0.0f;
+#pragma unroll
     for (int j = 0; j < N; ++j) {
       float d = zv[j] - mean;
       var += d * d;
@@ -659,31 +669,41 @@ __global__ void k_resln_fwd(const short* __restrict__ x,
     float rstd = rsqrtf(var / N + eps);
     short* yp = y + r * N;
     short* zp = z + r * N;
+#pragma unroll
     for (int j = 0; j < N; ++j) {
       zp[j] = f2bf_u16(zv[j]);
-      yp[j] = f2bf_u16((zv[j] - mean) * rstd * gamma[j] + beta[j]);
+      yp[j] = f2bf_u16((zv[j] - mean) * rstd * gm[j] + bt[j]);
     }
     stats[r * 2] = mean;
     stats[r * 2 + 1] = rstd;
   }
 }
 
-// backward: dz per row (flows to BOTH residual inputs); dgamma/dbeta via
-// LDS block partials + one global atomic per (block, column).
+// backward: dz per row (flows to BOTH residual inputs). dgamma/dbeta
+// partials accumulate in REGISTERS across this thread's rows; one LDS
+// merge + one global atomic per (block, column) at the end.
+template <int N>
 __global__ void k_resln_bwd(const short* __restrict__ dy,
                             const short* __restrict__ z,
                             const float* __restrict__ stats,
                             const float* __restrict__ gamma, int64_t M,
-                            int N, short* __restrict__ dz,
+                            short* __restrict__ dz,
                             float* __restrict__ dgamma,
                             float* __restrict__ dbeta) {
-  __shared__ float ldg[64];
-  __shared__ float ldb[64];
+  __shared__ float ldg[N];
+  __shared__ float ldb[N];
   for (int j = threadIdx.x; j < N; j += blockDim.x) {
     ldg[j] = 0.0f;
     ldb[j] = 0.0f;
   }
   __syncthreads();
+  float gm[N], pg[N], pb[N];
+#pragma unroll
+  for (int j = 0; j < N; ++j) {
+    gm[j] = gamma[j];
+    pg[j] = 0.0f;
+    pb[j] = 0.0f;
+  }
   int64_t r = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; r < M; r += stride) {
@@ -692,21 +712,28 @@ __global__ void k_resln_bwd(const short* __restrict__ dy,
     const float mean = stats[r * 2];
     const float rstd = stats[r * 2 + 1];
     float s1 = 0.0f, s2 = 0.0f;
-    float xh[64], g[64];
+    float xh[N], g[N];
+#pragma unroll
     for (int j = 0; j < N; ++j) {
       xh[j] = (bf2f_u16(zp[j]) - mean) * rstd;
       g[j] = bf2f_u16(dyp[j]);
-      float gg = g[j] * gamma[j];
+      float gg = g[j] * gm[j];
       s1 += gg;
       s2 += gg * xh[j];
-      atomicAdd(&ldg[j], g[j] * xh[j]);
-      atomicAdd(&ldb[j], g[j]);
+      pg[j] += g[j] * xh[j];
+      pb[j] += g[j];
     }
     s1 /= N;
     s2 /= N;
     short* dzp = dz + r * N;
+#pragma unroll
     for (int j = 0; j < N; ++j)
-      dzp[j] = f2bf_u16((g[j] * gamma[j] - s1 - xh[j] * s2) * rstd);
+      dzp[j] = f2bf_u16((g[j] * gm[j] - s1 - xh[j] * s2) * rstd);
+  }
+#pragma unroll
+  for (int j = 0; j < N; ++j) {
+    atomicAdd(&ldg[j], pg[j]);
+    atomicAdd(&ldb[j], pb[j]);
   }
   __syncthreads();
   for (int j = threadIdx.x; j < N; j += blockDim.x) {
@@ -971,11 +998,17 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> resln_fwd(
   auto z = torch::empty_like(x);
   auto stats = torch::empty({M, 2}, x.options().dtype(torch::kFloat32));
   if (M == 0) return {y, z, stats};
+  TORCH_CHECK(N == 16 || N == 32 || N == 64, "resln: N must be 16/32/64");
   int blocks = (int)std::min<int64_t>((M + 255) / 256, 4096);
-  k_resln_fwd<<<blocks, 256, 0, dense_stream()>>>(
-      bf_ptr(x), bf_ptr(a), gamma.data_ptr<float>(),
-      beta.data_ptr<float>(), M, N, (float)eps, bf_ptr_mut(y),
-      bf_ptr_mut(z), stats.data_ptr<float>());
+#define RESLN_FWD(NV) \
+  k_resln_fwd<NV><<<blocks, 256, 0, dense_stream()>>>( \
+      bf_ptr(x), bf_ptr(a), gamma.data_ptr<float>(), \
+      beta.data_ptr<float>(), M, (float)eps, bf_ptr_mut(y), \
+      bf_ptr_mut(z), stats.data_ptr<float>())
+  if (N == 16) RESLN_FWD(16);
+  else if (N == 32) RESLN_FWD(32);
+  else RESLN_FWD(64);
+#undef RESLN_FWD
   return {y, z, stats};
 }
 
@@ -993,11 +1026,18 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> resln_bwd(
   k_zero_f32d<<<1, 64, 0, stream>>>(dbeta.data_ptr<float>(), N);
   if (M == 0) return {dz, dgamma, dbeta};
   auto dyc = dy.contiguous();
-  int blocks = (int)std::min<int64_t>((M + 255) / 256, 4096);
-  k_resln_bwd<<<blocks, 256, 0, stream>>>(
-      bf_ptr(dyc), bf_ptr(z), stats.data_ptr<float>(),
-      gamma.data_ptr<float>(), M, N, bf_ptr_mut(dz),
-      dgamma.data_ptr<float>(), dbeta.data_ptr<float>());
+  // ~8 rows per thread amortize the dgamma/dbeta partial merges
+  int blocks = (int)std::min<int64_t>((M + 256 * 8 - 1) / (256 * 8), 4096);
+  blocks = std::max(blocks, 64);
+#define RESLN_BWD(NV) \
+  k_resln_bwd<NV><<<blocks, 256, 0, stream>>>( \
+      bf_ptr(dyc), bf_ptr(z), stats.data_ptr<float>(), \
+      gamma.data_ptr<float>(), M, bf_ptr_mut(dz), \
+      dgamma.data_ptr<float>(), dbeta.data_ptr<float>())
+  if (N == 16) RESLN_BWD(16);
+  else if (N == 32) RESLN_BWD(32);
+  else RESLN_BWD(64);
+#undef RESLN_BWD
   return {dz, dgamma, dbeta};
 }
 
