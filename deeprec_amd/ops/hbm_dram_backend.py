@@ -53,6 +53,33 @@ class _ColdView:
         return self._p._cold_slab(name, width, init_value, dtype)
 
 
+class _ColdGpuView:
+    """GPU-resident working copy of selected cold rows: the fused sparse
+    applies run unmodified against it (values + slot-aligned slabs), then
+    flush() scatters everything back into the pinned cold slabs."""
+
+    def __init__(self, parent: "HbmDramStorage", cs: torch.Tensor):
+        self._p = parent
+        self._cs = cs  # int64 cuda, cold-slab row indices
+        self.dim = parent.dim
+        self.values = parent.ext.gather_host_rows(parent.values_cold, cs)
+        self._gathered = {}
+
+    def get_slab(self, name, width, init_value, dtype=torch.float32):
+        if name not in self._gathered:
+            host = self._p._cold_slab(name, width, init_value, dtype)
+            self._gathered[name] = self._p.ext.gather_host_rows(host,
+                                                                self._cs)
+        return self._gathered[name]
+
+    def flush(self):
+        self._p.ext.scatter_host_rows(self.values, self._cs,
+                                      self._p.values_cold)
+        for name, rows in self._gathered.items():
+            self._p.ext.scatter_host_rows(rows, self._cs,
+                                          self._p.cold_slabs[name])
+
+
 class HbmDramStorage(HbmStorage):
     def __init__(self, dim: int, ev_option: EmbeddingVariableOption,
                  value_dtype=torch.float32, device=None, generator=None):
@@ -194,9 +221,15 @@ class HbmDramStorage(HbmStorage):
         if bool(hot.any()):
             out[hot] = self.values[slots[hot].long()]
         if bool(cold.any()):
-            cs = (slots[cold].cpu().long() - self.hot_rows)
-            staged = self.values_cold[cs]  # host gather (pinned source)
-            out[cold] = staged.to(self.device, non_blocking=True)
+            cs = slots[cold].long() - self.hot_rows
+            if self.values_cold.is_pinned():
+                # zero-copy: the GPU gathers cold rows straight out of
+                # pinned DRAM (one kernel at interconnect bandwidth;
+                # the CPU-gather path measured 0.4 GB/s)
+                out[cold] = self.ext.gather_host_rows(self.values_cold, cs)
+            else:  # mmap SSD tier: page-cache-bound host gather
+                staged = self.values_cold[cs.cpu()]
+                out[cold] = staged.to(self.device, non_blocking=True)
         if bool(none.any()):
             ku = keys[none]
             out[none] = self.ext.ev_gather(
@@ -231,10 +264,24 @@ class HbmDramStorage(HbmStorage):
             hip_backend.sparse_apply(name, self, slots[hot].to(torch.int32),
                                      grad[hot], dict(hyper))
         if bool(cold.any()):
-            cold_slots = (slots[cold].cpu().long() - self.hot_rows)
-            cold_grad = grad[cold].cpu().float()
-            getattr(sparse_optim_cpu, f"apply_{name}")(
-                _ColdView(self), cold_slots, cold_grad, **hyper)
+            if self.values_cold.is_pinned():
+                # zero-copy round trip: gather cold rows + optimizer
+                # slabs to GPU, run the SAME fused apply kernels as the
+                # hot tier, scatter back — the cold tier stays
+                # interconnect-bound instead of CPU-bound
+                cs = slots[cold].long() - self.hot_rows
+                view = _ColdGpuView(self, cs)
+                m = cs.numel()
+                hip_backend.sparse_apply(
+                    name, view,
+                    torch.arange(m, dtype=torch.int32, device=self.device),
+                    grad[cold], dict(hyper))
+                view.flush()
+            else:  # mmap SSD tier
+                cold_slots = (slots[cold].cpu().long() - self.hot_rows)
+                cold_grad = grad[cold].cpu().float()
+                getattr(sparse_optim_cpu, f"apply_{name}")(
+                    _ColdView(self), cold_slots, cold_grad, **hyper)
 
     # ---------------- export / import / shrink ----------------
     def shrink(self, step: int) -> int:

@@ -224,28 +224,39 @@ def test_flat_dense_adam_matches_torch_adam():
         (m(x) ** 2).mean().backward()
         o.step()
 
-    # After ONE step: TF-style (sqrt(v)+eps) and torch-style
-    # (sqrt(v/bc2)+eps) Adam differ ANALYTICALLY for near-zero gradients
-    # (|g| between eps and eps/sqrt(1-beta2) gets a materially different
-    # update ratio), so compare only saturated updates (|u| near lr —
-    # the overwhelming majority). Exact kernel correctness against the
+    # After ONE step: Adam's first update is ~ +-lr * sign(g), so any
+    # element whose gradient is at fp32-atomic-rounding noise level can
+    # flip SIGN between the two models' (independently accumulated)
+    # backward passes — and the eps-placement styles also diverge for
+    # |g| < eps/sqrt(1-beta2). Compare only where the reference gradient
+    # is decisively nonzero. Exact kernel correctness against the
     # TF-style formula is test_dense_adam_kernel_exact.
     before = [pb.detach().clone() for pb in mb.parameters()]
-    one_step(ma, oa, 0)
-    one_step(mb, ob, 0)
-    for pa, pb, p0 in zip(ma.parameters(), mb.parameters(), before):
-        u_ref = pb.detach() - p0
-        mask = u_ref.abs() > 0.005  # half of the +-lr saturation
-        assert mask.float().mean() > 0.8
+    x0 = torch.randn(128, 16, device="cuda",
+                     generator=torch.Generator("cuda").manual_seed(0))
+    oa.zero_grad()
+    (ma(x0) ** 2).mean().backward()
+    ob.zero_grad()
+    (mb(x0) ** 2).mean().backward()
+    gmasks = [pb.grad.abs() > 1e-4 for pb in mb.parameters()]
+    oa.step()
+    ob.step()
+    for pa, pb, p0, mask in zip(ma.parameters(), mb.parameters(), before,
+                                gmasks):
+        assert mask.float().mean() > 0.5
         torch.testing.assert_close(pa.detach()[mask], pb.detach()[mask],
                                    rtol=1e-3, atol=1e-4)
-    # several more steps: trajectories stay bounded-close (per-element
-    # divergence at relu/eps boundaries compounds chaotically)
+    # several more steps: trajectories stay statistically close (per-
+    # element divergence at sign/relu/eps boundaries compounds
+    # chaotically, so bound the relative Frobenius distance instead)
     for step in range(1, 5):
         one_step(ma, oa, step)
         one_step(mb, ob, step)
     for pa, pb in zip(ma.parameters(), mb.parameters()):
-        torch.testing.assert_close(pa, pb, rtol=0.5, atol=3e-2)
+        num = (pa.detach() - pb.detach()).norm()
+        den = pb.detach().norm().clamp(min=1e-6)
+        assert float(num / den) < 0.2, float(num / den)
+        assert torch.isfinite(pa).all()
     # shadows track the master weights
     for mod in ma.modules():
         if hasattr(mod, "w16_cache") and mod.w16_cache is not None:
