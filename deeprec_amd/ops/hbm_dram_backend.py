@@ -62,6 +62,7 @@ class _ColdGpuView:
         self._p = parent
         self._cs = cs  # int64 cuda, cold-slab row indices
         self.dim = parent.dim
+        self.ext = parent.ext
         self.values = parent.ext.gather_host_rows(parent.values_cold, cs)
         self._gathered = {}
 
