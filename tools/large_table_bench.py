@@ -107,12 +107,20 @@ def train_steps(ev, args, optimizer="adagrad", label=""):
     gen = torch.Generator("cuda").manual_seed(7)
     hyper = {"lr": 0.01, "initial_accumulator": 0.1, "epsilon": 1e-8}
     times = []
-    for step in range(args.warmup + args.steps):
+    parts = [0.0, 0.0, 0.0]  # lookup, gather, apply (timed steps only)
+
+    def tick():
         torch.cuda.synchronize()
-        t0 = time.perf_counter()
+        return time.perf_counter()
+
+    for step in range(args.warmup + args.steps):
+        timed_step = step >= args.warmup
+        t0 = tick()
         keys = batch_ids(args.batch, args.per_sample, args.ids, gen=gen)
         uniq, inverse, counts, slots = ev.storage.dedup_lookup(keys, step)
+        t1 = tick()
         emb = ev.storage.gather(uniq, slots)
+        t2 = tick()
         grad = emb * 1e-4  # decay-shaped grads keep values bounded
         if hasattr(ev.storage, "apply_split"):
             ev.storage.apply_split(optimizer, slots, grad, dict(hyper))
@@ -120,13 +128,20 @@ def train_steps(ev, args, optimizer="adagrad", label=""):
             hip_backend.sparse_apply(optimizer, ev.storage,
                                      slots.to(torch.int32), grad,
                                      dict(hyper))
-        torch.cuda.synchronize()
-        times.append(time.perf_counter() - t0)
+        t3 = tick()
+        times.append(t3 - t0)
+        if timed_step:
+            parts[0] += t1 - t0
+            parts[1] += t2 - t1
+            parts[2] += t3 - t2
     timed = times[args.warmup:]
     ms = sum(timed) / len(timed) * 1000
+    n = len(timed)
     sps = args.batch / (ms / 1000)
     log(f"{label} step: {ms:.2f} ms  ({sps/1e6:.2f}M samples/s, "
-        f"nnz/step={args.batch * args.per_sample})")
+        f"nnz/step={args.batch * args.per_sample}; "
+        f"lookup {parts[0]/n*1000:.2f} / gather {parts[1]/n*1000:.2f} / "
+        f"apply {parts[2]/n*1000:.2f} ms)")
     return ms, sps
 
 
