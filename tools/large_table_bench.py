@@ -100,6 +100,52 @@ def batch_ids(batch, per_sample, ids, hot_frac=0.8, hot_space=1 << 20,
     return torch.where(pick, hot, cold)
 
 
+def debug_lookup(ev, args, label=""):
+    """One finely-timed dedup_lookup to localize host/device cost."""
+    st = ev.storage
+    gen = torch.Generator("cuda").manual_seed(123)
+    keys = batch_ids(args.batch, args.per_sample, args.ids, gen=gen)
+    torch.cuda.synchronize()
+    marks = [("start", time.perf_counter())]
+
+    def mark(name):
+        torch.cuda.synchronize()
+        marks.append((name, time.perf_counter()))
+
+    nnz = keys.numel()
+    prev = st._watermark()
+    mark("watermark")
+    st._ensure_capacity(nnz)
+    mark("ensure_cap")
+    st._epoch += 1
+    uniq_buf = torch.empty(nnz, dtype=torch.int64, device=st.device)
+    centry_buf = torch.empty(nnz, dtype=torch.int64, device=st.device)
+    m_counter = torch.zeros(1, dtype=torch.int32, device=st.device)
+    mark("alloc")
+    st.ext.ht_dedup_a(keys, st.ht_keys, st.ht_freq, st.ht_version,
+                      st.ht_epoch, st.ht_compact, st._epoch, 0,
+                      st.entry_counter, m_counter, uniq_buf, centry_buf,
+                      st.error_flag)
+    mark("pass_a")
+    c = torch.cat([m_counter, st.entry_counter, st.slot_counter]).cpu()
+    m = int(c[0])
+    mark("sync_m")
+    inverse, counts, rank = st.ext.ht_dedup_c(keys, st.ht_keys,
+                                              st.ht_compact, m)
+    mark("pass_c")
+    slots = st.ext.ht_dedup_b(
+        centry_buf[:m], uniq_buf[:m], st.ht_slot, st.ht_freq,
+        st.ht_version, counts, 0, st.slot_counter, st.max_slots,
+        st.values, st.default_values, st.dvd_per_table, st.key_bits,
+        st._init_limit(), st.filter_freq, st.error_flag)
+    mark("pass_b")
+    st._init_cold_rows(uniq_buf[:m], slots, prev)
+    mark("cold_init")
+    parts = "  ".join(f"{n}={1000*(t1-t0):.2f}ms" for (_, t0), (n, t1)
+                      in zip(marks, marks[1:]))
+    log(f"{label} lookup breakdown (m={m}): {parts}")
+
+
 def train_steps(ev, args, optimizer="adagrad", label=""):
     """Training-shaped engine steps: dedup+probe+admit, gather, grad
     scatter-equivalent (rows += g), fused sparse apply."""
@@ -188,6 +234,7 @@ def phase_tier(args):
     cold_rows = args.ids - st.hot_rows
     log(f"hot rows {st.hot_rows/1e6:.1f}M, cold rows ~{cold_rows/1e6:.1f}M")
     ms, sps = train_steps(ev, args, label="tier")
+    debug_lookup(ev, args, label="tier")
     # staging bandwidth: materialize a pure-cold batch
     gen = torch.Generator("cuda").manual_seed(9)
     m = 1 << 20
@@ -214,6 +261,7 @@ def phase_tier(args):
     reb_dt = time.perf_counter() - t0
     log(f"rebalance: {moved/1e6:.2f}M rows changed tier in {reb_dt:.2f}s")
     ms2, sps2 = train_steps(ev, args, label="tier(after-rebalance)")
+    debug_lookup(ev, args, label="tier(after-rebalance)")
     mu = st.memory_usage()
     out = {"phase": "tier", "ids": args.ids, "dim": dim,
            "hot_gb": args.hot_gb, "hot_rows": int(st.hot_rows),
