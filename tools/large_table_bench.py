@@ -139,11 +139,16 @@ def debug_lookup(ev, args, label=""):
         st.values, st.default_values, st.dvd_per_table, st.key_bits,
         st._init_limit(), st.filter_freq, st.error_flag)
     mark("pass_b")
+    cold_new = (slots >= st.hot_rows) & (slots >= max(st.hot_rows, prev))
+    n_cold_new = int(cold_new.sum())
+    mark("cold_mask")
     st._init_cold_rows(uniq_buf[:m], slots, prev)
     mark("cold_init")
     parts = "  ".join(f"{n}={1000*(t1-t0):.2f}ms" for (_, t0), (n, t1)
                       in zip(marks, marks[1:]))
-    log(f"{label} lookup breakdown (m={m}): {parts}")
+    log(f"{label} lookup breakdown (m={m}, prev={prev}, "
+        f"slot_counter={int(st.slot_counter.cpu())}, "
+        f"max_slot={int(slots.max())}, n_cold_new={n_cold_new}): {parts}")
 
 
 def train_steps(ev, args, optimizer="adagrad", label=""):
