@@ -66,6 +66,10 @@ DEV_INLINE __hip_bfloat16 f2bf(float v) { return __float2bfloat16(v); }
 //   - admit a value slot once freq >= filter_freq; initialize the value
 //     row from default_values[key % default_value_dim]
 // out_slots[i] = slot or -1 (not admitted).
+// NOTE every kernel whose grid n_blocks() caps at 65535 blocks MUST
+// grid-stride: a plain `if (i >= n) return` silently dropped the tail of
+// batches beyond 16.7M keys (found when a 120M-key rebalance rebuild
+// lost most of the table).
 __global__ void k_lookup_insert(
     const int64_t* __restrict__ keys, const int32_t* __restrict__ counts,
     int n, int64_t* __restrict__ ht_keys, int32_t* __restrict__ ht_slot,
@@ -76,8 +80,9 @@ __global__ void k_lookup_insert(
     int dim, int default_value_dim, int key_bits, int init_limit,
     int filter_freq, int64_t step, int train,
     int32_t* __restrict__ out_slots, int32_t* __restrict__ error_flag) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t gstride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += gstride) {
   const int64_t key = keys[i];
   uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
   int32_t slot = -1;
@@ -86,7 +91,7 @@ __global__ void k_lookup_insert(
     int64_t cur = ht_keys[idx];
     if (cur != key) {
       if (cur != EMPTY_KEY) continue;
-      if (!train) { out_slots[i] = -1; return; }
+      if (!train) { out_slots[i] = -1; goto done_i; }
       int64_t prev = (int64_t)atomicCAS(
           (unsigned long long*)&ht_keys[idx],
           (unsigned long long)EMPTY_KEY, (unsigned long long)key);
@@ -106,14 +111,14 @@ __global__ void k_lookup_insert(
       if (slot >= max_slots) {  // out of slab space: host must grow
         atomicExch(error_flag, 1);
         out_slots[i] = -1;
-        return;
+        goto done_i;
       }
       ht_slot[idx] = slot;
       // multi-tier (HBM_DRAM): rows at slot >= init_limit live in the host
       // cold slab; the host initializes those (kernel must not touch them)
       if (slot >= init_limit) {
         out_slots[i] = slot;
-        return;
+        goto done_i;
       }
       // composite keys (EmbeddingCollection): default row is
       // table * dvd + (raw_key % dvd); key_bits == 0 means plain keys
@@ -131,10 +136,12 @@ __global__ void k_lookup_insert(
       for (int d = 0; d < dim; ++d) dst[d] = src[d];
     }
     out_slots[i] = slot;
-    return;
+    goto done_i;
   }
   atomicExch(error_flag, 2);  // table full (host sizing bug)
   out_slots[i] = -1;
+  done_i:;
+  }
 }
 
 // Bulk import used by restore: entries are created admitted with the given
@@ -146,27 +153,30 @@ __global__ void k_insert_bulk(
     int32_t* __restrict__ ht_freq, int64_t* __restrict__ ht_version,
     int64_t cap_mask, int32_t* __restrict__ entry_counter,
     int32_t* __restrict__ error_flag) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  const int64_t key = keys[i];
-  uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
-  for (int64_t probe = 0; probe <= cap_mask; ++probe) {
-    int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
-    int64_t cur = ht_keys[idx];
-    if (cur != key) {
-      if (cur != EMPTY_KEY) continue;
-      int64_t prev = (int64_t)atomicCAS(
-          (unsigned long long*)&ht_keys[idx],
-          (unsigned long long)EMPTY_KEY, (unsigned long long)key);
-      if (prev != EMPTY_KEY && prev != key) continue;
-      if (prev == EMPTY_KEY) atomicAdd(entry_counter, 1);
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t gstride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += gstride) {
+    const int64_t key = keys[i];
+    uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
+    bool placed = false;
+    for (int64_t probe = 0; probe <= cap_mask && !placed; ++probe) {
+      int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
+      int64_t cur = ht_keys[idx];
+      if (cur != key) {
+        if (cur != EMPTY_KEY) continue;
+        int64_t prev = (int64_t)atomicCAS(
+            (unsigned long long*)&ht_keys[idx],
+            (unsigned long long)EMPTY_KEY, (unsigned long long)key);
+        if (prev != EMPTY_KEY && prev != key) continue;
+        if (prev == EMPTY_KEY) atomicAdd(entry_counter, 1);
+      }
+      ht_slot[idx] = slots[i];
+      if (freqs) ht_freq[idx] = freqs[i];
+      if (versions) ht_version[idx] = versions[i];
+      placed = true;
     }
-    ht_slot[idx] = slots[i];
-    if (freqs) ht_freq[idx] = freqs[i];
-    if (versions) ht_version[idx] = versions[i];
-    return;
+    if (!placed) atomicExch(error_flag, 2);
   }
-  atomicExch(error_flag, 2);
 }
 
 // ---------------------------------------------------------------------
@@ -295,8 +305,9 @@ __global__ void k_dedup_pass_b(
     int dim, int default_value_dim, int key_bits, int init_limit,
     int filter_freq, int32_t* __restrict__ out_slots,
     int32_t* __restrict__ error_flag) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= m) return;
+  int64_t c = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t gstride = gridDim.x * (int64_t)blockDim.x;
+  for (; c < m; c += gstride) {
   int64_t idx = compact_entry[c];
   ht_freq[idx] += batch_counts[c];
   ht_version[idx] = step;
@@ -306,7 +317,7 @@ __global__ void k_dedup_pass_b(
     if (slot >= max_slots) {
       atomicExch(error_flag, 1);
       out_slots[c] = -1;
-      return;
+      continue;
     }
     ht_slot[idx] = slot;
     if (slot < init_limit) {
@@ -326,6 +337,7 @@ __global__ void k_dedup_pass_b(
     }
   }
   out_slots[c] = slot;
+  }
 }
 
 // Padded pass B for hipGraph capture: grid covers n_cap (= nnz upper
@@ -691,22 +703,28 @@ __global__ void k_lookup(
     const int64_t* __restrict__ ht_keys, const int32_t* __restrict__ ht_slot,
     int64_t cap_mask, int32_t* __restrict__ out_slots,
     int64_t* __restrict__ out_entry) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= n) return;
-  const int64_t key = keys[i];
-  uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
-  for (int64_t probe = 0; probe <= cap_mask; ++probe) {
-    int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
-    int64_t cur = ht_keys[idx];
-    if (cur == key) {
-      out_slots[i] = ht_slot[idx];
-      if (out_entry) out_entry[i] = idx;
-      return;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t gstride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < n; i += gstride) {
+    const int64_t key = keys[i];
+    uint64_t h = mix_hash((uint64_t)key) & (uint64_t)cap_mask;
+    bool found = false;
+    for (int64_t probe = 0; probe <= cap_mask; ++probe) {
+      int64_t idx = (int64_t)((h + probe) & (uint64_t)cap_mask);
+      int64_t cur = ht_keys[idx];
+      if (cur == key) {
+        out_slots[i] = ht_slot[idx];
+        if (out_entry) out_entry[i] = idx;
+        found = true;
+        break;
+      }
+      if (cur == EMPTY_KEY) break;
     }
-    if (cur == EMPTY_KEY) break;
+    if (!found) {
+      out_slots[i] = -1;
+      if (out_entry) out_entry[i] = -1;
+    }
   }
-  out_slots[i] = -1;
-  if (out_entry) out_entry[i] = -1;
 }
 
 // Export scan: compact live hash entries into dense output arrays.
@@ -1049,10 +1067,12 @@ __global__ void k_apply_adagrad_decay(
 __global__ void k_commit_period(float* __restrict__ period_slab,
                                 const int32_t* __restrict__ slots, int m,
                                 float cur_period) {
-  int i = blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= m) return;
-  int32_t s = slots[i];
-  if (s >= 0) period_slab[s] = cur_period;
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t gstride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < m; i += gstride) {
+    int32_t sl = slots[i];
+    if (sl >= 0) period_slab[sl] = cur_period;
+  }
 }
 
 __global__ void k_apply_adam(float* __restrict__ w, float* __restrict__ mom,

@@ -229,3 +229,31 @@ def test_memory_usage_gpu():
     assert mu["slab_bytes"] >= 2 * mu["values_bytes"] // 2
     assert mu["total_bytes"] == sum(
         v for k, v in mu.items() if k != "total_bytes")
+
+
+@pytest.mark.gpu
+def test_bulk_insert_lookup_beyond_grid_cap():
+    """Regression: kernels whose grid caps at 65535 blocks must
+    grid-stride — a 120M-key rebalance rebuild silently lost every key
+    past 16.7M (insert_bulk/lookup had no stride loop)."""
+    from deeprec_amd.embedding.options import EmbeddingVariableOption
+    from deeprec_amd.ops.hip_backend import HbmStorage
+
+    n = 20_000_000  # > 65535 * 256
+    st = HbmStorage(8, EmbeddingVariableOption(init_capacity=1 << 26),
+                    device="cuda")
+    st._grow_slots(n)
+    keys = torch.arange(n, dtype=torch.int64, device="cuda")
+    slots = torch.arange(n, dtype=torch.int32, device="cuda")
+    st.ext.ht_insert_bulk(keys, slots, torch.Tensor(), torch.Tensor(),
+                          st.ht_keys, st.ht_slot, st.ht_freq,
+                          st.ht_version, st.entry_counter, st.error_flag)
+    st._check_error()
+    assert st.total_count() == n
+    # probe the TAIL (the region the capped grid used to drop)
+    probe = torch.arange(n - 1_000_000, n, dtype=torch.int64,
+                         device="cuda")
+    got = st.lookup(probe)
+    assert bool((got >= 0).all()), "tail keys lost by bulk insert"
+    torch.testing.assert_close(got.long(), probe - (n - 1_000_000)
+                               + (n - 1_000_000))
