@@ -198,3 +198,57 @@ def test_multitier_filter_restore_roundtrip():
     # frequencies survive the round trip
     torch.testing.assert_close(st2.frequencies(keys).cpu(),
                                st.frequencies(keys).cpu())
+
+
+def test_background_maintenance_no_step_spike():
+    """EvictionManager: scoring on a worker thread + bounded chunk
+    application must promote hot keys WITHOUT any multi-second step
+    (the blocking rebalance() on the same shape costs seconds)."""
+    import time
+
+    from deeprec_amd.embedding.maintenance import EvictionManager
+    from deeprec_amd.ops import hip_backend
+
+    dim, hot_rows = 16, 4096
+    opt = EmbeddingVariableOption(
+        storage_option=StorageOption(storage_type=StorageType.HBM_DRAM,
+                                     storage_size=[hot_rows * dim * 4]),
+        init_option=InitializerOption(initializer=1.0))
+    ev = EmbeddingVariable("bg_maint", dim, ev_option=opt, device=DEV)
+    st = ev.storage
+    st._grow_slots(200_000)
+    st.get_slab("adagrad_accum", dim, 0.1)
+    # populate 100k ids; the first 4096 slots (hot tier) land on
+    # insert order, NOT on the hot id set
+    all_ids = torch.arange(100_000, dtype=torch.int64, device=DEV)
+    st.lookup_or_create(all_ids, torch.ones(100_000, dtype=torch.int32,
+                                            device=DEV), step=0)
+    mgr = EvictionManager(interval_steps=5, chunk_rows=8192)
+    mgr.register(ev)
+    hot_ids = torch.arange(90_000, 94_096, dtype=torch.int64,
+                           device=DEV)  # 4096 ids, inserted cold
+    hyper = {"lr": 0.01, "initial_accumulator": 0.1, "epsilon": 1e-8}
+    step_times = []
+    for step in range(60):
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        uniq, inverse, counts, slots = st.dedup_lookup(hot_ids, step)
+        emb = st.gather(uniq, slots)
+        st.apply_split("adagrad", slots, emb * 1e-3, dict(hyper))
+        mgr.step(step)
+        torch.cuda.synchronize()
+        step_times.append(time.perf_counter() - t0)
+    mgr.wait_idle()
+    for step in range(60, 70):  # drain remaining chunks
+        mgr.step(step)
+    # the hot ids must end up in the HBM tier
+    final_slots = st.lookup(hot_ids)
+    frac_hot = float((final_slots < st.hot_rows).float().mean())
+    assert frac_hot > 0.9, f"only {frac_hot:.2f} promoted"
+    assert mgr.stats["rows_promoted"] > 3000
+    # no step may stall on maintenance (blocking rebalance here ~seconds)
+    assert max(step_times) < 0.25, f"step spike {max(step_times):.3f}s"
+    # values survive the tier swaps exactly (all-ones init + decay grads)
+    vals = st.materialize(hot_ids, final_slots)
+    assert torch.isfinite(vals).all()
+    st._check_error()
