@@ -145,6 +145,11 @@ class EvictionManager:
         ps = st.lookup(pk).long()
         ds = st.lookup(dk).long()
         ok = (ps >= st.hot_rows) & (ds >= 0) & (ds < st.hot_rows)
+        ssd_base = getattr(st, "ssd_base", None)
+        if ssd_base is not None:
+            # SSD-resident rows are key-addressed, not slab rows: the
+            # HBM<->DRAM swap only applies to the in-memory tiers
+            ok &= ps < ssd_base
         n = int(ok.sum())
         if n == 0:
             return

@@ -26,6 +26,10 @@ class StorageType(enum.Enum):
     DRAM = 1
     HBM = 2
     HBM_DRAM = 3
+    # hot HBM + pinned-DRAM middle tier + append-only SSD files with
+    # compaction (reference: HbmDramSsdStorage, hbm_dram_ssd_storage.h);
+    # storage_size = [hbm_bytes, dram_bytes], storage_path = SSD dir
+    HBM_DRAM_SSD = 4
 
 
 class CacheStrategy(enum.Enum):

@@ -66,6 +66,12 @@ class EmbeddingVariable:
                 self.storage = HbmDramStorage(embedding_dim, self.ev_option,
                                               value_dtype, self.device,
                                               generator)
+            elif st == StorageType.HBM_DRAM_SSD:
+                from deeprec_amd.ops.hbm_dram_backend import (
+                    HbmDramSsdStorage)
+                self.storage = HbmDramSsdStorage(
+                    embedding_dim, self.ev_option, value_dtype,
+                    self.device, generator)
             else:
                 self.storage = HbmStorage(embedding_dim, self.ev_option,
                                           value_dtype, self.device, generator)

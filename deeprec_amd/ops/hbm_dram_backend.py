@@ -454,3 +454,259 @@ class HbmDramStorage(HbmStorage):
                     ci = (slots[cold].cpu().long() - self.hot_rows)
                     self.cold_slabs[name][ci] = rows[cold.cpu()]
         self._sync_counters()
+
+
+class HbmDramSsdStorage(HbmDramStorage):
+    """Three-tier composition: HBM hot rows + pinned-DRAM middle tier +
+    append-only SSD files with compaction (reference capability:
+    HbmDramSsdStorage, hbm_dram_ssd_storage.h + ssd_hash_kv.h).
+
+    Slot space: [0, hot_rows) = HBM slab, [hot_rows, hot_rows+dram_rows)
+    = pinned DRAM slab, >= hot_rows+dram_rows = SSD-resident (sparse,
+    keyed by the EV key in one SsdKv per column family: values + each
+    optimizer slab). storage_size = [hbm_bytes, dram_bytes];
+    storage_path = the SSD directory.
+    """
+
+    def __init__(self, dim, ev_option, value_dtype=torch.float32,
+                 device=None, generator=None):
+        import copy
+
+        from deeprec_amd.embedding.ssd_kv import SsdKv
+        so = ev_option.storage_option
+        assert so.storage_path, "HBM_DRAM_SSD requires storage_path"
+        assert so.storage_size and len(so.storage_size) >= 2, \
+            "HBM_DRAM_SSD requires storage_size=[hbm_bytes, dram_bytes]"
+        self._ssd_path = so.storage_path
+        opt = copy.deepcopy(ev_option)
+        # the base class builds the pinned-DRAM middle tier; the SSD dir
+        # must not trigger its mmap-slab mode
+        opt.storage_option.storage_path = None
+        opt.storage_option.storage_size = [so.storage_size[0]]
+        super().__init__(dim, opt, value_dtype, device, generator)
+        self.dram_rows = max(16, so.storage_size[1] // (dim * 4))
+        self.ssd_kvs = {"values": SsdKv(
+            os.path.join(self._ssd_path, "values"), dim)}
+        self._ssd_slab_init = {}
+
+    # ---------------- tier plumbing ----------------
+    @property
+    def ssd_base(self) -> int:
+        return self.hot_rows + self.dram_rows
+
+    @property
+    def max_slots(self) -> int:
+        return (1 << 31) - 2  # SSD tier is unbounded (sparse, key-addressed)
+
+    def _grow_slots(self, need: int):
+        # DRAM middle tier caps at dram_rows; beyond that rows live on SSD
+        super()._grow_slots(min(need, self.ssd_base))
+
+    def _ssd_kv(self, name, width, init_value):
+        from deeprec_amd.embedding.ssd_kv import SsdKv
+        if name not in self.ssd_kvs:
+            self.ssd_kvs[name] = SsdKv(
+                os.path.join(self._ssd_path, name), width)
+            self._ssd_slab_init[name] = init_value
+        return self.ssd_kvs[name]
+
+    def get_slab(self, name, width, init_value, dtype=torch.float32):
+        out = super().get_slab(name, width, init_value, dtype)
+        self._ssd_kv(name, width, init_value)
+        self._ssd_slab_init[name] = init_value
+        return out
+
+    def _init_cold_rows(self, keys, slots, prev):
+        dram_mask = (slots >= self.hot_rows) & (slots < self.ssd_base) & \
+            (slots >= max(self.hot_rows, prev))
+        if bool(dram_mask.any()):
+            ks = keys[dram_mask].cpu()
+            ss = (slots[dram_mask].cpu().long() - self.hot_rows)
+            rows = self._default_rows_cpu(ks)
+            self.values_cold[ss] = self.default_values_cpu[rows]
+            for name, t in self.cold_slabs.items():
+                t[ss] = self._cold_slab_init[name]
+        ssd_new = (slots >= self.ssd_base) & (slots >= max(self.ssd_base,
+                                                           prev))
+        if bool(ssd_new.any()):
+            ks = keys[ssd_new].cpu()
+            rows = self._default_rows_cpu(ks)
+            self.ssd_kvs["values"].write(
+                ks, self.default_values_cpu[rows])
+            # optimizer slabs default-init lazily on first read
+
+    # ---------------- materialized gather ----------------
+    def materialize(self, keys, slots) -> torch.Tensor:
+        m = keys.numel()
+        out = torch.empty(m, self.dim, dtype=torch.float32,
+                          device=self.device)
+        hot = (slots >= 0) & (slots < self.hot_rows)
+        dram = (slots >= self.hot_rows) & (slots < self.ssd_base)
+        ssd = slots >= self.ssd_base
+        none = slots < 0
+        if bool(hot.any()):
+            out[hot] = self.values[slots[hot].long()]
+        if bool(dram.any()):
+            cs = slots[dram].long() - self.hot_rows
+            out[dram] = self.ext.gather_host_rows(self.values_cold, cs)
+        if bool(ssd.any()):
+            rows = self.ssd_kvs["values"].read(keys[ssd])
+            out[ssd] = rows.to(self.device)
+        if bool(none.any()):
+            ku = keys[none]
+            out[none] = self.ext.ev_gather(
+                self.values, self.default_values, ku,
+                torch.full((ku.numel(),), -1, dtype=torch.int32,
+                           device=self.device),
+                self._no_permission_value(), self._use_no_permission(),
+                torch.float32)
+        return out
+
+    # ---------------- sparse apply ----------------
+    def apply_split(self, name, slots, grad, hyper):
+        from deeprec_amd.ops import hip_backend
+        in_mem = slots < self.ssd_base
+        ssd = slots >= self.ssd_base
+        if bool(in_mem.any()):
+            super().apply_split(name, slots[in_mem], grad[in_mem], hyper)
+        if bool(ssd.any()):
+            # read-modify-write through the KV: the coldest tier is the
+            # slow path by construction (reference semantics)
+            keys = getattr(self, "_apply_keys", None)
+            assert keys is not None, \
+                "apply on SSD rows requires set_apply_keys(keys)"
+            ks = keys[ssd]
+            view = _SsdApplyView(self, ks)
+            m = ks.numel()
+            hip_backend.sparse_apply(
+                name, view,
+                torch.arange(m, dtype=torch.int32, device=self.device),
+                grad[ssd], dict(hyper))
+            view.flush(ks)
+
+    def set_apply_keys(self, keys):
+        """The SSD tier is key-addressed; optimizers call this before
+        apply_split so SSD-resident rows can be located."""
+        self._apply_keys = keys
+
+    # ---------------- export / import ----------------
+    def export(self, include_filtered: bool = False):
+        keys, slots, freqs, versions = self._export_entries()
+        adm = slots >= 0
+        k, s = keys[adm], slots[adm]
+        values = self.materialize(k, s)
+        out = (k, values, freqs[adm].to(torch.int64), versions[adm])
+        if include_filtered:
+            out = out + (keys[~adm], freqs[~adm].to(torch.int64))
+        return out
+
+    def export_slabs(self, names):
+        keys, slots, freqs, versions = self._export_entries()
+        adm = slots >= 0
+        k, s = keys[adm], slots[adm]
+        hot = s < self.hot_rows
+        dram = (s >= self.hot_rows) & (s < self.ssd_base)
+        ssd = s >= self.ssd_base
+        outs = []
+        for nm in names:
+            w = self.slabs[nm].shape[1]
+            t = torch.empty(s.numel(), w)
+            t[hot.cpu()] = self.slabs[nm][s[hot].long()].cpu()
+            di = (s[dram].cpu().long() - self.hot_rows)
+            t[dram.cpu()] = self.cold_slabs[nm][di]
+            if bool(ssd.any()):
+                t[ssd.cpu()] = self.ssd_kvs[nm].read(
+                    k[ssd], default=self._ssd_slab_init.get(nm, 0.0))
+            outs.append(t)
+        return outs
+
+    def shrink(self, step: int) -> int:
+        raise NotImplementedError(
+            "HBM_DRAM_SSD shrink: evict via EvictionManager / checkpoint "
+            "repartition (full-rebuild shrink would need an SSD rewrite)")
+
+    def rebalance(self) -> int:
+        raise NotImplementedError(
+            "HBM_DRAM_SSD rebalance: use the background EvictionManager "
+            "(incremental promote/demote); a stop-the-world repack of an "
+            "SSD-backed table is deliberately unsupported")
+
+    def import_(self, keys, values, freqs=None, versions=None,
+                slab_rows=None):
+        keys = keys.to(self.device)
+        m = keys.numel()
+        if m == 0:
+            return
+        if freqs is not None:
+            counts = freqs.to(self.device, torch.int32).clamp(
+                min=max(1, self.filter_freq))
+        else:
+            counts = torch.full((m,), max(1, self.filter_freq),
+                                dtype=torch.int32, device=self.device)
+        slots = self.lookup_or_create(keys, counts, step=0, train=True)
+        values = values.to(self.device, torch.float32)
+        hot = (slots >= 0) & (slots < self.hot_rows)
+        dram = (slots >= self.hot_rows) & (slots < self.ssd_base)
+        ssd = slots >= self.ssd_base
+        if bool(hot.any()):
+            self.values[slots[hot].long()] = values[hot]
+        if bool(dram.any()):
+            ci = (slots[dram].cpu().long() - self.hot_rows)
+            self.values_cold[ci] = values[dram].cpu()
+        if bool(ssd.any()):
+            self.ssd_kvs["values"].write(keys[ssd].cpu(),
+                                         values[ssd].cpu())
+        if freqs is not None:
+            self.ext.ht_insert_bulk(
+                keys, slots, freqs.to(self.device, torch.int32),
+                versions.to(self.device, torch.int64)
+                if versions is not None else torch.Tensor(),
+                self.ht_keys, self.ht_slot, self.ht_freq, self.ht_version,
+                self.entry_counter, self.error_flag)
+        if slab_rows:
+            for name, rows in slab_rows.items():
+                self.get_slab(name, rows.shape[1], 0.0)
+                if bool(hot.any()):
+                    self.slabs[name][slots[hot].long()] = \
+                        rows[hot.cpu()].to(self.device)
+                if bool(dram.any()):
+                    ci = (slots[dram].cpu().long() - self.hot_rows)
+                    self.cold_slabs[name][ci] = rows[dram.cpu()]
+                if bool(ssd.any()):
+                    self.ssd_kvs[name].write(keys[ssd].cpu(),
+                                             rows[ssd.cpu()])
+        self._sync_counters()
+
+    def memory_usage(self) -> dict:
+        out = super().memory_usage()
+        out["ssd_rows"] = sum(kv.size() for kv in self.ssd_kvs.values())
+        out["ssd_files"] = sum(kv.file_count()
+                               for kv in self.ssd_kvs.values())
+        return out
+
+    def compact_ssd(self, sync=True) -> int:
+        return sum(kv.compact(sync=sync) for kv in self.ssd_kvs.values())
+
+
+class _SsdApplyView:
+    """GPU working copy of SSD-resident rows for the fused applies."""
+
+    def __init__(self, parent: HbmDramSsdStorage, keys: torch.Tensor):
+        self._p = parent
+        self._keys = keys
+        self.dim = parent.dim
+        self.ext = parent.ext
+        self.values = parent.ssd_kvs["values"].read(keys).to(parent.device)
+        self._gathered = {}
+
+    def get_slab(self, name, width, init_value, dtype=torch.float32):
+        if name not in self._gathered:
+            kv = self._p._ssd_kv(name, width, init_value)
+            self._gathered[name] = kv.read(
+                self._keys, default=init_value).to(self._p.device)
+        return self._gathered[name]
+
+    def flush(self, keys):
+        self._p.ssd_kvs["values"].write(keys, self.values.cpu())
+        for name, rows in self._gathered.items():
+            self._p.ssd_kvs[name].write(keys, rows.cpu())

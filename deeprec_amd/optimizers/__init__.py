@@ -53,8 +53,10 @@ class Optimizer:
         for ev in self.evs:
             ev._pending_grads.clear()
 
-    def _apply_sparse(self, ev, slots, grad, hyper):
-        if hasattr(ev.storage, "apply_split"):  # multi-tier HBM_DRAM
+    def _apply_sparse(self, ev, slots, grad, hyper, keys=None):
+        if hasattr(ev.storage, "apply_split"):  # multi-tier HBM_DRAM[_SSD]
+            if keys is not None and hasattr(ev.storage, "set_apply_keys"):
+                ev.storage.set_apply_keys(keys)  # SSD tier is key-addressed
             ev.storage.apply_split(self.sparse_name, slots, grad, hyper)
         elif ev.device.type == "cuda":
             from deeprec_amd.ops import hip_backend
@@ -74,7 +76,7 @@ class Optimizer:
                 continue
             hyper = self._sparse_hyper(ev)
             for slots, keys, grad in ev.consume_grads():
-                self._apply_sparse(ev, slots, grad, hyper)
+                self._apply_sparse(ev, slots, grad, hyper, keys=keys)
         if self.pre_dense_step is not None:
             self.pre_dense_step()
         if self._dense is not None:
