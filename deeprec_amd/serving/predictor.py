@@ -57,11 +57,16 @@ class Predictor:
     optionally hot-updating from new full/incremental checkpoints."""
 
     def __init__(self, model, checkpoint_dir: str,
-                 num_sessions: int = 2, device=None):
+                 num_sessions: int = 2, device=None,
+                 remote_sparse: bool = False):
         self.model = model
         self.dir = checkpoint_dir
+        # remote_sparse = RemoteSessionInstance mode: sparse weights live
+        # in an external feature store (attach_remote_store), so only the
+        # dense module restores from checkpoints here
         self.saver = Saver(module=model,
-                           embedding_variables=model.embedding_variables())
+                           embedding_variables=[] if remote_sparse
+                           else model.embedding_variables())
         self.group = SessionGroup(num_sessions, device)
         self._applied = set()
         self._loaded_full: Optional[str] = None
