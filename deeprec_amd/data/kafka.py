@@ -5,12 +5,14 @@ kafka_dataset_op.cc, python kafka_dataset_ops): iterate messages from
 topic partitions starting at stored offsets, resume exactly from a
 checkpointed offset state.
 
-Transport: this image has no Kafka broker or librdkafka, so the transport
-is pluggable and the built-in one is file-backed: `servers="file:///dir"`
-maps topic partition `t:p` to newline-delimited `dir/t-p.log`. The
-offset/resume/checkpoint semantics — the part the training loop depends
-on — are identical to the broker-backed version; a librdkafka consumer
-drops into `_FileConsumer`'s seat unchanged.
+Transports:
+- `servers="host:port"` — the REAL Kafka wire protocol (Metadata/Fetch
+  v0 framing with CRC'd message sets, data/kafka_wire.py) against any
+  broker serving the baseline APIs; offline tests run it against the
+  in-process MiniKafkaBroker speaking the same wire format.
+- `servers="file:///dir"` — file-backed: topic partition `t:p` maps to
+  newline-delimited `dir/t-p.log` (zero-dependency local tier).
+The offset/resume/checkpoint semantics are identical across transports.
 """
 from __future__ import annotations
 
@@ -50,11 +52,13 @@ class KafkaDataset:
                  group: str = "", eof: bool = True,
                  message_parser: Optional[Callable[[str], object]] = None,
                  batch_size: int = 1):
-        if not servers.startswith("file://"):
-            raise NotImplementedError(
-                "only the file-backed transport is available in this "
-                "environment (servers='file:///path')")
-        self.consumer = _FileConsumer(servers[len("file://"):])
+        if servers.startswith("file://"):
+            self.consumer = _FileConsumer(servers[len("file://"):])
+        else:
+            # real Kafka wire protocol (Metadata/Fetch v0 over TCP) —
+            # works against any broker serving the baseline APIs
+            from deeprec_amd.data.kafka_wire import KafkaWireConsumer
+            self.consumer = KafkaWireConsumer(servers)
         self.offsets: Dict[str, int] = {}
         self.parts = []
         for t in topics:
