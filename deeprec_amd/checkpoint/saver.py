@@ -55,13 +55,22 @@ class Saver:
     def __init__(self, module: Optional[torch.nn.Module] = None,
                  embedding_variables: Optional[List] = None,
                  optimizer=None, keep_checkpoint_max: int = 5,
-                 rank: int = 0, world_size: int = 1):
+                 rank: int = 0, world_size: int = 1,
+                 save_filtered: Optional[bool] = None):
         self.module = module
         self.evs = list(embedding_variables or [])
         self.optimizer = optimizer
         self.keep_checkpoint_max = keep_checkpoint_max
         self.rank = rank
         self.world_size = world_size
+        # save sub-threshold admission counters so a restore continues
+        # the filter exactly (reference: TF_EV_SAVE_FILTERED_FEATURES,
+        # embedding_var.h:533)
+        if save_filtered is None:
+            save_filtered = os.environ.get(
+                "DEEPREC_EV_SAVE_FILTERED_FEATURES",
+                "0") not in ("0", "", "false")
+        self.save_filtered = bool(save_filtered)
 
     # ------------- save -------------
     def save(self, directory: str, global_step: int) -> str:
@@ -97,7 +106,13 @@ class Saver:
                        os.path.join(path, "optimizer.pt"))
 
     def _ev_payload(self, base) -> Dict[str, torch.Tensor]:
-        keys, values, freqs, versions = base.export(include_filtered=False)
+        if self.save_filtered:
+            keys, values, freqs, versions, fkeys, ffreqs = \
+                base.export(include_filtered=True)
+        else:
+            keys, values, freqs, versions = \
+                base.export(include_filtered=False)
+            fkeys = None
         present = [n for n in _SLAB_NAMES if n in base.storage.slabs]
         slab_rows = base.storage.export_slabs(present) if present else []
         payload = {
@@ -107,8 +122,22 @@ class Saver:
             "versions": versions.cpu(),
             "buckets": (keys.cpu() % NUM_BUCKETS).to(torch.int32),
         }
+        if fkeys is not None:
+            payload["filtered_keys"] = fkeys.cpu()
+            payload["filtered_freqs"] = ffreqs.cpu().to(torch.int64)
+            cbf = getattr(base.storage, "_cbf", None)
+            if cbf is not None:
+                # counting-bloom pre-admission state is the filter itself
+                import numpy as _np
+                payload["cbf_counters"] = torch.from_numpy(
+                    cbf.counters.astype(_np.int64))
         for n, rows in zip(present, slab_rows):
             payload[f"slab/{n}"] = rows.cpu()
+            # slab fill value for keys admitted AFTER restore (e.g.
+            # adagrad's initial_accumulator) — without it, post-restore
+            # admissions start from 0.0 and trajectories diverge
+            payload[f"slabinit/{n}"] = torch.tensor(
+                float(base.storage._slab_init.get(n, 0.0)))
         return payload
 
     def _save_ev(self, path: str, ev):
@@ -182,11 +211,34 @@ class Saver:
             mask = self._owned_mask(ev, keys)
             slab_rows = {k[len("slab/"):]: v[mask] for k, v in data.items()
                          if k.startswith("slab/")}
+            self._precreate_slabs(base, data, slab_rows)
             base.storage.import_(
                 keys[mask].to(base.device),
                 data["values"][mask].to(base.device),
                 data["freqs"][mask], data["versions"][mask],
                 slab_rows=slab_rows or None)
+            self._restore_filtered(base, ev, data)
+
+    def _precreate_slabs(self, base, data, slab_rows):
+        for n, rows in slab_rows.items():
+            init = data.get(f"slabinit/{n}")
+            base.storage.get_slab(n, rows.shape[1],
+                                  float(init) if init is not None else 0.0)
+
+    def _restore_filtered(self, base, ev, data):
+        cbf_state = data.get("cbf_counters")
+        cbf = getattr(base.storage, "_cbf", None)
+        if cbf_state is not None and cbf is not None:
+            cbf.counters[:] = cbf_state.numpy().astype(
+                cbf.counters.dtype)
+        fk = data.get("filtered_keys")
+        if fk is None or fk.numel() == 0:
+            return
+        if not hasattr(base.storage, "import_filtered"):
+            return
+        fmask = self._owned_mask(ev, fk)
+        base.storage.import_filtered(fk[fmask].to(base.device),
+                                     data["filtered_freqs"][fmask])
 
     def _restore_ev_files(self, path: str, replay: bool = False):
         for ev in self.evs:
@@ -201,6 +253,7 @@ class Saver:
                 slab_rows = {k[len("slab/"):]: v[mask]
                              for k, v in data.items()
                              if k.startswith("slab/")}
+                self._precreate_slabs(base, data, slab_rows)
                 base.storage.import_(
                     keys[mask].to(base.device),
                     data["values"][mask].to(base.device),

@@ -332,6 +332,27 @@ class CpuStorage:
                 for name, rows in slab_rows.items():
                     self.get_slab(name, rows.shape[1], 0.0)[s] = rows[i]
 
+    def import_filtered(self, keys, freqs):
+        """Restore sub-threshold admission counters (reference capability:
+        TF_EV_SAVE_FILTERED_FEATURES, embedding_var.h:533): entries exist
+        with their frequency but no value slot until they cross the
+        filter threshold."""
+        m = keys.numel()
+        if m == 0:
+            return
+        self._grow_entries(m)
+        for i in range(m):
+            k = int(keys[i])
+            e = self._key2entry.get(k)
+            if e is None:
+                e = self.n_entries
+                self.n_entries += 1
+                self._key2entry[k] = e
+                self.entry_key[e] = k
+                self.entry_alive[e] = True
+                self.entry_slot[e] = -1
+            self.entry_freq[e] = int(freqs[i])
+
     def memory_usage(self) -> dict:
         values = self.values.numel() * self.values.element_size()
         slabs = sum(t.numel() * t.element_size()

@@ -486,6 +486,23 @@ class HbmStorage:
         self._check_error()
         self._sync_counters()
 
+    def import_filtered(self, keys, freqs):
+        """Restore sub-threshold admission counters (reference capability:
+        TF_EV_SAVE_FILTERED_FEATURES): hash entries with slot -1 carry
+        the frequency until admission."""
+        m = keys.numel()
+        if m == 0:
+            return
+        keys = keys.to(self.device)
+        self._ensure_capacity(m)
+        slots = torch.full((m,), -1, dtype=torch.int32, device=self.device)
+        self.ext.ht_insert_bulk(
+            keys, slots, freqs.to(self.device, torch.int32),
+            torch.Tensor(), self.ht_keys, self.ht_slot, self.ht_freq,
+            self.ht_version, self.entry_counter, self.error_flag)
+        self._check_error()
+        self._sync_counters()
+
     def memory_usage(self) -> dict:
         """Byte accounting (reference capability:
         embedding_variable_memory_test.cc): hash-table arrays, value slab,
