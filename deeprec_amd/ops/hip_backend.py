@@ -171,6 +171,13 @@ class HbmStorage:
             keep = keys != self.ext.PAD_KEY
             keys, slots = keys[keep], slots[keep]
             freqs, versions = freqs[keep], versions[keep]
+        # the scan compacts via an atomic cursor, so raw order differs
+        # between CALLS; key-sort it so export()/export_slabs() (separate
+        # calls) stay row-aligned in checkpoints
+        if n:
+            order = torch.argsort(keys)
+            keys, slots = keys[order], slots[order]
+            freqs, versions = freqs[order], versions[order]
         return keys, slots, freqs, versions
 
     def _check_error(self):
