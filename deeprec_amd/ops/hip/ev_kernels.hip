@@ -198,7 +198,7 @@ __global__ void k_dedup_pass_a(
     int64_t cap_mask, int epoch, int64_t step,
     int32_t* __restrict__ entry_counter, int32_t* __restrict__ m_counter,
     int64_t* __restrict__ uniq_keys, int64_t* __restrict__ compact_entry,
-    int32_t* __restrict__ error_flag) {
+    int64_t* __restrict__ occ_entry, int32_t* __restrict__ error_flag) {
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
   for (; j < nnz; j += stride) {
@@ -229,6 +229,7 @@ __global__ void k_dedup_pass_a(
           compact_entry[c] = idx;
         }
       }
+      occ_entry[j] = idx;  // pass C reads compact ids WITHOUT re-probing
       goto next_j;
     }
     atomicExch(error_flag, 2);
@@ -255,7 +256,7 @@ __global__ void k_dedup_pass_a_dev(
     const int64_t* __restrict__ step_dev,
     int32_t* __restrict__ entry_counter, int32_t* __restrict__ m_counter,
     int64_t* __restrict__ uniq_keys, int64_t* __restrict__ compact_entry,
-    int32_t* __restrict__ error_flag) {
+    int64_t* __restrict__ occ_entry, int32_t* __restrict__ error_flag) {
   const int epoch = *epoch_dev;
   const int64_t step = *step_dev;
   int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
@@ -284,6 +285,7 @@ __global__ void k_dedup_pass_a_dev(
           compact_entry[c] = idx;
         }
       }
+      occ_entry[j] = idx;
       goto next_j2;
     }
     atomicExch(error_flag, 2);
@@ -630,6 +632,44 @@ __global__ void k_dedup_pass_c(
         break;
       }
     }
+  }
+}
+
+// Index-direct pass C: pass A recorded each occurrence's hash entry, so
+// inverse/rank need NO re-probe of the table (the probe was ~half of
+// the dedup cost at 213k occurrences over a 16M-entry table).
+__global__ void k_dedup_pass_c_idx(
+    const int64_t* __restrict__ occ_entry, int nnz,
+    const int32_t* __restrict__ ht_compact,
+    int32_t* __restrict__ inverse, int32_t* __restrict__ counts,
+    int32_t* __restrict__ rank) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; j < nnz; j += stride) {
+    int c = ht_compact[occ_entry[j]];
+    inverse[j] = c;
+    rank[j] = atomicAdd(&counts[c], 1);
+  }
+}
+
+// PAD-aware variant for the owner side of the sharded exchange.
+__global__ void k_dedup_pass_c_idx_pad(
+    const int64_t* __restrict__ keys,
+    const int64_t* __restrict__ occ_entry, int nnz,
+    const int32_t* __restrict__ ht_compact,
+    int32_t* __restrict__ inverse, int32_t* __restrict__ counts,
+    int32_t* __restrict__ rank) {
+  int64_t j = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; j < nnz; j += stride) {
+    if (keys[j] == PAD_KEY) {
+      inverse[j] = -1;
+      rank[j] = 0;
+      continue;
+    }
+    int c = ht_compact[occ_entry[j]];
+    inverse[j] = c;
+    rank[j] = atomicAdd(&counts[c], 1);
   }
 }
 
@@ -1220,7 +1260,7 @@ torch::Tensor ht_dedup_a(torch::Tensor keys, torch::Tensor ht_keys,
                          int64_t epoch, int64_t step,
                          torch::Tensor entry_counter, torch::Tensor m_counter,
                          torch::Tensor uniq_keys, torch::Tensor compact_entry,
-                         torch::Tensor error_flag) {
+                         torch::Tensor occ_entry, torch::Tensor error_flag) {
   int64_t nnz = keys.numel();
   if (nnz == 0) return m_counter;
   auto stream = current_stream();
@@ -1231,7 +1271,7 @@ torch::Tensor ht_dedup_a(torch::Tensor keys, torch::Tensor ht_keys,
       ht_keys.numel() - 1, (int)epoch, step,
       entry_counter.data_ptr<int32_t>(), m_counter.data_ptr<int32_t>(),
       uniq_keys.data_ptr<int64_t>(), compact_entry.data_ptr<int64_t>(),
-      error_flag.data_ptr<int32_t>());
+      occ_entry.data_ptr<int64_t>(), error_flag.data_ptr<int32_t>());
   return m_counter;
 }
 
@@ -1408,7 +1448,8 @@ torch::Tensor ht_dedup_a_dev(
     torch::Tensor ht_compact, torch::Tensor epoch_dev,
     torch::Tensor step_dev, torch::Tensor entry_counter,
     torch::Tensor m_counter, torch::Tensor uniq_keys,
-    torch::Tensor compact_entry, torch::Tensor error_flag) {
+    torch::Tensor compact_entry, torch::Tensor occ_entry,
+    torch::Tensor error_flag) {
   int64_t nnz = keys.numel();
   if (nnz == 0) return m_counter;
   auto stream = current_stream();
@@ -1419,7 +1460,8 @@ torch::Tensor ht_dedup_a_dev(
       ht_keys.numel() - 1, epoch_dev.data_ptr<int32_t>(),
       step_dev.data_ptr<int64_t>(), entry_counter.data_ptr<int32_t>(),
       m_counter.data_ptr<int32_t>(), uniq_keys.data_ptr<int64_t>(),
-      compact_entry.data_ptr<int64_t>(), error_flag.data_ptr<int32_t>());
+      compact_entry.data_ptr<int64_t>(), occ_entry.data_ptr<int64_t>(),
+      error_flag.data_ptr<int32_t>());
   return m_counter;
 }
 
@@ -1464,6 +1506,41 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ht_dedup_c(
       ht_compact.data_ptr<int32_t>(), ht_keys.numel() - 1,
       inverse.data_ptr<int32_t>(), counts.data_ptr<int32_t>(),
       rank.data_ptr<int32_t>());
+  return {inverse, counts, rank};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ht_dedup_c_idx(
+    torch::Tensor occ_entry, torch::Tensor ht_compact, int64_t m) {
+  int64_t nnz = occ_entry.numel();
+  auto inverse = torch::empty({nnz}, ht_compact.options());
+  auto rank = torch::empty({nnz}, ht_compact.options());
+  auto counts = torch::empty({m}, ht_compact.options());
+  if (nnz == 0) return {inverse, counts, rank};
+  auto stream = current_stream();
+  k_zero_f32<<<n_blocks((m + 3) / 4), kBlock, 0, stream>>>(
+      reinterpret_cast<float*>(counts.data_ptr<int32_t>()), m);
+  k_dedup_pass_c_idx<<<n_blocks(nnz), kBlock, 0, stream>>>(
+      occ_entry.data_ptr<int64_t>(), (int)nnz,
+      ht_compact.data_ptr<int32_t>(), inverse.data_ptr<int32_t>(),
+      counts.data_ptr<int32_t>(), rank.data_ptr<int32_t>());
+  return {inverse, counts, rank};
+}
+
+std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> ht_dedup_c_idx_pad(
+    torch::Tensor keys, torch::Tensor occ_entry, torch::Tensor ht_compact,
+    int64_t m) {
+  int64_t nnz = occ_entry.numel();
+  auto inverse = torch::empty({nnz}, ht_compact.options());
+  auto rank = torch::empty({nnz}, ht_compact.options());
+  auto counts = torch::empty({m}, ht_compact.options());
+  if (nnz == 0) return {inverse, counts, rank};
+  auto stream = current_stream();
+  k_zero_f32<<<n_blocks((m + 3) / 4), kBlock, 0, stream>>>(
+      reinterpret_cast<float*>(counts.data_ptr<int32_t>()), m);
+  k_dedup_pass_c_idx_pad<<<n_blocks(nnz), kBlock, 0, stream>>>(
+      keys.data_ptr<int64_t>(), occ_entry.data_ptr<int64_t>(), (int)nnz,
+      ht_compact.data_ptr<int32_t>(), inverse.data_ptr<int32_t>(),
+      counts.data_ptr<int32_t>(), rank.data_ptr<int32_t>());
   return {inverse, counts, rank};
 }
 
@@ -1938,6 +2015,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   mod.def("gather_host_rows", &gather_host_rows);
   mod.def("scatter_host_rows", &scatter_host_rows);
   mod.def("ht_dedup_c", &ht_dedup_c);
+  mod.def("ht_dedup_c_idx", &ht_dedup_c_idx);
+  mod.def("ht_dedup_c_idx_pad", &ht_dedup_c_idx_pad);
   mod.def("csr_order", &csr_order);
   mod.def("csr_scatter", &csr_scatter);
   mod.def("ht_insert_bulk", &ht_insert_bulk);

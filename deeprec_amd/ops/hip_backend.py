@@ -237,6 +237,7 @@ class HbmStorage:
         self.ext.bump_epoch(self._epoch_dev, self._step_dev)
         uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
         centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        occ_entry = torch.empty(nnz, dtype=torch.int64, device=self.device)
         m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._last_m_dev = m_counter
         self.ext.ht_dedup_a_dev(values_cat, self.ht_keys, self.ht_freq,
@@ -244,12 +245,13 @@ class HbmStorage:
                                 self.ht_compact, self._epoch_dev,
                                 self._step_dev, self.entry_counter,
                                 m_counter, uniq_buf, centry_buf,
-                                self.error_flag)
+                                occ_entry, self.error_flag)
         # pass C first: its exact per-batch counts feed pass B's single
         # per-unique freq update (pass A is claim-only — hot keys no
-        # longer serialize per-occurrence atomics)
-        inverse, counts, rank = self.ext.ht_dedup_c(
-            values_cat, self.ht_keys, self.ht_compact, nnz)
+        # longer serialize per-occurrence atomics; index-direct: no
+        # table re-probe)
+        inverse, counts, rank = self.ext.ht_dedup_c_idx(
+            occ_entry, self.ht_compact, nnz)
         self._last_rank = rank  # consumed by collection._prep_backward
         slots = self.ext.ht_dedup_b_padded(
             centry_buf, uniq_buf, m_counter, self.ht_slot, self.ht_freq,
@@ -273,6 +275,7 @@ class HbmStorage:
         self.ext.bump_epoch(self._epoch_dev, self._step_dev)
         uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
         centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        occ_entry = torch.empty(nnz, dtype=torch.int64, device=self.device)
         m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
         self._last_m_dev = m_counter
         self.ext.ht_dedup_a_dev(values_cat, self.ht_keys, self.ht_freq,
@@ -280,9 +283,9 @@ class HbmStorage:
                                 self.ht_compact, self._epoch_dev,
                                 self._step_dev, self.entry_counter,
                                 m_counter, uniq_buf, centry_buf,
-                                self.error_flag)
-        inverse, counts, rank = self.ext.ht_dedup_c(
-            values_cat, self.ht_keys, self.ht_compact, nnz)
+                                occ_entry, self.error_flag)
+        inverse, counts, rank = self.ext.ht_dedup_c_idx(
+            occ_entry, self.ht_compact, nnz)
         self._last_rank = rank
         return uniq_buf, inverse, counts, m_counter
 
@@ -299,18 +302,19 @@ class HbmStorage:
         self.ext.bump_epoch_only(self._epoch_dev)
         uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
         centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        occ_entry = torch.empty(nnz, dtype=torch.int64, device=self.device)
         m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
         self.ext.ht_dedup_a_dev(recv_keys, self.ht_keys, self.ht_freq,
                                 self.ht_version, self.ht_epoch,
                                 self.ht_compact, self._epoch_dev,
                                 self._step_dev, self.entry_counter,
                                 m_counter, uniq_buf, centry_buf,
-                                self.error_flag)
-        # PAD-aware pass C: pad elements get inverse -1 and never touch
-        # the shared counters (a naive path serializes ~10^5 atomics on
-        # PAD_KEY's single cache line)
-        inverse, _counts_c, _rank = self.ext.ht_dedup_c_pad(
-            recv_keys, self.ht_keys, self.ht_compact, nnz)
+                                occ_entry, self.error_flag)
+        # PAD-aware, index-direct pass C: pad elements get inverse -1
+        # and never touch the shared counters (a naive path serialized
+        # ~10^5 atomics on PAD_KEY's single cache line)
+        inverse, _counts_c, _rank = self.ext.ht_dedup_c_idx_pad(
+            recv_keys, occ_entry, self.ht_compact, nnz)
         cnt_sum = self.ext.cnt_sum_pad(inverse, recv_cnt, nnz)
         slots = self.ext.ht_dedup_b_padded(
             centry_buf, uniq_buf, m_counter, self.ht_slot, self.ht_freq,
@@ -337,11 +341,13 @@ class HbmStorage:
         self._epoch += 1
         uniq_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
         centry_buf = torch.empty(nnz, dtype=torch.int64, device=self.device)
+        occ_entry = torch.empty(nnz, dtype=torch.int64, device=self.device)
         m_counter = torch.zeros(1, dtype=torch.int32, device=self.device)
         self.ext.ht_dedup_a(values_cat, self.ht_keys, self.ht_freq,
                             self.ht_version, self.ht_epoch, self.ht_compact,
                             self._epoch, step, self.entry_counter, m_counter,
-                            uniq_buf, centry_buf, self.error_flag)
+                            uniq_buf, centry_buf, occ_entry,
+                            self.error_flag)
         # one D2H sync (as torch.unique pays); piggyback the real counters
         # so capacity hints don't inflate by nnz per step (which caused
         # needless rehashes on high-uniqueness workloads)
@@ -353,8 +359,9 @@ class HbmStorage:
         self.observe_uniq_ratio(m, nnz)
         uniq = uniq_buf[:m]
         # pass C first: exact counts feed pass B's per-unique freq update
-        inverse, counts, rank = self.ext.ht_dedup_c(
-            values_cat, self.ht_keys, self.ht_compact, m)
+        # (index-direct: pass A recorded each occurrence's entry)
+        inverse, counts, rank = self.ext.ht_dedup_c_idx(
+            occ_entry, self.ht_compact, m)
         self._last_rank = rank  # consumed by collection._prep_backward
         slots = self.ext.ht_dedup_b(
             centry_buf[:m], uniq, self.ht_slot, self.ht_freq,
