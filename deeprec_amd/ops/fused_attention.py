@@ -68,6 +68,32 @@ def fused_mha(q, k, v, key_pad: torch.Tensor, n_heads: int,
     return _torch_mha(q, k, v, key_pad, n_heads, scale)
 
 
+class _DinFeat(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, seq, tgt):
+        from deeprec_amd.ops.build_ext import require_extension
+        ext = require_extension()
+        ctx.ext = ext
+        ctx.save_for_backward(seq, tgt)
+        return ext.din_feat_fwd(seq.contiguous(), tgt.contiguous())
+
+    @staticmethod
+    def backward(ctx, g):
+        seq, tgt = ctx.saved_tensors
+        dseq, dtgt = ctx.ext.din_feat_bwd(g, seq, tgt)
+        return dseq, dtgt
+
+
+def din_att_features(seq: torch.Tensor, tgt: torch.Tensor):
+    """[B,T,D] seq + [B,D] target -> bf16 [B*T, 4D] attention features
+    [s, t, s-t, s*t] in one fused pass on GPU (torch path elsewhere)."""
+    if seq.device.type == "cuda":
+        return _DinFeat.apply(seq.float(), tgt.float())
+    t = tgt.unsqueeze(1).expand_as(seq)
+    return torch.cat([seq, t, seq - t, seq * t],
+                     dim=2).reshape(seq.shape[0] * seq.shape[1], -1)
+
+
 class _ResLN(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x16, a16, gamma, beta, eps):

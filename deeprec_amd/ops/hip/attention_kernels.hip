@@ -231,6 +231,56 @@ __global__ void k_mha_bwd(const short* __restrict__ dout,
   }
 }
 
+// ---------------------------------------------------------------------
+// DIN attention features: att_in[b,t] = [s, tgt, s - tgt, s * tgt]
+// built in ONE pass (the torch path materializes the expand, two
+// elementwise intermediates and a 4-way cat — 5 kernels and ~2x the
+// HBM traffic at [B, T, 4D]). ≙ modelzoo/din/train.py attention input.
+__global__ void k_din_feat_fwd(const float* __restrict__ seq,
+                               const float* __restrict__ tgt, int B, int T,
+                               int D, short* __restrict__ out) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)B * T * D;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < total; i += stride) {
+    int d = (int)(i % D);
+    int64_t bt = i / D;
+    int b = (int)(bt / T);
+    float s = seq[i];
+    float t = tgt[(int64_t)b * D + d];
+    short* o = out + bt * 4 * D + d;
+    o[0] = fbu(s);
+    o[D] = fbu(t);
+    o[2 * D] = fbu(s - t);
+    o[3 * D] = fbu(s * t);
+  }
+}
+
+// backward: ds = g0 + g2 + g3*t ; dt[b] = sum_t (g1 - g2 + g3*s)
+__global__ void k_din_feat_bwd(const short* __restrict__ g,
+                               const float* __restrict__ seq,
+                               const float* __restrict__ tgt, int B, int T,
+                               int D, float* __restrict__ dseq,
+                               float* __restrict__ dtgt) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)B * T * D;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < total; i += stride) {
+    int d = (int)(i % D);
+    int64_t bt = i / D;
+    int b = (int)(bt / T);
+    const short* gp = g + bt * 4 * D + d;
+    float g0 = bfu(gp[0]);
+    float g1 = bfu(gp[D]);
+    float g2 = bfu(gp[2 * D]);
+    float g3 = bfu(gp[3 * D]);
+    float t = tgt[(int64_t)b * D + d];
+    float s = seq[i];
+    dseq[i] = g0 + g2 + g3 * t;
+    atomicAdd(&dtgt[(int64_t)b * D + d], g1 - g2 + g3 * s);
+  }
+}
+
 }  // namespace
 
 static const short* att_bf_ptr(const torch::Tensor& t) {
@@ -302,7 +352,37 @@ std::tuple<torch::Tensor, torch::Tensor, torch::Tensor> mha_bwd(
   return {dq, dk, dv};
 }
 
+torch::Tensor din_feat_fwd(torch::Tensor seq, torch::Tensor tgt) {
+  TORCH_CHECK(seq.is_contiguous() && seq.scalar_type() == torch::kFloat32);
+  int B = seq.size(0), T = seq.size(1), D = seq.size(2);
+  auto out = torch::empty({(int64_t)B * T, 4 * (int64_t)D},
+                          seq.options().dtype(torch::kBFloat16));
+  int64_t total = (int64_t)B * T * D;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+  k_din_feat_fwd<<<blocks, 256, 0, att_stream()>>>(
+      seq.data_ptr<float>(), tgt.data_ptr<float>(), B, T, D,
+      att_bf_ptr_mut(out));
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> din_feat_bwd(torch::Tensor g,
+                                                      torch::Tensor seq,
+                                                      torch::Tensor tgt) {
+  int B = seq.size(0), T = seq.size(1), D = seq.size(2);
+  auto dseq = torch::empty_like(seq);
+  auto dtgt = torch::zeros_like(tgt);
+  int64_t total = (int64_t)B * T * D;
+  int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
+  k_din_feat_bwd<<<blocks, 256, 0, att_stream()>>>(
+      att_bf_ptr(g.contiguous()), seq.data_ptr<float>(),
+      tgt.data_ptr<float>(), B, T, D, dseq.data_ptr<float>(),
+      dtgt.data_ptr<float>());
+  return {dseq, dtgt};
+}
+
 void register_attention(pybind11::module_& mod) {
   mod.def("mha_fwd", &mha_fwd);
   mod.def("mha_bwd", &mha_bwd);
+  mod.def("din_feat_fwd", &din_feat_fwd);
+  mod.def("din_feat_bwd", &din_feat_bwd);
 }

@@ -120,3 +120,24 @@ def test_fused_residual_ln_matches_torch():
                                    rtol=5e-2, atol=5e-1)
         ln.weight.grad = None
         ln.bias.grad = None
+
+
+def test_din_att_features_matches_torch():
+    from deeprec_amd.ops.fused_attention import din_att_features
+
+    torch.manual_seed(4)
+    b, t, d = 33, 17, 32
+    seq = torch.randn(b, t, d, device=DEV, requires_grad=True)
+    tgt = torch.randn(b, d, device=DEV, requires_grad=True)
+    out = din_att_features(seq, tgt)
+    sr = seq.detach().clone().requires_grad_(True)
+    tr = tgt.detach().clone().requires_grad_(True)
+    te = tr.unsqueeze(1).expand(b, t, d)
+    ref = torch.cat([sr, te, sr - te, sr * te], dim=2).reshape(b * t,
+                                                               4 * d)
+    torch.testing.assert_close(out.float(), ref, rtol=1e-2, atol=1e-2)
+    g = torch.randn(b * t, 4 * d, device=DEV)
+    out.backward(g.to(torch.bfloat16))
+    ref.backward(g)
+    torch.testing.assert_close(seq.grad, sr.grad, rtol=2e-2, atol=2e-2)
+    torch.testing.assert_close(tgt.grad, tr.grad, rtol=2e-2, atol=2e-1)
