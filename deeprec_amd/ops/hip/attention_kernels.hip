@@ -256,12 +256,14 @@ __global__ void k_din_feat_fwd(const float* __restrict__ seq,
   }
 }
 
-// backward: ds = g0 + g2 + g3*t ; dt[b] = sum_t (g1 - g2 + g3*s)
-__global__ void k_din_feat_bwd(const short* __restrict__ g,
-                               const float* __restrict__ seq,
-                               const float* __restrict__ tgt, int B, int T,
-                               int D, float* __restrict__ dseq,
-                               float* __restrict__ dtgt) {
+// backward: ds = g0 + g2 + g3*t (elementwise); dt[b] = sum_t
+// (g1 - g2 + g3*s) via one thread per (b, d) looping T — NO atomics
+// (the per-element atomicAdd variant measured 132 us/step at B*T*D
+// 13M atomics).
+__global__ void k_din_feat_bwd_ds(const short* __restrict__ g,
+                                  const float* __restrict__ tgt, int B,
+                                  int T, int D,
+                                  float* __restrict__ dseq) {
   int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
   int64_t total = (int64_t)B * T * D;
   int64_t stride = gridDim.x * (int64_t)blockDim.x;
@@ -270,14 +272,29 @@ __global__ void k_din_feat_bwd(const short* __restrict__ g,
     int64_t bt = i / D;
     int b = (int)(bt / T);
     const short* gp = g + bt * 4 * D + d;
-    float g0 = bfu(gp[0]);
-    float g1 = bfu(gp[D]);
-    float g2 = bfu(gp[2 * D]);
-    float g3 = bfu(gp[3 * D]);
-    float t = tgt[(int64_t)b * D + d];
-    float s = seq[i];
-    dseq[i] = g0 + g2 + g3 * t;
-    atomicAdd(&dtgt[(int64_t)b * D + d], g1 - g2 + g3 * s);
+    dseq[i] = bfu(gp[0]) + bfu(gp[2 * D])
+              + bfu(gp[3 * D]) * tgt[(int64_t)b * D + d];
+  }
+}
+
+__global__ void k_din_feat_bwd_dt(const short* __restrict__ g,
+                                  const float* __restrict__ seq, int B,
+                                  int T, int D,
+                                  float* __restrict__ dtgt) {
+  int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+  int64_t total = (int64_t)B * D;
+  int64_t stride = gridDim.x * (int64_t)blockDim.x;
+  for (; i < total; i += stride) {
+    int d = (int)(i % D);
+    int b = (int)(i / D);
+    float acc = 0.0f;
+    for (int t = 0; t < T; ++t) {
+      int64_t bt = (int64_t)b * T + t;
+      const short* gp = g + bt * 4 * D + d;
+      acc += bfu(gp[D]) - bfu(gp[2 * D])
+             + bfu(gp[3 * D]) * seq[bt * D + d];
+    }
+    dtgt[i] = acc;
   }
 }
 
@@ -370,12 +387,16 @@ std::tuple<torch::Tensor, torch::Tensor> din_feat_bwd(torch::Tensor g,
                                                       torch::Tensor tgt) {
   int B = seq.size(0), T = seq.size(1), D = seq.size(2);
   auto dseq = torch::empty_like(seq);
-  auto dtgt = torch::zeros_like(tgt);
+  auto dtgt = torch::empty_like(tgt);
+  auto gc = g.contiguous();
   int64_t total = (int64_t)B * T * D;
   int blocks = (int)std::min<int64_t>((total + 255) / 256, 8192);
-  k_din_feat_bwd<<<blocks, 256, 0, att_stream()>>>(
-      att_bf_ptr(g.contiguous()), seq.data_ptr<float>(),
-      tgt.data_ptr<float>(), B, T, D, dseq.data_ptr<float>(),
+  k_din_feat_bwd_ds<<<blocks, 256, 0, att_stream()>>>(
+      att_bf_ptr(gc), tgt.data_ptr<float>(), B, T, D,
+      dseq.data_ptr<float>());
+  int blocks2 = (int)std::min<int64_t>(((int64_t)B * D + 255) / 256, 8192);
+  k_din_feat_bwd_dt<<<blocks2, 256, 0, att_stream()>>>(
+      att_bf_ptr(gc), seq.data_ptr<float>(), B, T, D,
       dtgt.data_ptr<float>());
   return {dseq, dtgt};
 }
