@@ -146,8 +146,15 @@ def main():
                 eager_step(i)
             torch.cuda.synchronize()
             # pad cap: worst observed split + 50% headroom (overflow trips
-            # engine error 4 and fails loudly after the run)
+            # engine error 4 and fails loudly after the run). The cap is
+            # part of the WIRE SHAPE, so every rank must agree: all-reduce
+            # the max over ranks
             cap = int(1.5 * max(coll.observed_max_split(), 1)) + 256
+            if world_size > 1:
+                import torch.distributed as dist
+                t = torch.tensor([cap], dtype=torch.int64, device=device)
+                dist.all_reduce(t, op=dist.ReduceOp.MAX)
+                cap = int(t.item())
             coll.enable_graph_mode(expected_entries=1 << 23,
                                    expected_slots=1 << 23, pad_cap=cap)
             st = coll.storage

@@ -167,8 +167,10 @@ class ShardedEmbeddingCollection:
         wire protocol is identical; only the local dedup engine differs.
         pad_cap must exceed the per-peer unique-key count of every step
         (observed_max_split() after eager warmup + slack is the intended
-        sizing); overflow raises via the engine error flag on GPU and
-        ValueError on CPU."""
+        sizing) and must be IDENTICAL on every rank — it defines the
+        wire shape, so agree on it with an all-reduce max before calling
+        (bench.py does). Overflow raises via the engine error flag on
+        GPU and ValueError on CPU."""
         self._pad_cap = int(pad_cap)
 
     def observed_max_split(self) -> int:
