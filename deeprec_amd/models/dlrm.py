@@ -129,12 +129,15 @@ class DLRM(nn.Module):
                 dense, (0, self.dense_in - dense.shape[1]))
         with amp:
             bot = self.mlp_bot(dense)
-            feats = torch.cat([bot.unsqueeze(1), emb_feats.to(bot.dtype)],
-                              dim=1)  # [B, F+1, D]
             if self.interaction_op == "dot":
-                inter = self._interact(feats)
-                top_in = torch.cat([bot, inter], dim=1)
+                # cat-fused interaction: [bot | pair dots] in one kernel
+                # per direction (no feats assembly, no top-input cat)
+                from deeprec_amd.ops.fused_mlp import dot_interaction_cat
+                top_in = dot_interaction_cat(bot, emb_feats,
+                                             self.n_pairs_pad)
             else:
+                feats = torch.cat(
+                    [bot.unsqueeze(1), emb_feats.to(bot.dtype)], dim=1)
                 top_in = feats.flatten(1)
             logits = self.mlp_top(top_in)
         return logits.squeeze(1).float()

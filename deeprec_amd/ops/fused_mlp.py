@@ -128,6 +128,40 @@ def dot_interaction(feats: torch.Tensor, p_pad: int = None) -> torch.Tensor:
     return out.to(feats.dtype)
 
 
+class _DotInteractionCat(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, bot, emb, p_pad):
+        from deeprec_amd.ops.build_ext import require_extension
+        ext = require_extension()
+        b16 = bot.to(torch.bfloat16).contiguous()
+        e16 = emb.to(torch.bfloat16).contiguous()
+        out = ext.interact_cat_fwd(b16, e16, p_pad)
+        ctx.ext = ext
+        ctx.dtypes = (bot.dtype, emb.dtype)
+        ctx.save_for_backward(b16, e16)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        b16, e16 = ctx.saved_tensors
+        dbot, demb = ctx.ext.interact_cat_bwd(
+            grad_out.to(torch.bfloat16), b16, e16)
+        bd, ed = ctx.dtypes
+        return dbot.to(bd), demb.to(ed), None
+
+
+def dot_interaction_cat(bot: torch.Tensor, emb: torch.Tensor,
+                        p_pad: int) -> torch.Tensor:
+    """bot [B,D] + emb [B,F,D] -> [B, D + p_pad] = [bot | pairwise dots
+    over [bot; emb] rows, i<j upper triangle, zero-padded] in ONE fused
+    kernel per direction — removes the feats-assembly cat AND the
+    top-MLP input cat. Torch fallback composes the same math."""
+    if bot.device.type == "cuda" and bot.shape[1] % 8 == 0:
+        return _DotInteractionCat.apply(bot, emb, p_pad)
+    feats = torch.cat([bot.unsqueeze(1), emb.to(bot.dtype)], dim=1)
+    return torch.cat([bot, dot_interaction(feats, p_pad)], dim=1)
+
+
 def fused_mlp(sizes: List[int], in_dim: int,
               final_activation: bool = True) -> nn.Sequential:
     layers = []

@@ -827,6 +827,114 @@ __global__ void k_interact_fwd(const short* __restrict__ feats, int B, int F,
 }
 
 // backward: dfeats[b,i,d] = sum_j!=i g[b, pair(i,j)] * feats[b,j,d]
+// Cat-fused variant: takes bot [B, D] and emb [B, Fe, D] separately and
+// emits top_in = [bot | pair dots] [B, D + P_pad] directly — removes the
+// two torch cats around the interaction (feats assembly + top-MLP input).
+__global__ void k_interact_cat_fwd(const short* __restrict__ bot,
+                                   const short* __restrict__ emb, int B,
+                                   int Fe, int D, int P, int P_pad,
+                                   short* __restrict__ out) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int b = blockIdx.x * 4 + wave;
+  const int F = Fe + 1;
+  extern __shared__ short lds[];
+  short* f = lds + wave * F * D;
+  const bool active = b < B;
+  if (active) {
+    const short* bsrc = bot + (int64_t)b * D;
+    for (int t = lane; t * 8 < D; t += 64)
+      *reinterpret_cast<bf16x8*>(f + t * 8) =
+          *reinterpret_cast<const bf16x8*>(bsrc + t * 8);
+    const short* esrc = emb + (int64_t)b * Fe * D;
+    for (int t = lane; t * 8 < Fe * D; t += 64)
+      *reinterpret_cast<bf16x8*>(f + D + t * 8) =
+          *reinterpret_cast<const bf16x8*>(esrc + t * 8);
+  }
+  __syncthreads();
+  if (!active) return;
+  short* dst = reinterpret_cast<short*>(out) + (int64_t)b * (D + P_pad);
+  for (int d = lane; d < D; d += 64) dst[d] = f[d];  // bot passthrough
+  dst += D;
+  for (int p = lane; p < P_pad; p += 64) {
+    if (p >= P) {
+      dst[p] = 0;
+      continue;
+    }
+    int i = 0, rem = p, row = F - 1;
+    while (rem >= row) {
+      rem -= row;
+      --row;
+      ++i;
+    }
+    int j = i + 1 + rem;
+    float acc = 0.0f;
+    const short* fi = f + i * D;
+    const short* fj = f + j * D;
+    for (int d = 0; d < D; ++d)
+      acc += bf2f_u16(fi[d]) * bf2f_u16(fj[d]);
+    dst[p] = f2bf_u16(acc);
+  }
+}
+
+__global__ void k_interact_cat_bwd(const short* __restrict__ grad,
+                                   const short* __restrict__ bot,
+                                   const short* __restrict__ emb, int B,
+                                   int Fe, int D, int P, int P_pad,
+                                   short* __restrict__ dbot,
+                                   short* __restrict__ demb) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int b = blockIdx.x * 4 + wave;
+  const int F = Fe + 1;
+  extern __shared__ short lds[];
+  short* f = lds + wave * (F * D + ((P + 7) & ~7));
+  short* g = f + F * D;
+  const bool active = b < B;
+  if (active) {
+    const short* bsrc = bot + (int64_t)b * D;
+    for (int t = lane; t * 8 < D; t += 64)
+      *reinterpret_cast<bf16x8*>(f + t * 8) =
+          *reinterpret_cast<const bf16x8*>(bsrc + t * 8);
+    const short* esrc = emb + (int64_t)b * Fe * D;
+    for (int t = lane; t * 8 < Fe * D; t += 64)
+      *reinterpret_cast<bf16x8*>(f + D + t * 8) =
+          *reinterpret_cast<const bf16x8*>(esrc + t * 8);
+    const short* gsrc = grad + (int64_t)b * (D + P_pad) + D;
+    for (int t = lane; t < P; t += 64) g[t] = gsrc[t];
+  }
+  __shared__ short plut[40 * 40];
+  for (int t = threadIdx.x; t < F * F; t += blockDim.x) {
+    int i = t / F, j = t % F;
+    int lo = i < j ? i : j, hi = i < j ? j : i;
+    plut[t] = (short)(i == j ? -1
+                             : lo * F - (lo * (lo + 1)) / 2 + (hi - lo - 1));
+  }
+  __syncthreads();
+  if (!active) return;
+  const short* gdirect = grad + (int64_t)b * (D + P_pad);
+  short* db = dbot + (int64_t)b * D;
+  short* de = demb + (int64_t)b * Fe * D;
+  for (int t = lane; t < F * D; t += 64) {
+    int i = t / D, d = t % D;
+    float acc = 0.0f;
+    const short* prow = &plut[i * F];
+    const short* fd = f + d;
+#pragma unroll 4
+    for (int j = 0; j < F; ++j) {
+      int p = prow[j];
+      if (p < 0) continue;
+      acc += bf2f_u16(g[p]) * bf2f_u16(fd[j * D]);
+    }
+    if (i == 0) {
+      // bot row: the direct top-MLP slice adds in
+      db[d] = f2bf_u16(acc + bf2f_u16(gdirect[d]));
+    } else {
+      de[(i - 1) * D + d] = f2bf_u16(acc);
+    }
+  }
+}
+
 __global__ void k_interact_bwd(const short* __restrict__ grad,
                                const short* __restrict__ feats, int B, int F,
                                int D, int P, int P_pad,
@@ -1094,6 +1202,39 @@ torch::Tensor interact_bwd(torch::Tensor grad, torch::Tensor feats) {
   return dfeats;
 }
 
+torch::Tensor interact_cat_fwd(torch::Tensor bot, torch::Tensor emb,
+                               int64_t p_pad) {
+  TORCH_CHECK(bot.scalar_type() == torch::kBFloat16 && bot.is_contiguous());
+  TORCH_CHECK(emb.scalar_type() == torch::kBFloat16 && emb.is_contiguous());
+  int B = emb.size(0), Fe = emb.size(1), D = emb.size(2);
+  int F = Fe + 1;
+  int P = F * (F - 1) / 2;
+  TORCH_CHECK(D % 8 == 0, "D must be a multiple of 8");
+  auto out = torch::empty({B, D + p_pad}, emb.options());
+  int blocks = (B + 3) / 4;
+  size_t lds = 4 * (size_t)F * D * sizeof(short);
+  k_interact_cat_fwd<<<blocks, 256, lds, dense_stream()>>>(
+      bf_ptr(bot), bf_ptr(emb), B, Fe, D, P, (int)p_pad, bf_ptr_mut(out));
+  return out;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> interact_cat_bwd(
+    torch::Tensor grad, torch::Tensor bot, torch::Tensor emb) {
+  int B = emb.size(0), Fe = emb.size(1), D = emb.size(2);
+  int F = Fe + 1;
+  TORCH_CHECK(F <= 40, "interact: pair LUT supports F <= 40");
+  int P = F * (F - 1) / 2;
+  int P_pad = grad.size(1) - D;
+  auto dbot = torch::empty_like(bot);
+  auto demb = torch::empty_like(emb);
+  int blocks = (B + 3) / 4;
+  size_t lds = 4 * (size_t)(F * D + ((P + 7) & ~7)) * sizeof(short);
+  k_interact_cat_bwd<<<blocks, 256, lds, dense_stream()>>>(
+      bf_ptr(grad.contiguous()), bf_ptr(bot), bf_ptr(emb), B, Fe, D, P,
+      P_pad, bf_ptr_mut(dbot), bf_ptr_mut(demb));
+  return {dbot, demb};
+}
+
 std::tuple<torch::Tensor, torch::Tensor> l2norm_fwd(torch::Tensor x,
                                                     double eps) {
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && x.is_contiguous());
@@ -1120,6 +1261,8 @@ void register_dense(py::module_& mod) {
   mod.def("l2norm_bwd", &l2norm_bwd);
   mod.def("interact_fwd", &interact_fwd);
   mod.def("interact_bwd", &interact_bwd);
+  mod.def("interact_cat_fwd", &interact_cat_fwd);
+  mod.def("interact_cat_bwd", &interact_cat_bwd);
   mod.def("linear_fwd", &linear_fwd, py::arg("x"), py::arg("w"),
           py::arg("bias"), py::arg("act"), py::arg("variant") = -1);
   mod.def("linear_dx", &linear_dx);

@@ -338,3 +338,31 @@ def test_dense_adam_kernel_exact():
     torch.testing.assert_close(w16.float(), wn.to(torch.bfloat16).float())
     # grads must be untouched (the all-reduce scaling folds into gscale)
     torch.testing.assert_close(g, g0)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("b,f,d", [(64, 26, 16), (100, 7, 8)])
+def test_dot_interaction_cat(b, f, d):
+    """Cat-fused interaction: [bot | pair dots] ≡ cat(bot,
+    dot_interaction(cat(bot, emb)))."""
+    from deeprec_amd.ops.fused_mlp import dot_interaction, dot_interaction_cat
+
+    torch.manual_seed(5)
+    p = (f + 1) * f // 2
+    p_pad = (p + 15) & ~15
+    bot = torch.randn(b, d, device=DEV).to(torch.bfloat16).float()
+    emb = torch.randn(b, f, d, device=DEV).to(torch.bfloat16).float()
+    bg = bot.clone().requires_grad_(True)
+    eg = emb.clone().requires_grad_(True)
+    out = dot_interaction_cat(bg, eg, p_pad)
+    assert out.shape == (b, d + p_pad)
+    br = bot.clone().requires_grad_(True)
+    er = emb.clone().requires_grad_(True)
+    feats = torch.cat([br.unsqueeze(1), er], dim=1)
+    ref = torch.cat([br, dot_interaction(feats, p_pad).float()], dim=1)
+    torch.testing.assert_close(out.float(), ref, rtol=2e-2, atol=2e-2)
+    g = torch.randn(b, d + p_pad, device=DEV)
+    out.backward(g.to(torch.bfloat16))
+    ref.backward(g)
+    torch.testing.assert_close(bg.grad, br.grad, rtol=3e-2, atol=2e-1)
+    torch.testing.assert_close(eg.grad, er.grad, rtol=3e-2, atol=2e-1)
