@@ -157,7 +157,10 @@ def dot_interaction_cat(bot: torch.Tensor, emb: torch.Tensor,
     kernel per direction — removes the feats-assembly cat AND the
     top-MLP input cat. Torch fallback composes the same math."""
     if bot.device.type == "cuda" and bot.shape[1] % 8 == 0:
-        return _DotInteractionCat.apply(bot, emb, p_pad)
+        out = _DotInteractionCat.apply(bot, emb, p_pad)
+        # keep the caller's compute dtype (the fp32 debug path feeds
+        # plain nn.Linear layers)
+        return out if out.dtype == bot.dtype else out.to(bot.dtype)
     feats = torch.cat([bot.unsqueeze(1), emb.to(bot.dtype)], dim=1)
     return torch.cat([bot, dot_interaction(feats, p_pad)], dim=1)
 
