@@ -72,6 +72,70 @@ class ElasticAgent:
         return int(self.store.get(_WORLD_KEY))
 
 
+def _export_full(base):
+    keys, values, freqs, versions = base.export()
+    names = list(base.storage.slabs)
+    rows = base.storage.export_slabs(names) if names else []
+    return {"keys": keys.cpu(), "values": values.cpu().float(),
+            "freqs": freqs.cpu(), "versions": versions.cpu(),
+            "slabs": {n: r.cpu() for n, r in zip(names, rows)},
+            "slab_init": {n: float(base.storage._slab_init.get(n, 0.0))
+                          for n in names}}
+
+
+def _import_owned(ev, shard, world, rank, collection):
+    base = getattr(ev, "local", ev)
+    keys = shard["keys"]
+    if collection:
+        from deeprec_amd.embedding.collection import KEY_BITS
+        raw = keys & ((1 << KEY_BITS) - 1)
+        mask = (raw % world) == rank
+    else:
+        mask = (keys % world) == rank
+    slab_rows = {n: r[mask] for n, r in shard["slabs"].items()}
+    for n, r in slab_rows.items():
+        base.storage.get_slab(n, r.shape[1], shard["slab_init"][n])
+    base.storage.import_(keys[mask].to(base.device),
+                         shard["values"][mask].to(base.device),
+                         shard["freqs"][mask], shard["versions"][mask],
+                         slab_rows=slab_rows or None)
+
+
+def live_resize(new_world: int, sharded_evs, reinit_fn) -> bool:
+    """IN-PROCESS cluster shrink (no restart): every rank exports its
+    shard state (values + optimizer slabs) and all-gathers it in the OLD
+    process group; the group is torn down; surviving ranks re-initialize
+    via `reinit_fn(rank, new_world)` and rebuild their shards under the
+    new routing. Departing ranks return False and should exit.
+
+    This is the live counterpart of the reference's UpdateServerDef
+    resize (elastic_grpc_server_lib.cc:294); growth still goes through
+    the restart path (new processes cannot be conjured in-process —
+    exactly torchrun's model)."""
+    import torch.distributed as dist
+
+    from deeprec_amd.parallel import comm
+    old_world = comm.world_size()
+    rank = comm.rank()
+    assert 0 < new_world <= old_world, \
+        "live_resize handles shrink; growth is restart-based"
+    payload = [_export_full(getattr(ev, "local", ev))
+               for ev in sharded_evs]
+    gathered = [None] * old_world
+    dist.all_gather_object(gathered, payload)
+    dist.destroy_process_group()
+    if rank >= new_world:
+        return False  # departed; shard state lives on in the survivors
+    reinit_fn(rank, new_world)
+    for i, ev in enumerate(sharded_evs):
+        collection = hasattr(ev, "export_tables")
+        ev.reshard(new_world, rank)
+        for rank_payload in gathered:
+            _import_owned(ev, rank_payload[i], new_world, rank,
+                          collection)
+    return True
+
+
 def elastic_step_hook(agent: ElasticAgent, saver, ckpt_dir: str,
                       global_step: int) -> bool:
     """Call between steps: on a scale event, write a full checkpoint and

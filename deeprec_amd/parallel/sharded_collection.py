@@ -92,6 +92,17 @@ class ShardedEmbeddingCollection:
     def export_tables(self, include_filtered=False):
         return self.local.export_tables(include_filtered)
 
+    def reshard(self, world: int, rank: int):
+        """Live-resize support: fresh local collection under the new
+        (world, rank) routing (parallel/elastic.live_resize re-imports)."""
+        old = self.local
+        self.world = world
+        self.rank = rank
+        self.local = EmbeddingCollection(
+            f"{self.name}/part_{rank}", list(old.table_names), self.dim,
+            old.ev_option, list(old.combiners), old.device,
+            trainable=old.trainable)
+
     def restore_table(self, table, keys, values, freqs=None, versions=None):
         mask = (keys % self.world) == self.rank
         self.local.restore_table(

@@ -85,6 +85,17 @@ class ShardedEmbeddingVariable:
     def get_slab(self, name, width=None, init_value=0.0, dtype=torch.float32):
         return self.local_ev.get_slab(name, width, init_value, dtype)
 
+    def reshard(self, world: int, rank: int):
+        """Live-resize support: swap in a FRESH local shard under the
+        new (world, rank) routing; the caller re-imports owned rows
+        (parallel/elastic.live_resize)."""
+        old = self.local_ev
+        self.world = world
+        self.rank = rank
+        self.local_ev = EmbeddingVariable(
+            f"{self.name}/part_{rank}", self.dim, old.value_dtype,
+            old.ev_option, old.device, trainable=old.trainable)
+
 
 class _ShardedPooledLookup(torch.autograd.Function):
     @staticmethod
