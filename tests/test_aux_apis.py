@@ -441,3 +441,60 @@ def test_fused_layer_norm_wrapper():
     torch.testing.assert_close(
         fused_layer_norm(x, w, b),
         torch.nn.functional.layer_norm(x, (16,), w, b))
+
+
+def test_feature_column_extended_surface():
+    """bucketized / crossed / weighted / indicator / identity /
+    vocabulary / adaptive columns through InputLayer (reference:
+    feature_column_v2.py surface)."""
+    import deeprec_amd.feature_column as fc
+    from deeprec_amd.embedding.variable import reset_registry
+
+    reset_registry()
+    torch.manual_seed(0)
+    b = 16
+    age = fc.numeric_column("age")
+    age_b = fc.bucketized_column(age, boundaries=[18, 25, 40, 65])
+    cross = fc.crossed_column(["c1", "c2"], hash_bucket_size=100)
+    cross_emb = fc.embedding_column(cross, dimension=8)
+    wcat = fc.weighted_categorical_column(
+        fc.categorical_column_with_embedding("items"), "item_w")
+    w_emb = fc.embedding_column(wcat, dimension=4, combiner="sum")
+    ind = fc.indicator_column(
+        fc.categorical_column_with_identity("slot", num_buckets=6))
+    voc = fc.categorical_column_with_vocabulary_list("voc",
+                                                     [10, 20, 30])
+    voc_emb = fc.embedding_column(voc, dimension=4)
+    ad = fc.categorical_column_with_adaptive_embedding(
+        "ad_ids", hash_bucket_size=50)
+    ad_emb = fc.adaptive_embedding_column(ad, dimension=8)
+
+    layer = fc.InputLayer([age_b, cross_emb, w_emb, ind, voc_emb, ad_emb])
+    feats = {
+        "age": torch.randint(10, 80, (b,)).float(),
+        "c1": torch.randint(0, 1000, (b,)),
+        "c2": torch.randint(0, 1000, (b,)),
+        "items": torch.randint(0, 40, (b, 3)),
+        "item_w": torch.rand(b, 3),
+        "slot": torch.randint(0, 6, (b, 2)),
+        "voc": torch.randint(0, 3, (b,)) * 10 + 10,
+        "ad_ids": torch.randint(0, 500, (b, 2)),
+    }
+    out = layer(feats, train=True)
+    # widths: 5 (bucketized one-hot) + 8 + 4 + 6 (indicator) + 4 + 8
+    assert out.shape == (b, 5 + 8 + 4 + 6 + 4 + 8)
+    assert torch.isfinite(out).all()
+    # bucketized one-hot is exactly one-hot
+    assert torch.equal(out[:, :5].sum(1), torch.ones(b))
+    # indicator multi-hot hits at most 2 slots
+    ind_part = out[:, 17:23]
+    assert ((ind_part.sum(1) >= 1) & (ind_part.sum(1) <= 2)).all()
+    # weighted column: scaling the weights scales the sum-combined rows
+    feats2 = dict(feats)
+    feats2["item_w"] = feats["item_w"] * 2
+    out2 = layer(feats2, train=False)
+    torch.testing.assert_close(out2[:, 13:17],
+                               layer(feats, train=False)[:, 13:17] * 2,
+                               rtol=1e-4, atol=1e-5)
+    # trainable params flow grads (crossed/weighted/adaptive EVs + static)
+    (out ** 2).sum().backward()
