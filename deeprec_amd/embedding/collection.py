@@ -246,7 +246,10 @@ class EmbeddingCollection:
 
     def _forward(self, uniq, slots, inverse, offsets_cat, weights_cat, batch,
                  out_dtype, emb_override=None):
-        if self.device.type == "cuda":
+        # serving storages without the HIP engine (CPU-offloaded or
+        # remote-KV rows under a CUDA model) take the generic gather+pool
+        # path below — it is pure torch and runs on either device
+        if self.device.type == "cuda" and hasattr(self.storage, "ext"):
             if emb_override is not None:
                 return self.storage.ext.group_pooled_fwd_direct(
                     emb_override.contiguous(), uniq, inverse, offsets_cat,
@@ -270,7 +273,8 @@ class EmbeddingCollection:
         nb = offsets_cat.numel() - 1  # == n_tables * batch
         lengths = (offsets_cat[1:] - offsets_cat[:-1]).long()
         row_ids_cat = torch.repeat_interleave(
-            torch.arange(nb, dtype=torch.int64), lengths)
+            torch.arange(nb, dtype=torch.int64,
+                         device=offsets_cat.device), lengths)
         for t in range(self.n_tables):
             lo, hi = t * batch, (t + 1) * batch
             jmask = (row_ids_cat >= lo) & (row_ids_cat < hi)

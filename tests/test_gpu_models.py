@@ -59,3 +59,36 @@ def test_native_extension_is_loaded():
     mod = be.load_extension()
     assert mod.__file__.endswith(".so")
     assert "deeprec_amd/_ext" in mod.__file__
+
+
+def test_device_placement_cpu_embeddings():
+    """Serving device-placement optimization: embeddings move to CPU,
+    dense stays on GPU; predictions match the all-GPU model exactly
+    (reference capability: Device-Placement.md)."""
+    from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.serving.device_placement import move_embeddings_to_cpu
+
+    torch.manual_seed(0)
+    m = DLRM(device=DEV, bf16=True, name_prefix="dp_test", num_sparse=6)
+    ds = CriteoSyntheticDataset(batch_size=128, device=DEV, seed=4,
+                                matrix_format=True)
+    opt = AdagradOptimizer(params=m.parameters(),
+                           embedding_variables=m.embedding_variables(),
+                           learning_rate=0.05)
+    for _ in range(2):
+        dense, ids, labels = ds.next_batch()
+        loss = m.loss_fn(m(dense, ids[:, :6]), labels)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    dense, ids, _ = ds.next_batch()
+    ref = m(dense, ids[:, :6], train=False)
+    before = torch.cuda.memory_allocated()
+    moved = move_embeddings_to_cpu(m)
+    assert moved > 0
+    out = m(dense, ids[:, :6], train=False)
+    torch.testing.assert_close(out, ref, rtol=1e-3, atol=1e-3)
+    assert m.collection.storage.memory_usage()["cpu_offloaded"]
+    del before
