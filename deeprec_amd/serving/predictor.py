@@ -142,13 +142,19 @@ class Predictor:
 
     def process(self, request: dict) -> dict:
         """JSON-ish request/response (processor.cc C-ABI contract shape):
-        {"dense": [[...]], "sparse": [[...ids...]]} -> {"probabilities":
-        [...]}"""
+        {"dense": [[...]], "sparse": [[...ids...]], "compress": bool}
+        -> {"probabilities": [...]}. compress=true deduplicates repeated
+        samples before the forward (sample-aware compression — ranking
+        batches repeat user/context rows across candidates)."""
         dense = torch.tensor(request["dense"], dtype=torch.float32,
                              device=self.group.device)
         sparse = torch.tensor(request["sparse"], dtype=torch.int64,
                               device=self.group.device)
-        probs = self.predict(dense, sparse)
+        if request.get("compress"):
+            from deeprec_amd.data.compression import compressed_forward
+            probs, _ = compressed_forward(self.predict, dense, sparse)
+        else:
+            probs = self.predict(dense, sparse)
         if isinstance(probs, list):
             return {"probabilities": [p.cpu().tolist() for p in probs]}
         return {"probabilities": probs.cpu().tolist()}

@@ -498,3 +498,53 @@ def test_feature_column_extended_surface():
                                rtol=1e-4, atol=1e-5)
     # trainable params flow grads (crossed/weighted/adaptive EVs + static)
     (out ** 2).sum().backward()
+
+
+def test_sample_aware_compression():
+    """Deduplicated forward is result-identical and runs the model on
+    the unique rows only (reference:
+    sample_awared_graph_compression.py)."""
+    from deeprec_amd.data.compression import (compress_batch,
+                                              compressed_forward)
+
+    torch.manual_seed(0)
+    base_d = torch.randn(4, 13)
+    base_s = torch.randint(0, 100, (4, 6))
+    idx = torch.tensor([0, 1, 0, 2, 1, 0, 3, 2])  # 8 samples, 4 unique
+    dense, sparse = base_d[idx], base_s[idx]
+    uniq, inverse = compress_batch(dense, sparse)
+    assert uniq[0].shape[0] == 4
+    torch.testing.assert_close(uniq[0][inverse], dense)
+    torch.testing.assert_close(uniq[1][inverse], sparse)
+
+    calls = {}
+
+    def fn(d, s):
+        calls["rows"] = d.shape[0]
+        return d.sum(1) + s.float().sum(1)
+
+    out, frac = compressed_forward(fn, dense, sparse)
+    assert calls["rows"] == 4 and abs(frac - 0.5) < 1e-6
+    torch.testing.assert_close(out, fn(dense, sparse))
+
+
+def test_predictor_compressed_process(tmp_path):
+    from deeprec_amd.checkpoint.saver import Saver
+    from deeprec_amd.models.dlrm import DLRM
+    from deeprec_amd.serving.predictor import Predictor
+
+    torch.manual_seed(3)
+    m = DLRM(device="cpu", bf16=False, num_sparse=4, name_prefix="cmp")
+    Saver(module=m,
+          embedding_variables=m.embedding_variables()).save(
+        str(tmp_path), 1)
+    pred = Predictor(m, str(tmp_path))
+    dense = torch.randn(2, 13)
+    ids = torch.randint(0, 50, (2, 4))
+    d = torch.cat([dense, dense]).tolist()
+    s = torch.cat([ids, ids]).tolist()
+    r1 = pred.process({"dense": d, "sparse": s})
+    r2 = pred.process({"dense": d, "sparse": s, "compress": True})
+    torch.testing.assert_close(torch.tensor(r1["probabilities"]),
+                               torch.tensor(r2["probabilities"]),
+                               rtol=1e-6, atol=1e-7)
