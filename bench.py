@@ -211,6 +211,20 @@ def main():
             model.collection.local.graph_mode = False
         for i in range(args.warmup):
             eager_step(i)
+      if distributed and world_size > 1:
+        # the fallback must be COORDINATED: a rank replaying a captured
+        # collective against a rank running the eager path would hang
+        import torch.distributed as dist
+        ok = torch.tensor([1 if use_graph else 0], device=device)
+        dist.all_reduce(ok)
+        if int(ok.item()) != world_size and use_graph:
+            print("bench: peer rank failed capture; using eager on all",
+                  flush=True)
+            use_graph = False
+            one_step = eager_step
+            model.collection.graph_mode = False
+            model.collection._pad_cap = None
+            model.collection.local.graph_mode = False
     else:
         for i in range(args.warmup):
             one_step(i)
