@@ -548,3 +548,32 @@ def test_predictor_compressed_process(tmp_path):
     torch.testing.assert_close(torch.tensor(r1["probabilities"]),
                                torch.tensor(r2["probabilities"]),
                                rtol=1e-6, atol=1e-7)
+
+
+def test_batch_caches_lru_lfu():
+    """BatchCache / LRUCache / LFUCache semantics (reference: cache.h,
+    incl. the LRU/LFU ordering cases of embedding_variable_ops_test)."""
+    from deeprec_amd.embedding.cache import LFUCache, LRUCache, make_cache
+    from deeprec_amd.embedding.options import CacheStrategy
+
+    lru = LRUCache()
+    lru.add_to_cache(torch.tensor([1, 2, 3]))
+    lru.add_to_cache(torch.tensor([2, 4]))
+    assert lru.size() == 4 and 3 in lru and 9 not in lru
+    # 1 and 3 are stalest; 2 was refreshed
+    assert lru.get_evict_ids(2).tolist() == [1, 3]
+    assert lru.get_cached_ids(2).tolist() == [4, 2]
+    # capacity auto-eviction
+    lru_cap = LRUCache(capacity=2)
+    lru_cap.add_to_cache(torch.tensor([1, 2, 3, 4]))
+    assert lru_cap.size() == 2 and 4 in lru_cap and 1 not in lru_cap
+
+    lfu = LFUCache()
+    lfu.add_to_cache(torch.tensor([5, 5, 6, 7]))
+    lfu.add_to_cache(torch.tensor([5, 6]))
+    # freq: 5->3, 6->2, 7->1
+    assert lfu.get_cached_ids(2).tolist() == [5, 6]
+    assert lfu.get_evict_ids(1).tolist() == [7]
+    assert lfu.size() == 2
+    assert isinstance(make_cache(CacheStrategy.LRU), LRUCache)
+    assert isinstance(make_cache(CacheStrategy.LFU), LFUCache)
