@@ -130,6 +130,40 @@ class Estimator:
             results.append(self.evaluate(eval_input_fn, steps=eval_steps))
         return results
 
+    @classmethod
+    def run_cluster(cls, model_fn: Callable, input_fn: Callable,
+                    tables: dict, tf_config=None, steps: int = 100,
+                    model_dir: str = None, config=None, params=None,
+                    ps_optimizer: str = "adagrad", ps_lr: float = 0.1):
+        """The multi-node Estimator story (reference: TF_CONFIG +
+        ClusterSpec + replica_device_setter, modelzoo train.py:858-913):
+        the SAME entry runs every task.
+
+        - task.type == "ps": host the EV shards behind the pull/push
+          plane (training/cluster.start_ps) and serve until killed;
+        - task.type in ("worker", "chief"): build the model via
+          model_fn(params) where params["embeddings"] maps table name ->
+          PsShardedEmbedding, then run the normal Estimator train loop
+          (sparse grads ride the PS pushes inside backward).
+        """
+        from deeprec_amd.training.cluster import (parse_tf_config,
+                                                  start_ps,
+                                                  worker_embeddings)
+        cfg = parse_tf_config(tf_config)
+        if cfg["type"] == "ps":
+            server = start_ps(cfg, tables, optimizer=ps_optimizer,
+                              lr=ps_lr, checkpoint_dir=model_dir)
+            server._thread.join()  # serve until the process is killed
+            return None
+        p = dict(params or {})
+        p["embeddings"] = worker_embeddings(cfg, tables)
+        p["cluster"] = cfg
+        est = cls(model_fn, model_dir=model_dir, config=config, params=p)
+        est.train(input_fn, steps=steps)
+        for emb in p["embeddings"].values():
+            emb.flush()
+        return est
+
     @torch.no_grad()
     def predict(self, input_fn: Callable):
         for batch in input_fn():
