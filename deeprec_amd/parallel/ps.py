@@ -148,7 +148,10 @@ class PsServer:
             g.index_add_(0, inverse, grads.float())
             slots = ev.storage.lookup(uniq)
             ev.accumulate_grad(slots, uniq, g)
-            self.opt.step()
+            # the PS owns its shard's optimizer state but NOT the
+            # training step counter (matters when a PS shares a process
+            # with a worker, e.g. tests)
+            self.opt.step(increment_global_step=False)
             self._applied += 1
 
     def save(self, directory: str, step: int) -> str:
