@@ -51,10 +51,17 @@ def build_argparser():
                    help="embedding-parallel across ranks")
     p.add_argument("--micro_batch", type=int, default=1,
                    help="gradient-accumulation sub-batches per step")
-    p.add_argument("--storage", choices=["hbm", "hbm_dram"], default="hbm",
-                   help="EV storage tier (hbm_dram = host cold tier)")
+    p.add_argument("--storage",
+                   choices=["hbm", "hbm_dram", "hbm_dram_ssd"],
+                   default="hbm",
+                   help="EV storage tier (hbm_dram = pinned host cold "
+                        "tier; hbm_dram_ssd adds append-only SSD files)")
     p.add_argument("--hot_bytes", type=int, default=256 << 20,
-                   help="HBM hot-tier budget for --storage hbm_dram")
+                   help="HBM hot-tier budget for multi-tier storage")
+    p.add_argument("--dram_bytes", type=int, default=1 << 30,
+                   help="DRAM middle-tier budget for hbm_dram_ssd")
+    p.add_argument("--storage_path", default=None,
+                   help="SSD directory for hbm_dram_ssd")
     p.add_argument("--parquet", default=None,
                    help="train from a parquet file (columns: label, "
                         "dense_0..12, sparse_0..N) instead of synthetic")
@@ -71,6 +78,13 @@ def make_ev_option(args) -> EmbeddingVariableOption:
         opt.storage_option = StorageOption(
             storage_type=StorageType.HBM_DRAM,
             storage_size=[args.hot_bytes])
+    elif args.storage == "hbm_dram_ssd":
+        from deeprec_amd.embedding.options import StorageOption, StorageType
+        assert args.storage_path, "--storage hbm_dram_ssd needs --storage_path"
+        opt.storage_option = StorageOption(
+            storage_type=StorageType.HBM_DRAM_SSD,
+            storage_size=[args.hot_bytes, args.dram_bytes],
+            storage_path=args.storage_path)
     if args.ev_filter == "counter":
         opt.filter_option = CounterFilter(filter_freq=args.filter_freq)
     elif args.ev_filter == "cbf":
