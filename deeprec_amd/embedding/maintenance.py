@@ -40,6 +40,7 @@ class EvictionManager:
         self._lock = threading.Lock()
         self._worker: Optional[threading.Thread] = None
         self._last_kick = -1
+        self._interval = interval_steps
         self.stats = {"chunks_applied": 0, "rows_promoted": 0,
                       "scores_computed": 0}
 
@@ -54,7 +55,7 @@ class EvictionManager:
         every interval; applies at most one bounded chunk."""
         if not self._evs:
             return
-        if (global_step - self._last_kick >= self.interval_steps
+        if (global_step - self._last_kick >= self._interval
                 and (self._worker is None or not self._worker.is_alive())):
             self._last_kick = global_step
             self._worker = threading.Thread(target=self._score_all,
@@ -122,11 +123,18 @@ class EvictionManager:
         ds = hot_scores[hot_order[:k]]
         keep = ps > ds
         promote_k, demote_k = promote_k[keep], demote_k[keep]
+        self.stats["scores_computed"] += 1
         if promote_k.numel() == 0:
+            # placement converged: back the scoring cadence off
+            # exponentially (a 120M-row score costs ~tens of ms of GPU
+            # contention; re-running it every interval after convergence
+            # taxed steady-state steps ~5x)
+            self._interval = min(self._interval * 2,
+                                 self.interval_steps * 64)
             return
+        self._interval = self.interval_steps
         with self._lock:
             self._plans[id(st)] = (promote_k.cpu(), demote_k.cpu())
-        self.stats["scores_computed"] += 1
 
     # ---------------- bounded chunk application ----------------
     def _apply_chunk(self, st, plan):
