@@ -138,7 +138,41 @@ __global__ void k_linear_fwd_t(const short* __restrict__ A,
                                                             0);
     }
   }
-  // D: col = lane%16, row = 4*(lane/16) + i
+  // D: col = lane%16, row = 4*(lane/16) + i. The MFMA output layout
+  // writes 32B fragments per instruction; for FULL 16x64 tiles, stage
+  // the tile in LDS and flush 128B row bursts (the fragmented stores
+  // capped this kernel at ~520 GB/s on an 8 MB output).
+  __shared__ short cstage[4][16][72];  // 72 = 64 + 8 (bank de-phase)
+  const bool full_tile =
+      (MT == 1) && (n0 + 64 <= N) && (m0 + 16 <= M) && ((N & 7) == 0);
+  if (full_tile) {
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      const int cn = t * 16 + (lane & 15);
+      const float bv = bias ? bias[n0 + cn] : 0.0f;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        float v = acc[0][t][i] + bv;
+        if (act == 1 && v < 0.0f) v = 0.0f;
+        else if (act == 2) v = 1.0f / (1.0f + __expf(-v));
+        cstage[wave][(lane >> 4) * 4 + i][cn] = f2bf_u16(v);
+      }
+    }
+    __builtin_amdgcn_s_waitcnt(0);  // LDS writes visible within the wave
+    // flush: 8 lanes per row x 8 cols each = one 16B store per lane,
+    // two row-groups of 8 -> full 128B bursts per row
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      const int row = half * 8 + (lane >> 3);
+      const int c8 = (lane & 7) * 8;
+      bf16x8 vreg;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) vreg[j] = cstage[wave][row][c8 + j];
+      *reinterpret_cast<bf16x8*>(
+          C + (int64_t)(m0 + row) * N + n0 + c8) = vreg;
+    }
+    return;
+  }
 #pragma unroll
   for (int r = 0; r < MT; ++r) {
 #pragma unroll
