@@ -21,7 +21,8 @@ _EXT_DIR = os.path.normpath(os.path.join(_HERE, "..", "_ext"))
 _SOURCES = [os.path.join(_HERE, "hip", "ev_kernels.hip"),
             os.path.join(_HERE, "hip", "dense_kernels.hip"),
             os.path.join(_HERE, "hip", "gru_kernels.hip"),
-            os.path.join(_HERE, "hip", "attention_kernels.hip")]
+            os.path.join(_HERE, "hip", "attention_kernels.hip"),
+            os.path.join(_HERE, "hip", "fp8_kernels.hip")]
 _HIPCC = os.environ.get("HIPCC", "/opt/rocm/bin/hipcc")
 _ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 

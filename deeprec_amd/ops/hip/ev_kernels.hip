@@ -1993,11 +1993,13 @@ void apply_ftrl(torch::Tensor w, torch::Tensor n, torch::Tensor z,
 void register_dense(pybind11::module_& mod);      // dense_kernels.hip
 void register_gru(pybind11::module_& mod);        // gru_kernels.hip
 void register_attention(pybind11::module_& mod);  // attention_kernels.hip
+void register_fp8(pybind11::module_& mod);        // fp8_kernels.hip
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, mod) {
   register_dense(mod);
   register_gru(mod);
   register_attention(mod);
+  register_fp8(mod);
   mod.def("ht_lookup_insert", &ht_lookup_insert);
   mod.def("ht_dedup_a", &ht_dedup_a);
   mod.def("ht_dedup_b", &ht_dedup_b);

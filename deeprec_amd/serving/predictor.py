@@ -58,9 +58,16 @@ class Predictor:
 
     def __init__(self, model, checkpoint_dir: str,
                  num_sessions: int = 2, device=None,
-                 remote_sparse: bool = False):
+                 remote_sparse: bool = False, fp8_mlp: bool = False):
         self.model = model
         self.dir = checkpoint_dir
+        # fp8_mlp: serve the dense MLPs as OCP e4m3 (ops/fp8.py) — the
+        # converter is reverted around full model updates so restores
+        # land in the fp32 originals, then re-quantized
+        self._fp8 = None
+        if fp8_mlp:
+            from deeprec_amd.ops.fp8 import Fp8MlpConverter
+            self._fp8 = Fp8MlpConverter(model)
         # remote_sparse = RemoteSessionInstance mode: sparse weights live
         # in an external feature store (attach_remote_store), so only the
         # dense module restores from checkpoints here
@@ -72,6 +79,8 @@ class Predictor:
         self._loaded_full: Optional[str] = None
         self._lock = threading.Lock()
         self.reload()
+        if self._fp8 is not None and not self._fp8._sites:
+            self._fp8.convert()  # no checkpoint yet: quantize init weights
 
     # ------------- model update -------------
     def reload(self) -> bool:
@@ -80,7 +89,11 @@ class Predictor:
         if ck is None or ck == self._loaded_full:
             return False
         with self._lock:
+            if self._fp8 is not None:
+                self._fp8.revert()
             self.saver.restore(ck)  # also replays newer incr deltas
+            if self._fp8 is not None:
+                self._fp8.convert()
             self._loaded_full = ck
             self._applied = {p for p in self._incr_paths()}
         return True
