@@ -87,6 +87,9 @@ class DIN(_SeqBase):
             t16 = target.to(cd).unsqueeze(1).expand_as(s16)
             att_in = torch.cat([s16, t16, s16 - t16, s16 * t16], dim=2)
             scores = self.att(att_in).float().squeeze(2)    # [B, T]
+        if seq.device.type == "cuda":
+            from deeprec_amd.ops.fused_attention import masked_softmax_pool
+            return masked_softmax_pool(scores, seq, mask)
         scores = scores.masked_fill(~mask, -1e9)
         w = torch.softmax(scores, dim=1)
         return (w.unsqueeze(2) * seq).sum(1)

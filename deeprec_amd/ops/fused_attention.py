@@ -176,3 +176,41 @@ class FusedTransformerEncoder(nn.Module):
         for layer in self.layers:
             x = layer(x, key_pad)
         return x
+
+
+class _MaskedSoftmaxPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, scores, seq, mask_u8):
+        from deeprec_amd.ops.build_ext import require_extension
+        ext = require_extension()
+        out, w = ext.msm_pool_fwd(scores, seq, mask_u8)
+        ctx.ext = ext
+        ctx.save_for_backward(seq, w, mask_u8)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        seq, w, mask_u8 = ctx.saved_tensors
+        dscores, dseq = ctx.ext.msm_pool_bwd(dout, seq, w, mask_u8)
+        return dscores, dseq, None
+
+
+def masked_softmax_pool(scores: torch.Tensor, seq: torch.Tensor,
+                        mask: torch.Tensor) -> torch.Tensor:
+    """softmax(scores masked to the valid positions) @ seq, fused.
+
+    scores [B, T] fp32, seq [B, T, D] fp32, mask [B, T] bool (True =
+    valid). A fully-masked row pools to ZEROS (masked positions train
+    nothing — torch softmax of an all -inf row would give uniform 1/T
+    instead). One kernel per direction (attention_kernels.hip) replaces
+    the masked_fill / softmax / broadcast-mul / sum chain and its
+    backward — the dominant torch-glue cost of the DIN step."""
+    if scores.is_cuda:
+        return _MaskedSoftmaxPool.apply(
+            scores.float().contiguous(), seq.float().contiguous(),
+            mask.to(torch.uint8).contiguous())
+    mf = mask.to(scores.dtype)
+    s = scores.float() - scores.float().amax(dim=1, keepdim=True)
+    e = torch.exp(s) * mf
+    w = e / (e.sum(dim=1, keepdim=True) + 1e-20)
+    return (w.unsqueeze(2) * seq.float()).sum(1)

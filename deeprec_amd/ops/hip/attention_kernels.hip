@@ -298,6 +298,111 @@ __global__ void k_din_feat_bwd_dt(const short* __restrict__ g,
   }
 }
 
+// ------------------------------------------------------------------
+// DIN attention tail: masked softmax over scores [B, T] + weighted
+// pooling of seq [B, T, D] fused into ONE kernel per direction. The
+// torch chain (masked_fill, softmax, [B,T,D] broadcast-mul
+// materialization, sum, + their backwards) was ~8 launches and the
+// dominant torch-glue cost of the DIN step (prof_din2.csv). A fully
+// masked row pools to zeros (matches "masked positions train nothing").
+// ------------------------------------------------------------------
+__global__ void k_msm_pool_fwd(const float* __restrict__ scores,
+                               const float* __restrict__ seq,
+                               const uint8_t* __restrict__ mask,
+                               int B, int T, int D,
+                               float* __restrict__ w_out,
+                               float* __restrict__ out) {
+  extern __shared__ float smw[];  // [T]
+  __shared__ float red[256];
+  const int b = blockIdx.x;
+  if (b >= B) return;
+  float lmax = -3.4e38f;
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    float s = mask[b * T + t] ? scores[b * T + t] : -3.4e38f;
+    smw[t] = s;
+    lmax = fmaxf(lmax, s);
+  }
+  red[threadIdx.x] = lmax;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off)
+      red[threadIdx.x] = fmaxf(red[threadIdx.x], red[threadIdx.x + off]);
+    __syncthreads();
+  }
+  const float m = red[0];
+  __syncthreads();
+  float lsum = 0.0f;
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    float e = (smw[t] <= -3.0e38f) ? 0.0f : __expf(smw[t] - m);
+    smw[t] = e;
+    lsum += e;
+  }
+  red[threadIdx.x] = lsum;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  const float inv = 1.0f / (red[0] + 1e-20f);
+  __syncthreads();
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    smw[t] *= inv;
+    w_out[b * T + t] = smw[t];
+  }
+  __syncthreads();
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float acc = 0.0f;
+    const float* sp = seq + (int64_t)b * T * D + d;
+    for (int t = 0; t < T; ++t) acc += smw[t] * sp[(int64_t)t * D];
+    out[(int64_t)b * D + d] = acc;
+  }
+}
+
+// dscores[t] = mask ? w[t] * (dout.seq[t] - sum_u w[u] (dout.seq[u])) : 0
+// dseq[t, d] = w[t] * dout[d]
+__global__ void k_msm_pool_bwd(const float* __restrict__ dout,
+                               const float* __restrict__ seq,
+                               const float* __restrict__ w,
+                               const uint8_t* __restrict__ mask,
+                               int B, int T, int D,
+                               float* __restrict__ dscores,
+                               float* __restrict__ dseq) {
+  extern __shared__ float sdot[];  // [T] row dots, then [D] dout row
+  __shared__ float red[256];
+  const int b = blockIdx.x;
+  if (b >= B) return;
+  float* sdo = sdot + T;
+  for (int d = threadIdx.x; d < D; d += blockDim.x)
+    sdo[d] = dout[(int64_t)b * D + d];
+  __syncthreads();
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    const float* sp = seq + ((int64_t)b * T + t) * D;
+    float acc = 0.0f;
+    for (int d = 0; d < D; ++d) acc += sdo[d] * sp[d];
+    sdot[t] = acc;
+  }
+  __syncthreads();
+  float lw = 0.0f;
+  for (int t = threadIdx.x; t < T; t += blockDim.x)
+    lw += w[b * T + t] * sdot[t];
+  red[threadIdx.x] = lw;
+  __syncthreads();
+  for (int off = blockDim.x >> 1; off > 0; off >>= 1) {
+    if (threadIdx.x < off) red[threadIdx.x] += red[threadIdx.x + off];
+    __syncthreads();
+  }
+  const float wdot = red[0];
+  for (int t = threadIdx.x; t < T; t += blockDim.x) {
+    const float wt = w[b * T + t];
+    dscores[b * T + t] =
+        mask[b * T + t] ? wt * (sdot[t] - wdot) : 0.0f;
+  }
+  for (int i = threadIdx.x; i < T * D; i += blockDim.x) {
+    const int t = i / D;
+    dseq[(int64_t)b * T * D + i] = w[b * T + t] * sdo[i - t * D];
+  }
+}
+
 }  // namespace
 
 static const short* att_bf_ptr(const torch::Tensor& t) {
@@ -401,9 +506,47 @@ std::tuple<torch::Tensor, torch::Tensor> din_feat_bwd(torch::Tensor g,
   return {dseq, dtgt};
 }
 
+std::tuple<torch::Tensor, torch::Tensor> msm_pool_fwd(torch::Tensor scores,
+                                                      torch::Tensor seq,
+                                                      torch::Tensor mask) {
+  TORCH_CHECK(scores.scalar_type() == torch::kFloat32
+                  && scores.is_contiguous());
+  TORCH_CHECK(seq.scalar_type() == torch::kFloat32 && seq.is_contiguous());
+  TORCH_CHECK(mask.scalar_type() == torch::kUInt8 && mask.is_contiguous());
+  int B = scores.size(0), T = scores.size(1), D = seq.size(2);
+  TORCH_CHECK(seq.size(0) == B && seq.size(1) == T && mask.numel() == B * T);
+  TORCH_CHECK(T <= 4096, "sequence too long for the per-sample block");
+  auto w = torch::empty_like(scores);
+  auto out = torch::empty({(int64_t)B, (int64_t)D}, seq.options());
+  if (B == 0) return {out, w};
+  k_msm_pool_fwd<<<B, 128, (size_t)T * sizeof(float), att_stream()>>>(
+      scores.data_ptr<float>(), seq.data_ptr<float>(),
+      mask.data_ptr<uint8_t>(), B, T, D, w.data_ptr<float>(),
+      out.data_ptr<float>());
+  return {out, w};
+}
+
+std::tuple<torch::Tensor, torch::Tensor> msm_pool_bwd(torch::Tensor dout,
+                                                      torch::Tensor seq,
+                                                      torch::Tensor w,
+                                                      torch::Tensor mask) {
+  int B = seq.size(0), T = seq.size(1), D = seq.size(2);
+  auto dc = dout.contiguous();
+  auto dscores = torch::empty_like(w);
+  auto dseq = torch::empty_like(seq);
+  if (B == 0) return {dscores, dseq};
+  k_msm_pool_bwd<<<B, 128, (size_t)(T + D) * sizeof(float), att_stream()>>>(
+      dc.data_ptr<float>(), seq.data_ptr<float>(), w.data_ptr<float>(),
+      mask.data_ptr<uint8_t>(), B, T, D, dscores.data_ptr<float>(),
+      dseq.data_ptr<float>());
+  return {dscores, dseq};
+}
+
 void register_attention(pybind11::module_& mod) {
   mod.def("mha_fwd", &mha_fwd);
   mod.def("mha_bwd", &mha_bwd);
   mod.def("din_feat_fwd", &din_feat_fwd);
   mod.def("din_feat_bwd", &din_feat_bwd);
+  mod.def("msm_pool_fwd", &msm_pool_fwd);
+  mod.def("msm_pool_bwd", &msm_pool_bwd);
 }

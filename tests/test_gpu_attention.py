@@ -141,3 +141,56 @@ def test_din_att_features_matches_torch():
     ref.backward(g)
     torch.testing.assert_close(seq.grad, sr.grad, rtol=2e-2, atol=2e-2)
     torch.testing.assert_close(tgt.grad, tr.grad, rtol=2e-2, atol=2e-1)
+
+
+def test_masked_softmax_pool_matches_torch():
+    from deeprec_amd.ops.fused_attention import masked_softmax_pool
+
+    torch.manual_seed(5)
+    b, t, d = 257, 50, 32
+    scores = torch.randn(b, t, device=DEV, requires_grad=True)
+    seq = torch.randn(b, t, d, device=DEV, requires_grad=True)
+    mask = torch.rand(b, t, device=DEV) > 0.3
+    mask[0] = False       # fully masked row -> zero pooled output
+    mask[:, 0] = True     # every other row has >=1 valid position
+    mask[0, :] = False
+
+    out = masked_softmax_pool(scores, seq, mask)
+    sr = scores.detach().clone().requires_grad_(True)
+    qr = seq.detach().clone().requires_grad_(True)
+    mf = mask.float()
+    e = torch.exp(sr - sr.amax(dim=1, keepdim=True)) * mf
+    w = e / (e.sum(dim=1, keepdim=True) + 1e-20)
+    ref = (w.unsqueeze(2) * qr).sum(1)
+    torch.testing.assert_close(out, ref, rtol=1e-4, atol=1e-5)
+    assert bool((out[0] == 0).all())
+
+    g = torch.randn(b, d, device=DEV)
+    out.backward(g)
+    ref.backward(g)
+    torch.testing.assert_close(scores.grad, sr.grad, rtol=1e-3, atol=1e-5)
+    torch.testing.assert_close(seq.grad, qr.grad, rtol=1e-4, atol=1e-6)
+    assert bool((scores.grad[0] == 0).all())
+
+
+def test_din_trains_with_fused_attend_tail():
+    from deeprec_amd.data.synthetic import CriteoSyntheticDataset
+    from deeprec_amd.models.sequence import DIN
+    from deeprec_amd.optimizers import AdamAsyncOptimizer
+    torch.manual_seed(6)
+    m = DIN(device=DEV, bf16=True)
+    ds = CriteoSyntheticDataset(batch_size=512, seed=7, device=DEV)
+    opt = AdamAsyncOptimizer(params=m.parameters(),
+                             embedding_variables=m.embedding_variables(),
+                             learning_rate=0.01)
+    first = None
+    for i in range(4):
+        dense, ids, seq, target, labels = ds.next_seq_batch(seq_len=50)
+        loss = m.loss_fn(m(dense, ids[:, :m.num_sparse], seq, target),
+                         labels)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+        if first is None:
+            first = float(loss)
+    assert torch.isfinite(loss) and float(loss) < first + 0.5
