@@ -146,3 +146,22 @@ def test_fp8_gather_output_halves_bytes():
     assert payload < bf16_payload
     deq = dequantize_fp8_rows(q, s)
     assert (deq - rows).abs().max() < rows.abs().max() * 0.1
+
+
+def test_masked_softmax_pool_cpu_fallback():
+    """CPU fallback of the fused DIN attend tail matches the plain torch
+    chain (and zeroes fully-masked rows)."""
+    from deeprec_amd.ops.fused_attention import masked_softmax_pool
+    torch.manual_seed(5)
+    b, t, d = 8, 11, 4
+    scores = torch.randn(b, t)
+    seq = torch.randn(b, t, d)
+    mask = torch.rand(b, t) > 0.4
+    mask[:, 0] = True
+    mask[3] = False
+    out = masked_softmax_pool(scores, seq, mask)
+    w = torch.softmax(scores.masked_fill(~mask, -1e9), 1)
+    ref = (w.unsqueeze(2) * seq).sum(1)
+    live = mask.any(1)
+    torch.testing.assert_close(out[live], ref[live], rtol=1e-5, atol=1e-6)
+    assert bool((out[3] == 0).all())
