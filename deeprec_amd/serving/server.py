@@ -93,14 +93,60 @@ class DynamicBatcher:
                     fut.set_exception(e)
 
 
+class _Metrics:
+    """Prometheus metrics for the serving shell (one registry per app so
+    tests can build many apps in one process). The reference ships no
+    Prometheus surface (SURVEY §5 'nothing Prometheus-like'); a
+    production serving deployment needs one, so this exceeds parity.
+    No-ops gracefully when prometheus_client is unavailable."""
+
+    def __init__(self):
+        try:
+            from prometheus_client import (CollectorRegistry, Counter,
+                                           Gauge, Histogram)
+        except ImportError:  # pragma: no cover
+            self.registry = None
+            return
+        self.registry = CollectorRegistry()
+        self.requests = Counter(
+            "deeprec_requests_total", "Predict requests",
+            ["endpoint", "status"], registry=self.registry)
+        self.latency = Histogram(
+            "deeprec_request_seconds", "Predict latency", ["endpoint"],
+            buckets=(.0005, .001, .0025, .005, .01, .025, .05, .1, .25,
+                     .5, 1.0, 2.5), registry=self.registry)
+        self.rows = Histogram(
+            "deeprec_request_rows", "Rows per request",
+            buckets=(1, 2, 4, 8, 16, 32, 64, 128, 256, 512, 1024),
+            registry=self.registry)
+        self.reloads = Counter(
+            "deeprec_model_reloads_total", "Full model updates applied",
+            registry=self.registry)
+
+    def observe(self, endpoint: str, status: str, secs: float,
+                rows: Optional[int] = None):
+        if self.registry is None:
+            return
+        self.requests.labels(endpoint, status).inc()
+        self.latency.labels(endpoint).observe(secs)
+        if rows is not None:
+            self.rows.observe(rows)
+
+    def expose(self):
+        from prometheus_client import generate_latest
+        return generate_latest(self.registry)
+
+
 def create_app(predictor: Predictor, batcher: Optional[DynamicBatcher] = None):
     """Build the FastAPI app. Endpoints are sync `def`s — FastAPI runs them
     on its threadpool, and the batcher coalesces across those threads."""
-    from fastapi import Body, FastAPI, HTTPException
+    from fastapi import Body, FastAPI, HTTPException, Response
 
     app = FastAPI(title="deeprec_amd serving")
     app.state.predictor = predictor
     app.state.batcher = batcher
+    metrics = _Metrics()
+    app.state.metrics = metrics
 
     @app.get("/health")
     def health():
@@ -108,8 +154,16 @@ def create_app(predictor: Predictor, batcher: Optional[DynamicBatcher] = None):
                 "checkpoint": predictor._loaded_full,
                 "device": str(predictor.group.device)}
 
+    @app.get("/metrics")
+    def prometheus_metrics():
+        if metrics.registry is None:
+            raise HTTPException(501, "prometheus_client not installed")
+        return Response(metrics.expose(),
+                        media_type="text/plain; version=0.0.4")
+
     @app.post("/v1/predict")
     def predict(request: dict = Body(...)):
+        t0 = time.monotonic()
         try:
             if batcher is not None:
                 dense = torch.tensor(request["dense"], dtype=torch.float32,
@@ -117,21 +171,34 @@ def create_app(predictor: Predictor, batcher: Optional[DynamicBatcher] = None):
                 sparse = torch.tensor(request["sparse"], dtype=torch.int64,
                                       device=predictor.group.device)
                 probs = batcher.submit(dense, sparse).result(timeout=30)
-                return {"probabilities": probs.cpu().tolist()}
-            return predictor.process(request)
+                out = {"probabilities": probs.cpu().tolist()}
+            else:
+                out = predictor.process(request)
+            metrics.observe("predict", "ok", time.monotonic() - t0,
+                            len(out["probabilities"]))
+            return out
         except KeyError as e:
+            metrics.observe("predict", "422", time.monotonic() - t0)
             raise HTTPException(422, f"missing field {e}")
         except (ValueError, TypeError, RuntimeError) as e:
             # ragged lists, wrong dtypes, shape mismatches
+            metrics.observe("predict", "422", time.monotonic() - t0)
             raise HTTPException(422, f"malformed request: {e}")
 
     @app.post("/v1/predict_batch")
     def predict_batch(requests: list = Body(...)):
-        return predictor.batch_process(requests)
+        t0 = time.monotonic()
+        out = predictor.batch_process(requests)
+        metrics.observe("predict_batch", "ok", time.monotonic() - t0,
+                        len(requests))
+        return out
 
     @app.post("/v1/reload")
     def reload():
-        return {"reloaded": predictor.reload(),
+        ok = predictor.reload()
+        if ok and metrics.registry is not None:
+            metrics.reloads.inc()
+        return {"reloaded": ok,
                 "checkpoint": predictor._loaded_full}
 
     return app

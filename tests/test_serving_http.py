@@ -161,3 +161,28 @@ def test_feature_store_publish_and_lookup(tmp_path):
     missing = store_backed_lookup(fstore2, name,
                                   torch.tensor([10 ** 12]), 4, default=0.5)
     assert torch.allclose(missing, torch.full((1, 4), 0.5))
+
+
+def test_prometheus_metrics_endpoint(tmp_path):
+    """/metrics exposes request counters and latency histograms
+    (observability beyond the reference's hook-based step metrics,
+    SURVEY §5)."""
+    pytest.importorskip("prometheus_client")
+    model = _make_ckpt(tmp_path)
+    pred = Predictor(model, str(tmp_path), num_sessions=1, device="cpu")
+    client = TestClient(create_app(pred))
+
+    req = {"dense": torch.randn(3, 13).tolist(),
+           "sparse": torch.randint(0, 100, (3, 4)).tolist()}
+    assert client.post("/v1/predict", json=req).status_code == 200
+    assert client.post("/v1/predict", json={"dense": [[1.0]]}
+                       ).status_code == 422
+    body = client.get("/metrics").text
+    assert 'deeprec_requests_total{endpoint="predict",status="ok"} 1.0' \
+        in body
+    assert 'deeprec_requests_total{endpoint="predict",status="422"} 1.0' \
+        in body
+    assert "deeprec_request_seconds_bucket" in body
+    # two apps in one process: per-app registries must not collide
+    client2 = TestClient(create_app(pred))
+    assert client2.get("/metrics").status_code == 200
