@@ -100,6 +100,12 @@ class EmbeddingVariable:
             self._recorded_ids.append(uniq_keys.detach())
         return slots
 
+    def lookup_tier(self, keys: torch.Tensor) -> torch.Tensor:
+        """Per-key storage tier: −1 absent, 0 HBM/resident, 1 DRAM,
+        2 SSD (reference: KvResourceLookupTier op,
+        kernels/kv_variable_lookup_ops.cc:537)."""
+        return self.storage.lookup_tier(keys)
+
     def gather(self, keys: torch.Tensor, out_dtype=None) -> torch.Tensor:
         """Inference read: no insert, default value for missing keys."""
         uniq, inverse = torch.unique(keys, return_inverse=True)

@@ -237,6 +237,14 @@ class CpuStorage:
         """Probe without insert (serving path)."""
         return self.lookup_or_create(keys, None, 0, train=False)
 
+    def lookup_tier(self, keys: torch.Tensor) -> torch.Tensor:
+        """Reference: KvResourceLookupTier
+        (kernels/kv_variable_lookup_ops.cc:537): −1 = not present,
+        0 = the (single) resident tier."""
+        slots = self.lookup(keys).long()
+        return torch.where(slots >= 0, torch.zeros_like(slots),
+                           torch.full_like(slots, -1))
+
     def gather(self, keys: torch.Tensor, slots: torch.Tensor,
                out_dtype=None) -> torch.Tensor:
         """values[slots] with default-value fill for slot<0."""

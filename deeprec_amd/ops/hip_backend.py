@@ -214,6 +214,23 @@ class HbmStorage:
         slots, _ = self.ext.ht_lookup(keys, self.ht_keys, self.ht_slot, False)
         return slots
 
+    def lookup_tier(self, keys) -> torch.Tensor:
+        """Per-key storage tier (reference: KvResourceLookupTier,
+        kernels/kv_variable_lookup_ops.cc:537): −1 = not present,
+        0 = HBM. Multi-tier storages override with 1 = DRAM, 2 = SSD."""
+        slots = self.lookup(keys).long()
+        hot = getattr(self, "hot_rows", None)
+        tier = torch.where(slots >= 0,
+                           torch.zeros_like(slots),
+                           torch.full_like(slots, -1))
+        if hot is not None:
+            tier = torch.where(slots >= hot, torch.ones_like(slots), tier)
+            ssd_base = getattr(self, "ssd_base", None)
+            if ssd_base is not None:
+                tier = torch.where(slots >= ssd_base,
+                                   torch.full_like(slots, 2), tier)
+        return tier
+
     def enable_graph_mode(self, expected_entries: int, expected_slots: int):
         """hipGraph capture prep: pre-size the table/slabs (no growth can
         happen inside a captured step) and move the dedup epoch/step

@@ -248,3 +248,14 @@ def test_memory_usage_accounting():
     assert mu["slab_bytes"] > 0
     assert mu["total_bytes"] == (mu["table_bytes"] + mu["values_bytes"]
                                  + mu["slab_bytes"])
+
+
+def test_lookup_tier_cpu():
+    """KvResourceLookupTier parity on the CPU tier: 0 resident, -1
+    absent (reference: kv_variable_lookup_ops.cc:537)."""
+    import torch
+    from deeprec_amd.embedding import EmbeddingVariable
+    ev = EmbeddingVariable("tier_cpu/ev", 4)
+    ev.lookup_or_create(torch.tensor([1, 2, 3]))
+    tier = ev.lookup_tier(torch.tensor([1, 2, 3, 99]))
+    assert tier.tolist() == [0, 0, 0, -1]

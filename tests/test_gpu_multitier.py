@@ -298,3 +298,27 @@ def test_three_tier_hbm_dram_ssd(tmp_path):
     # SSD compaction round-trips under the storage API
     st.compact_ssd(sync=True)
     torch.testing.assert_close(embedding_lookup(ev, ids), after)
+
+
+def test_lookup_tier_reports_placement():
+    """KvResourceLookupTier parity (reference:
+    kv_variable_lookup_ops.cc:537): -1 absent, 0 HBM, 1 DRAM cold."""
+    from deeprec_amd.embedding.options import (EmbeddingVariableOption,
+                                               StorageOption, StorageType)
+    from deeprec_amd.ops.hbm_dram_backend import HbmDramStorage
+
+    dim, hot_rows = 8, 32
+    opt = EmbeddingVariableOption(storage_option=StorageOption(
+        storage_type=StorageType.HBM_DRAM,
+        storage_size=[hot_rows * dim * 4]))
+    st = HbmDramStorage(dim, opt, device="cuda")
+    keys = torch.arange(64, dtype=torch.int64, device="cuda")
+    st.lookup_or_create(keys, torch.ones(64, dtype=torch.int32,
+                                         device="cuda"), step=0)
+    probe = torch.cat([keys, torch.tensor([1 << 40], device="cuda")])
+    tier = st.lookup_tier(probe)
+    assert int(tier[-1]) == -1                      # never inserted
+    counts = torch.bincount(tier[:-1].clamp(min=0), minlength=2)
+    assert int(counts[0]) == hot_rows               # hot tier full
+    assert int(counts[1]) == 64 - hot_rows          # rest in DRAM
+    assert bool((tier[:-1] >= 0).all())
