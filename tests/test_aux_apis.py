@@ -175,6 +175,32 @@ def test_quantize_embeddings_roundtrip():
     assert (v - v2).abs().max() < v.abs().max() / 100
 
 
+def test_quantize_embeddings_cli_fp8(tmp_path):
+    """The low-precision tool's fp8 path: quantized EV files written
+    alongside originals, reconstructable within e4m3 tolerance."""
+    import subprocess
+    import sys as _sys
+    from safetensors.torch import load_file, save_file
+    from deeprec_amd.ops.fp8 import dequantize_fp8_rows
+    d = tmp_path / "ckpt-1"
+    d.mkdir()
+    v = torch.randn(50, 8)
+    save_file({"keys": torch.arange(50), "values": v,
+               "freqs": torch.ones(50, dtype=torch.int64),
+               "versions": torch.zeros(50, dtype=torch.int64)},
+              str(d / "ev-emb-0.safetensors"))
+    r = subprocess.run(
+        [_sys.executable, "tools/quantize_embeddings.py", str(d),
+         "--apply", "--format", "fp8"],
+        capture_output=True, text=True, cwd=os.path.dirname(
+            os.path.dirname(os.path.abspath(__file__))))
+    assert r.returncode == 0, r.stderr
+    out = load_file(str(d / "ev-emb-0.fp8.safetensors"))
+    deq = dequantize_fp8_rows(out["values_fp8"], out["values_scale"])
+    assert (deq - v).abs().max() < v.abs().max() * 0.1
+    assert bool((out["keys"] == torch.arange(50)).all())
+
+
 def test_csv_dataset(tmp_path):
     from deeprec_amd.data.parquet import CsvDataset
     fn = os.path.join(tmp_path, "train.csv")

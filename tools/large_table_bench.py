@@ -27,7 +27,15 @@ import json
 import sys
 import time
 
+import os
+import sys
+
 import torch
+
+# runnable from anywhere: the repo root is the import root
+_REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+if _REPO_ROOT not in sys.path:
+    sys.path.insert(0, _REPO_ROOT)
 
 sys.path.insert(0, ".")
 

@@ -1,9 +1,11 @@
 """Post-training embedding quantization (reference capability:
 tools/low_precision_optimize): converts a checkpoint's EV value tensors to
-int8 with per-row scales, shrinking serving checkpoints ~4x; also verifies
-reconstruction error.
+int8 (per-row scales, ~4x smaller) or OCP fp8 e4m3 (--format fp8, ~4x
+smaller and directly consumable by the fp8 serving kernels, ops/fp8.py);
+also verifies reconstruction error.
 
 Usage: python tools/quantize_embeddings.py <ckpt_dir>/ckpt-N [--apply]
+       [--format int8|fp8]
 """
 import argparse
 import glob
@@ -11,6 +13,11 @@ import os
 import sys
 
 import torch
+
+# runnable from anywhere: the repo root is the import root
+_REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+if _REPO_ROOT not in sys.path:
+    sys.path.insert(0, _REPO_ROOT)
 from safetensors.torch import load_file, save_file
 
 
@@ -28,19 +35,26 @@ def main():
     p = argparse.ArgumentParser()
     p.add_argument("ckpt")
     p.add_argument("--apply", action="store_true",
-                   help="write <file>.int8.safetensors alongside originals")
+                   help="write <file>.<fmt>.safetensors alongside originals")
+    p.add_argument("--format", choices=["int8", "fp8"], default="int8")
     args = p.parse_args()
     files = sorted(glob.glob(os.path.join(args.ckpt, "ev-*.safetensors")))
     if not files:
         print("no EV files found", file=sys.stderr)
         sys.exit(1)
     for fn in files:
-        if fn.endswith(".int8.safetensors"):
+        if fn.endswith((".int8.safetensors", ".fp8.safetensors")):
             continue
         data = load_file(fn)
         v = data["values"]
-        q, scale = quantize_rows(v)
-        err = (dequantize_rows(q, scale) - v).abs().max()
+        if args.format == "fp8":
+            from deeprec_amd.ops.fp8 import (dequantize_fp8_rows,
+                                             quantize_fp8_rows)
+            q, scale = quantize_fp8_rows(v)
+            err = (dequantize_fp8_rows(q, scale) - v).abs().max()
+        else:
+            q, scale = quantize_rows(v)
+            err = (dequantize_rows(q, scale) - v).abs().max()
         orig_b = v.numel() * 4
         new_b = q.numel() + scale.numel() * 4
         print(f"{os.path.basename(fn)}: rows={v.shape[0]} dim={v.shape[1]} "
@@ -48,9 +62,10 @@ def main():
               f"max_abs_err={float(err):.5f}")
         if args.apply:
             out = {k: t for k, t in data.items() if k != "values"}
-            out["values_int8"] = q
+            out[f"values_{args.format}"] = q
             out["values_scale"] = scale
-            save_file(out, fn.replace(".safetensors", ".int8.safetensors"))
+            save_file(out, fn.replace(".safetensors",
+                                      f".{args.format}.safetensors"))
 
 
 if __name__ == "__main__":
