@@ -104,6 +104,21 @@ def _chunked_ev_backward(ev, grad_out, inverse, offsets, row_ids, counts,
         False, splits)
 
 
+def _drop_positions(sp_ids: RaggedIds, keep: torch.Tensor) -> RaggedIds:
+    """New RaggedIds with only the kept positions (offsets rebuilt)."""
+    row_ids = sp_ids.row_ids()[keep]
+    values = sp_ids.values[keep]
+    weights = sp_ids.weights[keep] if sp_ids.weights is not None else None
+    lengths = torch.zeros(sp_ids.batch_size, dtype=torch.int64,
+                          device=values.device)
+    lengths.index_add_(0, row_ids.long(),
+                       torch.ones_like(row_ids, dtype=torch.int64))
+    offsets = torch.zeros(sp_ids.batch_size + 1, dtype=sp_ids.offsets.dtype,
+                          device=values.device)
+    offsets[1:] = lengths.cumsum(0)
+    return RaggedIds(values, offsets, weights)
+
+
 def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,
                             combiner: str = "mean",
                             out_dtype=None,
@@ -114,6 +129,12 @@ def embedding_lookup_sparse(ev: EmbeddingVariable, sp_ids: RaggedIds,
     if isinstance(ev, ShardedEmbeddingVariable):
         return sharded_embedding_lookup_sparse(ev, sp_ids, combiner,
                                                out_dtype, train)
+    inv = getattr(ev, "invalid_key", None)
+    if inv is not None and bool((sp_ids.values == inv).any()):
+        # the EV's "no feature" sentinel: dropped before the lookup —
+        # never admitted, never trained, zeros in the pooled output
+        # (reference: invalid-key semantics of get_embedding_variable)
+        sp_ids = _drop_positions(sp_ids, sp_ids.values != inv)
     import os
     use_dedup = (train and ev.trainable
                  and hasattr(ev.storage, "dedup_lookup")
@@ -151,18 +172,7 @@ def safe_embedding_lookup_sparse(ev, sp_ids: RaggedIds, combiner="mean",
     pooled kernels' behavior) and negative ids are dropped.
     (reference: fused_safe_embedding_lookup_sparse, embedding_ops.py:1306)"""
     if bool((sp_ids.values < 0).any()):
-        keep = sp_ids.values >= 0
-        row_ids = sp_ids.row_ids()[keep]
-        values = sp_ids.values[keep]
-        weights = sp_ids.weights[keep] if sp_ids.weights is not None else None
-        lengths = torch.zeros(sp_ids.batch_size, dtype=torch.int64,
-                              device=values.device)
-        lengths.index_add_(0, row_ids.long(),
-                           torch.ones_like(row_ids, dtype=torch.int64))
-        offsets = torch.zeros(sp_ids.batch_size + 1, dtype=sp_ids.offsets.dtype,
-                              device=values.device)
-        offsets[1:] = lengths.cumsum(0)
-        sp_ids = RaggedIds(values, offsets, weights)
+        sp_ids = _drop_positions(sp_ids, sp_ids.values >= 0)
     return embedding_lookup_sparse(ev, sp_ids, combiner, out_dtype, train)
 
 

@@ -49,9 +49,15 @@ class EmbeddingVariable:
     def __init__(self, name: str, embedding_dim: int,
                  value_dtype=torch.float32,
                  ev_option: Optional[EmbeddingVariableOption] = None,
-                 device=None, generator=None, trainable: bool = True):
+                 device=None, generator=None, trainable: bool = True,
+                 invalid_key: Optional[int] = None):
         self.name = name
         self.dim = embedding_dim
+        # sentinel id for "no feature" (reference:
+        # tf.get_embedding_variable invalid-key per dtype,
+        # variable_scope.py:2146): lookups drop it (zeros in pooled
+        # output), it is never admitted and never trains
+        self.invalid_key = invalid_key
         self.value_dtype = value_dtype
         self.ev_option = ev_option or EmbeddingVariableOption()
         self.device = torch.device(device or "cpu")
@@ -176,7 +182,9 @@ def get_embedding_variable(name: str, embedding_dim: int,
                            initializer=None,
                            ev_option: Optional[EmbeddingVariableOption] = None,
                            device=None, trainable: bool = True,
-                           reuse: bool = True) -> EmbeddingVariable:
+                           reuse: bool = True,
+                           invalid_key: Optional[int] = None
+                           ) -> EmbeddingVariable:
     """Create-or-reuse an EV by name (reference: tf.get_embedding_variable,
     python/ops/variable_scope.py:2147)."""
     with _REGISTRY_LOCK:
@@ -193,7 +201,8 @@ def get_embedding_variable(name: str, embedding_dim: int,
         if initializer is not None:
             ev_option.init_option.initializer = initializer
         ev = EmbeddingVariable(name, embedding_dim, value_dtype, ev_option,
-                               device, trainable=trainable)
+                               device, trainable=trainable,
+                               invalid_key=invalid_key)
         _REGISTRY[name] = ev
         return ev
 

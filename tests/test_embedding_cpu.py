@@ -259,3 +259,29 @@ def test_lookup_tier_cpu():
     ev.lookup_or_create(torch.tensor([1, 2, 3]))
     tier = ev.lookup_tier(torch.tensor([1, 2, 3, 99]))
     assert tier.tolist() == [0, 0, 0, -1]
+
+
+def test_invalid_key_sentinel():
+    """get_embedding_variable(invalid_key=K): K is dropped from lookups
+    (zeros in pooled output), never admitted, never trained (reference:
+    the invalid-key sentinel of tf.get_embedding_variable,
+    variable_scope.py:2146)."""
+    import torch
+    from deeprec_amd.embedding import (RaggedIds, embedding_lookup_sparse,
+                                       get_embedding_variable)
+    ev = get_embedding_variable("invkey/ev", 4, invalid_key=-1)
+    ids = RaggedIds(torch.tensor([5, -1, 6, -1, -1]),
+                    torch.tensor([0, 2, 4, 5]))
+    out = embedding_lookup_sparse(ev, ids, combiner="sum")
+    ref = embedding_lookup_sparse(
+        ev, RaggedIds(torch.tensor([5, 6]), torch.tensor([0, 1, 2, 2])),
+        combiner="sum")
+    torch.testing.assert_close(out.detach(), ref.detach())
+    assert bool((out[2].detach() == 0).all())  # all-invalid row -> zeros
+    tier = ev.lookup_tier(torch.tensor([-1, 5, 6]))
+    assert tier.tolist() == [-1, 0, 0]  # sentinel never admitted
+    # and it never accumulates gradients
+    out.sum().backward()
+    grads = ev.consume_grads()
+    for _, keys, _ in grads:
+        assert bool((keys != -1).all())
