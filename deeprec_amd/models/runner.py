@@ -68,6 +68,26 @@ def build_argparser():
     p.add_argument("--timeline", type=int, default=None)
     p.add_argument("--log_steps", type=int, default=50)
     p.add_argument("--seed", type=int, default=42)
+    # reference train.py flags that are ALWAYS-ON here by architecture
+    # (accepted for drop-in CLI compatibility, modelzoo/dlrm/README.md):
+    # --ev            every sparse feature is an EmbeddingVariable
+    # --emb_fusion    the collection path fuses all tables' lookups
+    # --op_fusion     fusion lives in the HIP kernels, not graph passes
+    # --group_embedding  the collection IS a group lookup
+    for always_on in ("ev", "emb_fusion", "op_fusion", "group_embedding"):
+        p.add_argument(f"--{always_on}", action="store_true", default=True,
+                       help="accepted for reference-CLI parity; always on")
+    p.add_argument("--adaptive_emb", action="store_true",
+                   help="adaptive (hash+vocab) embedding — available via "
+                        "feature_column.adaptive_embedding_column; the zoo "
+                        "runner's collection path notes and ignores it")
+    p.add_argument("--dynamic_ev", action="store_true",
+                   help="dynamic-dimension EV — available via "
+                        "embedding.extras.DynamicEmbeddingVariable; noted "
+                        "and ignored by the zoo runner")
+    p.add_argument("--workqueue", default=None, metavar="GLOB",
+                   help="shard parquet files across workers through the "
+                        "checkpointable WorkQueue (reference: --workqueue)")
     return p
 
 
@@ -97,13 +117,14 @@ def make_ev_option(args) -> EmbeddingVariableOption:
     return opt
 
 
-def _parquet_batches(path, batch_size, device, num_sparse):
+def _parquet_batches(path, batch_size, device, num_sparse,
+                     num_epochs=1 << 30):
     """Adapt ParquetDataset rows to the (dense, ids, labels) batch shape."""
     from deeprec_amd.data.parquet import ParquetDataset
 
     class _Wrap:
         def __iter__(self):
-            ds = ParquetDataset(path, batch_size, num_epochs=1 << 30)
+            ds = ParquetDataset(path, batch_size, num_epochs=num_epochs)
             for cols in ds:
                 dense = torch.stack(
                     [cols[f"dense_{i}"].float() for i in range(13)],
@@ -152,7 +173,33 @@ def main(argv=None):
         broadcast_parameters(model.parameters())
         reducer = DenseGradAllreducer(model.parameters())
 
-    if args.parquet and not is_seq:
+    if args.adaptive_emb or args.dynamic_ev:
+        logging.getLogger("deeprec_amd").warning(
+            "--adaptive_emb/--dynamic_ev: the zoo runner's collection "
+            "path does not use them; see "
+            "feature_column.adaptive_embedding_column and "
+            "embedding.extras.DynamicEmbeddingVariable")
+    if args.workqueue and not is_seq:
+        import glob as _glob
+        from deeprec_amd.data.parquet import WorkQueue
+        files = sorted(_glob.glob(args.workqueue))
+        if not files:
+            raise SystemExit(f"--workqueue matched no files: "
+                             f"{args.workqueue}")
+        wq = WorkQueue(files, shuffle=True, seed=args.seed)
+
+        class _WqBatches:
+            def __iter__(self):
+                while True:
+                    fn = wq.take()
+                    if fn is None:
+                        return
+                    yield from _parquet_batches(fn, args.batch_size,
+                                                device, model.num_sparse,
+                                                num_epochs=1)
+
+        ds = _WqBatches()
+    elif args.parquet and not is_seq:
         from deeprec_amd.data.parquet import ParquetDataset
         ds = _parquet_batches(args.parquet, args.batch_size, device,
                               model.num_sparse)
@@ -213,7 +260,13 @@ def main(argv=None):
                 30 if args.incremental_ckpt else None),
             max_steps=args.steps) as sess:
         while not sess.should_stop():
-            sess.run(step_fn)
+            try:
+                sess.run(step_fn)
+            except StopIteration:
+                # finite source (e.g. --workqueue drained): clean stop
+                logging.getLogger("deeprec_amd").info(
+                    "input exhausted; stopping")
+                break
     dt = time.perf_counter() - t0
     if rank == 0:
         sps = args.steps * args.batch_size * world / dt

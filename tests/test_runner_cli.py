@@ -38,3 +38,33 @@ def test_runner_micro_batch_and_parquet(tmp_path):
          "--micro_batch", "2", "--parquet", fn],
         capture_output=True, text=True, timeout=240)
     assert r.returncode == 0 and "RESULT" in r.stdout, r.stderr[-500:]
+
+
+def test_runner_workqueue_and_parity_flags(tmp_path):
+    """--workqueue shards parquet files through the checkpointable
+    WorkQueue and stops cleanly when the queue drains; the reference
+    always-on flags (--ev --emb_fusion --op_fusion --group_embedding)
+    are accepted (modelzoo train.py CLI parity)."""
+    import subprocess
+    import sys
+
+    import numpy as np
+    import pandas as pd
+
+    for part in range(2):
+        df = pd.DataFrame(
+            {"label": np.random.randint(0, 2, 64).astype("float32")})
+        for i in range(13):
+            df[f"dense_{i}"] = np.random.randn(64).astype("float32")
+        for i in range(26):
+            df[f"sparse_{i}"] = np.random.randint(0, 50, 64)
+        df.to_parquet(str(tmp_path / f"part-{part}.parquet"))
+    r = subprocess.run(
+        [sys.executable, "-m", "deeprec_amd.models.runner", "--model",
+         "dlrm", "--steps", "100", "--batch_size", "32", "--no_bf16",
+         "--ev", "--emb_fusion", "--op_fusion", "--group_embedding",
+         "--adaptive_emb",
+         "--workqueue", str(tmp_path / "part-*.parquet")],
+        capture_output=True, text=True, timeout=240)
+    # 4 batches exist; --steps 100 must stop at input exhaustion, not die
+    assert r.returncode == 0 and "RESULT" in r.stdout, r.stderr[-800:]
