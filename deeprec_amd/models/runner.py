@@ -85,6 +85,12 @@ def build_argparser():
                    help="dynamic-dimension EV — available via "
                         "embedding.extras.DynamicEmbeddingVariable; noted "
                         "and ignored by the zoo runner")
+    p.add_argument("--protocol", default="rccl",
+                   choices=["rccl", "grpc", "grpc++", "star_server"],
+                   help="reference --protocol parity: every choice maps "
+                        "to the MI355X data planes (RCCL collectives "
+                        "single-node; the TCP PS plane via "
+                        "Estimator.run_cluster multi-node)")
     p.add_argument("--workqueue", default=None, metavar="GLOB",
                    help="shard parquet files across workers through the "
                         "checkpointable WorkQueue (reference: --workqueue)")

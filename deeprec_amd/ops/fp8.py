@@ -120,7 +120,10 @@ class Fp8Linear(nn.Module):
             elif self.act == "sigmoid":
                 y = torch.sigmoid(y)
             y = y.to(torch.bfloat16)
-        return y.reshape(*shape, self.out_features)
+        y = y.reshape(*shape, self.out_features)
+        # keep the caller's compute dtype (fp32 debug/serving models feed
+        # plain torch ops downstream)
+        return y if y.dtype == x.dtype else y.to(x.dtype)
 
     def extra_repr(self):
         return (f"in={self.in_features}, out={self.out_features}, "
