@@ -95,3 +95,74 @@ def test_cpu_engine_matches_dict_oracle(seed):
             assert sorted(k2.tolist()) == sorted(k.tolist())
 
     assert st.size() == len(oracle.rows)
+
+
+class FilteredOracle:
+    """Dict oracle for counter-filter admission: a key is admitted (row
+    allocated, default-initialized) in the SAME call where its
+    accumulated count reaches filter_freq; non-admitted lookups return
+    default_value_no_permission (0.0) and never train."""
+
+    def __init__(self, dim, default, filter_freq):
+        self.dim = dim
+        self.default = default
+        self.F = filter_freq
+        self.counts = {}
+        self.rows = {}
+
+    def lookup(self, keys, counts):
+        out = torch.zeros(len(keys), self.dim)
+        admitted = []
+        for i, (k, c) in enumerate(zip(keys, counts)):
+            self.counts[k] = self.counts.get(k, 0) + c
+            if self.counts[k] >= self.F:
+                if k not in self.rows:
+                    self.rows[k] = self.default[
+                        k % self.default.shape[0]].clone()
+                out[i] = self.rows[k]
+                admitted.append(i)
+        return out, admitted
+
+
+@pytest.mark.parametrize("seed", [0, 1, 2])
+def test_cpu_engine_counter_filter_matches_oracle(seed):
+    """Counter-filter admission under random lookup/train sequences
+    (reference semantics: CounterFilterPolicy, counter_filter_policy.h
+    — metadata accumulates pre-admission, values only post-admission)."""
+    from deeprec_amd.embedding.options import CounterFilter
+    from deeprec_amd.ops.sparse_optim_cpu import apply_sgd
+
+    rng = random.Random(100 + seed)
+    torch.manual_seed(seed)
+    dim, F = 5, 3
+    opt = EmbeddingVariableOption(
+        init_option=InitializerOption(default_value_dim=8),
+        filter_option=CounterFilter(filter_freq=F))
+    st = CpuStorage(dim, opt)
+    oracle = FilteredOracle(dim, st.default_values, F)
+
+    for step in range(50):
+        ks = sorted(set(rng.randrange(0, 40)
+                        for _ in range(rng.randrange(1, 7))))
+        cs = [rng.randrange(1, 3) for _ in ks]
+        kt = torch.tensor(ks, dtype=torch.int64)
+        ct = torch.tensor(cs, dtype=torch.int64)
+        slots = st.lookup_or_create(kt, ct, step=step)
+        got = st.gather(kt, slots)
+        want, admitted = oracle.lookup(ks, cs)
+        torch.testing.assert_close(got, want, rtol=1e-5, atol=1e-6)
+        # engine and oracle agree on WHO is admitted
+        assert (slots >= 0).nonzero().squeeze(1).tolist() == admitted
+
+        if rng.random() < 0.7 and admitted:
+            grads = torch.randn(kt.numel(), dim)
+            apply_sgd(st, slots, grads, lr=0.1)
+            for i in admitted:
+                oracle.rows[ks[i]] -= 0.1 * grads[i]
+
+    # sub-threshold keys were counted but never materialized
+    assert st.size() == len(oracle.rows)
+    every = torch.arange(40, dtype=torch.int64)
+    freqs = st.frequencies(every)
+    for k in range(40):
+        assert int(freqs[k]) == oracle.counts.get(k, 0), k
