@@ -2,7 +2,6 @@
 (hypothesis; the reference validates the same surfaces with hand-picked
 cases — random op sequences catch the interleavings hand-written tests
 miss)."""
-import numpy as np
 import pytest
 import torch
 
