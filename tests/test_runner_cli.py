@@ -68,3 +68,21 @@ def test_runner_workqueue_and_parity_flags(tmp_path):
         capture_output=True, text=True, timeout=240)
     # 4 batches exist; --steps 100 must stop at input exhaustion, not die
     assert r.returncode == 0 and "RESULT" in r.stdout, r.stderr[-800:]
+
+
+def test_runner_distributed_gloo_two_ranks():
+    """The runner's world>1 path (init_distributed, sharded collection,
+    dense reducer, rank-0 RESULT) through a real 2-process torchrun
+    launch on gloo — the CPU analog of the driver's multi-GPU launch."""
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29573", "-m", "deeprec_amd.models.runner",
+         "--model", "dlrm", "--steps", "3", "--batch_size", "64",
+         "--no_bf16", "--sharded", "--log_steps", "1"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, (r.stdout[-400:], r.stderr[-800:])
+    assert "RESULT" in r.stdout and "world=2" in r.stdout
