@@ -49,3 +49,22 @@ def test_bench_two_rank_gloo_contract():
     assert d["config"]["global_batch"] == 256  # whole node, not per rank
     assert d["config"]["parallelism"] == "dp2+ep2"
     assert d["value"] > 0
+
+
+def test_bench_force_dist_single_process():
+    """--force-dist at world=1 on CPU: the sharded+collective code path
+    (gloo self-exchange) behind the same JSON contract."""
+    r = subprocess.run(
+        [sys.executable, "bench.py", "--force-dist", "--steps", "2",
+         "--warmup", "1", "--batch", "128"],
+        capture_output=True, text=True, timeout=300)
+    assert r.returncode == 0, r.stderr[-800:]
+    d = _last_json(r.stdout)
+    assert d["config"]["parallelism"] == "dp1+ep1"
+    assert d["value"] > 0
+
+
+def test_graft_entry_surface():
+    import __graft_entry__
+    assert callable(__graft_entry__.build)
+    assert callable(__graft_entry__.smoke)
