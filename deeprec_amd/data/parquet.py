@@ -127,17 +127,31 @@ class WorkQueue:
                 item = self._items[self._cursor]
                 self._cursor += 1
                 return item
-        # rank 0 assigns: every rank calls take() collectively
-        idx = torch.tensor([0], dtype=torch.int64)
-        if dist.get_rank() == 0:
-            with self._lock:
-                idx[0] = self._cursor
-                self._cursor += dist.get_world_size()
-        dist.broadcast(idx, src=0)
-        my = int(idx[0]) + dist.get_rank()
-        if dist.get_rank() != 0:
-            self._cursor = int(idx[0]) + dist.get_world_size()
-        return self._items[my] if my < len(self._items) else None
+        # rank 0 assigns: every rank calls take() collectively. A rank
+        # may only STOP calling take() when a round starts past the end
+        # of the list (every rank sees the same idx, so all agree the
+        # queue is drained in the same round). In a TAIL round — idx
+        # still in range but this rank's slot past the end — the rank
+        # must keep participating in broadcast rounds, or the peers that
+        # DID draw items would block forever on their next take().
+        while True:
+            idx = torch.tensor([0], dtype=torch.int64)
+            if dist.get_rank() == 0:
+                with self._lock:
+                    idx[0] = self._cursor
+                    if self._cursor < len(self._items):
+                        self._cursor += dist.get_world_size()
+            dist.broadcast(idx, src=0)
+            start = int(idx[0])
+            if start >= len(self._items):
+                return None  # unanimous: drained
+            if dist.get_rank() != 0:
+                self._cursor = min(start + dist.get_world_size(),
+                                   len(self._items))
+            my = start + dist.get_rank()
+            if my < len(self._items):
+                return self._items[my]
+            # tail round: no item for this rank; rejoin the next round
 
     def remaining(self) -> List[str]:
         return self._items[self._cursor:]

@@ -270,6 +270,31 @@ def test_work_queue_distributed():
     _run_dist(_body_work_queue_dist, world_size=2, port=29542)
 
 
+def _body_work_queue_tail_drain(rank, world):
+    """Odd item count: one rank draws None in the TAIL round while the
+    peer still drew an item. The drained rank must keep participating
+    in assignment rounds until ALL ranks see the end — the naive
+    return-None-and-stop protocol deadlocked the peer's next take()."""
+    from deeprec_amd.data.parquet import WorkQueue
+    wq = WorkQueue([f"f{i}" for i in range(3)])
+    got = []
+    while True:
+        item = wq.take()
+        if item is None:
+            break
+        got.append(item)
+    gathered = [None, None]
+    dist.all_gather_object(gathered, got)
+    assert sorted(gathered[0] + gathered[1]) == ["f0", "f1", "f2"]
+    # both ranks reached None (this line executing on both IS the
+    # no-deadlock proof); a further collective round still agrees
+    assert wq.take() is None
+
+
+def test_work_queue_tail_drain():
+    _run_dist(_body_work_queue_tail_drain, world_size=2, port=29547)
+
+
 def _body_bf16_exchange(rank, world):
     from deeprec_amd.parallel.sharded_collection import (
         ShardedEmbeddingCollection)
