@@ -94,6 +94,10 @@ def build_argparser():
     p.add_argument("--workqueue", default=None, metavar="GLOB",
                    help="shard parquet files across workers through the "
                         "checkpointable WorkQueue (reference: --workqueue)")
+    p.add_argument("--eval_steps", type=int, default=0,
+                   help="after training, evaluate N batches (no inserts, "
+                        "no training) and print loss/accuracy/AUC — the "
+                        "reference train.py eval loop")
     return p
 
 
@@ -279,6 +283,36 @@ def main(argv=None):
         print(f"RESULT model={args.model} steps={args.steps} "
               f"batch={args.batch_size} world={world} "
               f"samples_per_sec={sps:.1f}")
+
+    if args.eval_steps > 0:
+        # reference train.py eval loop: no inserts, no training; report
+        # loss / accuracy / streaming AUC
+        from deeprec_amd.training.metrics import (StreamingAccuracy,
+                                                  StreamingAUC)
+        auc, acc = StreamingAUC(), StreamingAccuracy()
+        losses = []
+        with torch.no_grad():
+            for _ in range(args.eval_steps):
+                if is_seq:
+                    dense, ids, seq, target, labels = ds.next_seq_batch()
+                    logits = model(dense, ids[:, :model.num_sparse], seq,
+                                   target, train=False)
+                else:
+                    try:
+                        dense, ids, labels = next(it)
+                    except StopIteration:
+                        break  # finite source drained during training
+                    logits = model(dense, ids, train=False)
+                if isinstance(logits, (list, tuple)):
+                    logits = logits[0]
+                losses.append(float(model.loss_fn(logits, labels)))
+                probs = torch.sigmoid(logits.float())
+                auc.update(probs, labels)
+                acc.update(probs, labels)
+        if rank == 0:
+            print(f"EVAL model={args.model} steps={args.eval_steps} "
+                  f"loss={sum(losses) / len(losses):.4f} "
+                  f"accuracy={acc.result():.4f} auc={auc.result():.4f}")
     if world > 1:
         import torch.distributed as dist
         dist.destroy_process_group()

@@ -86,3 +86,14 @@ def test_runner_distributed_gloo_two_ranks():
         capture_output=True, text=True, timeout=300)
     assert r.returncode == 0, (r.stdout[-400:], r.stderr[-800:])
     assert "RESULT" in r.stdout and "world=2" in r.stdout
+
+
+def test_runner_eval_loop(capsys):
+    """--eval_steps: post-training eval (no inserts) printing
+    loss/accuracy/AUC — the reference train.py eval loop."""
+    main(["--model", "wdl", "--steps", "3", "--batch_size", "64",
+          "--no_bf16", "--eval_steps", "3"])
+    out = capsys.readouterr().out
+    assert "EVAL model=wdl" in out and "auc=" in out
+    auc = float(out.split("auc=")[1].split()[0])
+    assert 0.0 <= auc <= 1.0
