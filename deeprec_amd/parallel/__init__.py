@@ -19,6 +19,8 @@ from deeprec_amd.parallel import comm  # noqa: F401
 from deeprec_amd.parallel.comm import is_initialized, rank, world_size  # noqa: F401
 from deeprec_amd.parallel.sharded_embedding import (  # noqa: F401
     ShardedEmbeddingVariable, sharded_embedding_lookup_sparse)
+from deeprec_amd.parallel.sharded_collection import (  # noqa: F401
+    ShardedEmbeddingCollection)
 
 
 def init_distributed(backend: str = None, device=None):
