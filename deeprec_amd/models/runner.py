@@ -303,9 +303,11 @@ def main(argv=None):
                     except StopIteration:
                         break  # finite source drained during training
                     logits = model(dense, ids, train=False)
+                # loss on the FULL head structure (multi-task models take
+                # the logits list); AUC/accuracy on the primary head
+                losses.append(float(model.loss_fn(logits, labels)))
                 if isinstance(logits, (list, tuple)):
                     logits = logits[0]
-                losses.append(float(model.loss_fn(logits, labels)))
                 probs = torch.sigmoid(logits.float())
                 auc.update(probs, labels)
                 acc.update(probs, labels)

@@ -97,3 +97,12 @@ def test_runner_eval_loop(capsys):
     assert "EVAL model=wdl" in out and "auc=" in out
     auc = float(out.split("auc=")[1].split()[0])
     assert 0.0 <= auc <= 1.0
+
+
+def test_runner_eval_multitask(capsys):
+    """Multi-task models evaluate with the full head structure passed
+    to loss_fn (AUC/accuracy on the primary head)."""
+    main(["--model", "esmm", "--steps", "2", "--batch_size", "64",
+          "--no_bf16", "--eval_steps", "2"])
+    out = capsys.readouterr().out
+    assert "EVAL model=esmm" in out and "auc=" in out
