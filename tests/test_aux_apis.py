@@ -603,3 +603,39 @@ def test_batch_caches_lru_lfu():
     assert lfu.size() == 2
     assert isinstance(make_cache(CacheStrategy.LRU), LRUCache)
     assert isinstance(make_cache(CacheStrategy.LFU), LFUCache)
+
+
+def test_quantized_checkpoint_serves(tmp_path):
+    """Quantize an EV checkpoint file offline, load the shrunken
+    artifact back and restore it into a fresh EV — gathers match within
+    quantization tolerance for both formats (completes the reference's
+    low-precision-optimize pipeline: the artifact is servable, not just
+    smaller)."""
+    import subprocess
+    import sys as _sys
+    from safetensors.torch import save_file
+    from deeprec_amd.embedding import EmbeddingVariable
+    from tools.quantize_embeddings import load_quantized
+
+    d = tmp_path / "ckpt-1"
+    d.mkdir()
+    torch.manual_seed(0)
+    keys = torch.arange(40, dtype=torch.int64) * 7
+    vals = torch.randn(40, 8)
+    save_file({"keys": keys, "values": vals,
+               "freqs": torch.ones(40, dtype=torch.int64),
+               "versions": torch.zeros(40, dtype=torch.int64)},
+              str(d / "ev-t-0.safetensors"))
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for fmt, tol in (("int8", 0.02), ("fp8", 0.1)):
+        r = subprocess.run(
+            [_sys.executable, "tools/quantize_embeddings.py", str(d),
+             "--apply", "--format", fmt],
+            capture_output=True, text=True, cwd=root)
+        assert r.returncode == 0, r.stderr
+        k, v, f, ver = load_quantized(
+            str(d / f"ev-t-0.{fmt}.safetensors"))
+        ev = EmbeddingVariable(f"quantserve/{fmt}", 8)
+        ev.restore(k, v, f, ver)
+        got = ev.gather(keys)
+        assert (got - vals).abs().max() < vals.abs().max() * tol, fmt

@@ -70,3 +70,23 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def load_quantized(fn: str):
+    """Load a `.int8.safetensors` / `.fp8.safetensors` EV file back into
+    servable tensors: (keys, values fp32 dequantized, freqs, versions).
+    Feed straight into `EmbeddingVariable.restore(...)` (or a remote KV
+    publish) — completes the low-precision pipeline: quantize offline,
+    serve the shrunken artifact."""
+    from safetensors.torch import load_file
+    data = load_file(fn)
+    scale = data["values_scale"]
+    if "values_int8" in data:
+        values = dequantize_rows(data["values_int8"], scale)
+    elif "values_fp8" in data:
+        from deeprec_amd.ops.fp8 import dequantize_fp8_rows
+        values = dequantize_fp8_rows(data["values_fp8"], scale)
+    else:
+        raise ValueError(f"{fn}: no quantized values tensor")
+    return (data["keys"], values, data.get("freqs"),
+            data.get("versions"))
