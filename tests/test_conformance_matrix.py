@@ -163,3 +163,36 @@ def test_matrix_repartition_cpu(worlds, tmp_path):
              tmp_path=tmp_path)
     run_cell("adam", "collection", "counter", "cpu",
              saver_worlds=worlds, tmp_path=tmp_path)
+
+
+@pytest.mark.parametrize("opt_name", ["adagrad", "adam_async", "ftrl"])
+def test_invalid_key_checkpoint_cell(opt_name, tmp_path):
+    """Matrix extension: invalid_key EVs train/checkpoint/restore like
+    plain EVs and the sentinel never appears in any checkpoint."""
+    from deeprec_amd.checkpoint.saver import Saver
+    from deeprec_amd.embedding import RaggedIds, embedding_lookup_sparse
+    from deeprec_amd.embedding.variable import get_embedding_variable
+
+    ev = get_embedding_variable(f"cm_inv/{opt_name}", DIM, invalid_key=-1)
+    opt = OPTIMIZERS[opt_name]([ev])
+    g = torch.Generator().manual_seed(3)
+    for step in range(3):
+        raw = torch.randint(0, 30, (12,), generator=g)
+        raw[::4] = -1  # sentinel sprinkled through the batch
+        ids = RaggedIds(raw, torch.arange(0, 13, 2))
+        out = embedding_lookup_sparse(ev, ids, combiner="sum")
+        out.sum().backward()
+        opt.step()
+    saver = Saver(embedding_variables=[ev], optimizer=opt)
+    path = saver.save(str(tmp_path), 3)
+    keys, vals, _, _ = ev.export()
+    assert bool((keys != -1).all())  # sentinel never admitted
+
+    ev2 = get_embedding_variable(f"cm_inv2/{opt_name}", DIM,
+                                 invalid_key=-1)
+    ev2.name = ev.name
+    opt2 = OPTIMIZERS[opt_name]([ev2])
+    Saver(embedding_variables=[ev2], optimizer=opt2).restore(path)
+    k2, v2, _, _ = ev2.export()
+    o1, o2 = torch.argsort(keys), torch.argsort(k2)
+    torch.testing.assert_close(vals[o1], v2[o2])
