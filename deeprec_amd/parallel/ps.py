@@ -69,6 +69,9 @@ class _PsHandler(socketserver.BaseRequestHandler):
                 elif op == "SAVE":
                     path = srv.ps.save(req["dir"], req["step"])
                     _send_frame(self.request, {"ok": True, "path": path})
+                elif op == "INCRSAVE":
+                    path = srv.ps.incremental_save(req["dir"], req["step"])
+                    _send_frame(self.request, {"ok": True, "path": path})
                 elif op == "STAT":
                     _send_frame(self.request, {"ok": True,
                                                "stat": srv.ps.stat()})
@@ -157,6 +160,13 @@ class PsServer:
     def save(self, directory: str, step: int) -> str:
         with self._lock:
             return self.saver.save(directory, step)
+
+    def incremental_save(self, directory: str, step: int) -> str:
+        """Delta checkpoint of keys touched since the last save
+        (reference failover contract: restore = last full + ordered
+        incremental replay, Incremental-Checkpoint.md)."""
+        with self._lock:
+            return self.saver.incremental_save(directory, step)
 
     def stat(self):
         return {"tables": {n: ev.size() for n, ev in self.evs.items()},

@@ -241,3 +241,35 @@ def test_estimator_run_cluster_worker(tmp_path):
         assert not torch.allclose(rows, torch.full_like(rows, 0.5))
     finally:
         server.close()
+
+
+def test_ps_incremental_failover(tmp_path):
+    """Async-PS failover with DELTAS (reference contract: restore = last
+    full + ordered incremental replay): full SAVE, more training,
+    INCRSAVE only; a fresh PS on the same port restores full+incr and
+    serves the post-delta rows."""
+    s0 = PsServer({"emb": 4}, ps_index=0)
+    port0 = s0.port
+    client = PsClient([("127.0.0.1", port0)])
+    emb = PsShardedEmbedding(client, "emb", 4, async_push=False)
+    keys = torch.arange(8, dtype=torch.int64)
+    rows = emb.lookup(keys)
+    (rows ** 2).sum().backward()
+    client.call(0, {"op": "SAVE", "dir": str(tmp_path), "step": 1})
+    # post-full-save training lands ONLY in the incremental delta
+    rows = emb.lookup(keys)
+    (rows ** 2).sum().backward()
+    trained = emb.lookup(keys, train=False).clone()
+    resp = client.call(0, {"op": "INCRSAVE", "dir": str(tmp_path),
+                           "step": 2})
+    assert resp["path"].endswith("ckpt-2.incr")
+    s0.close()
+
+    s1 = PsServer({"emb": 4}, ps_index=0, port=port0,
+                  checkpoint_dir=str(tmp_path))
+    try:
+        got = emb.lookup(keys, train=False)
+        torch.testing.assert_close(got, trained)  # delta replayed
+    finally:
+        s1.close()
+        client.close()
