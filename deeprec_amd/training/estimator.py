@@ -162,6 +162,14 @@ class Estimator:
         est.train(input_fn, steps=steps)
         for emb in p["embeddings"].values():
             emb.flush()
+        if model_dir and int(cfg.get("index", 0)) == 0:
+            # the chief checkpoints every PS shard at the end of its run
+            # — the failover artifact a restarted PS restores from
+            # (reference: async-PS failover, Incremental-Checkpoint.md)
+            client = next(iter(p["embeddings"].values())).client
+            for i in range(client.world):
+                client.call(i, {"op": "SAVE", "dir": model_dir,
+                                "step": steps})
         return est
 
     @torch.no_grad()

@@ -232,8 +232,12 @@ def test_estimator_run_cluster_worker(tmp_path):
 
         est = Estimator.run_cluster(
             model_fn, input_fn, tables={"emb": 8}, tf_config=cfg,
-            steps=5, config=RunConfig(log_step_count_steps=1000))
+            steps=5, model_dir=str(tmp_path),
+            config=RunConfig(log_step_count_steps=1000))
         assert est is not None
+        # the chief checkpointed the PS shard (failover artifact)
+        import glob as _glob
+        assert _glob.glob(str(tmp_path / "ckpt-5" / "ev-*"))
         st = server.stat()
         assert st["applied"] >= 5 and st["tables"]["emb"] > 0
         # PS rows actually trained (moved off the 0.5 init)
