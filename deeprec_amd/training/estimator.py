@@ -159,6 +159,12 @@ class Estimator:
         p["embeddings"] = worker_embeddings(cfg, tables)
         p["cluster"] = cfg
         est = cls(model_fn, model_dir=model_dir, config=config, params=p)
+        if cfg["type"] == "evaluator":
+            # evaluator task: read-only pulls (no inserts, no pushes),
+            # reports loss/accuracy/AUC against the live PS shards
+            metrics = est.evaluate(input_fn, steps=steps)
+            est.eval_metrics = metrics
+            return est
         est.train(input_fn, steps=steps)
         for emb in p["embeddings"].values():
             emb.flush()

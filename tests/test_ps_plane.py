@@ -277,3 +277,57 @@ def test_ps_incremental_failover(tmp_path):
     finally:
         s1.close()
         client.close()
+
+
+def test_estimator_run_cluster_evaluator():
+    """Evaluator role: read-only metrics against live PS shards (no
+    inserts, no pushes — the TF cluster evaluator task)."""
+    from deeprec_amd.optimizers import GradientDescentOptimizer
+    from deeprec_amd.training.estimator import Estimator, RunConfig
+
+    server = PsServer({"emb": 8}, ps_index=0)
+    try:
+        class M(torch.nn.Module):
+            def __init__(self, emb):
+                super().__init__()
+                self.emb = emb
+                self.dense = torch.nn.Linear(8, 1)
+
+            def embedding_variables(self):
+                return []
+
+            def forward(self, ids, train=True):
+                return self.dense(
+                    self.emb.lookup(ids, train=train).float()).squeeze(1)
+
+            def loss_fn(self, logits, labels):
+                return torch.nn.functional.\
+                    binary_cross_entropy_with_logits(logits, labels)
+
+        def model_fn(params):
+            m = M(params["embeddings"]["emb"])
+            return m, GradientDescentOptimizer(params=m.parameters(),
+                                               learning_rate=0.05)
+
+        def input_fn():
+            g = torch.Generator().manual_seed(1)
+            while True:
+                yield (torch.randint(0, 40, (16,), generator=g),
+                       torch.rand(16, generator=g).round())
+
+        cfg = {"cluster": {"ps": [f"127.0.0.1:{server.port}"],
+                           "worker": []},
+               "task": {"type": "evaluator", "index": 0}}
+        before = server.stat()["tables"]["emb"]
+        est = Estimator.run_cluster(model_fn, input_fn, tables={"emb": 8},
+                                    tf_config=cfg, steps=4,
+                                    config=RunConfig(
+                                        log_step_count_steps=1000))
+        m = est.eval_metrics
+        assert {"loss", "accuracy", "auc"} <= set(m)
+        assert 0.0 <= m["auc"] <= 1.0
+        # read-only: the evaluator admitted nothing and applied nothing
+        assert server.stat()["tables"]["emb"] == before
+        assert server.stat()["applied"] == 0
+    finally:
+        server.close()
