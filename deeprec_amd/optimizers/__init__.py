@@ -47,6 +47,18 @@ class Optimizer:
     def _sparse_hyper(self, ev) -> dict:
         raise NotImplementedError
 
+    def set_learning_rate(self, lr: float):
+        """Update the learning rate in place (LR schedules — reference:
+        tf.train.exponential_decay fed into optimizer constructors).
+        Takes effect on the NEXT eager step for both sparse applies
+        (hyper is rebuilt per step) and the dense optimizer (param
+        groups updated). A hipGraph-captured step bakes the lr it was
+        captured with — re-capture (or run eager) to change it."""
+        self.lr = lr
+        if self._dense is not None:
+            for g in self._dense.param_groups:
+                g["lr"] = lr
+
     def zero_grad(self, set_to_none: bool = True):
         if self._dense is not None:
             self._dense.zero_grad(set_to_none=set_to_none)

@@ -639,3 +639,48 @@ def test_quantized_checkpoint_serves(tmp_path):
         ev.restore(k, v, f, ver)
         got = ev.gather(keys)
         assert (got - vals).abs().max() < vals.abs().max() * tol, fmt
+
+
+def test_lr_schedules():
+    """exponential/polynomial/piecewise schedules (reference:
+    tf.train.exponential_decay family) + the hook driving an optimizer
+    through MonitoredTrainingSession."""
+    from deeprec_amd.training.schedules import (
+        LearningRateScheduleHook, exponential_decay, piecewise_constant,
+        polynomial_decay)
+    exp = exponential_decay(0.1, decay_steps=10, decay_rate=0.5)
+    assert abs(exp(0) - 0.1) < 1e-9 and abs(exp(10) - 0.05) < 1e-9
+    stair = exponential_decay(0.1, 10, 0.5, staircase=True)
+    assert abs(stair(9) - 0.1) < 1e-9 and abs(stair(10) - 0.05) < 1e-9
+    poly = polynomial_decay(0.1, 100, end_learning_rate=0.01)
+    assert abs(poly(0) - 0.1) < 1e-9 and abs(poly(100) - 0.01) < 1e-9
+    pw = piecewise_constant([5, 10], [0.1, 0.01, 0.001])
+    assert pw(0) == 0.1 and pw(7) == 0.01 and pw(50) == 0.001
+
+    from deeprec_amd.embedding import (EmbeddingVariable, RaggedIds,
+                                       embedding_lookup_sparse)
+    from deeprec_amd.optimizers import AdagradOptimizer
+    from deeprec_amd.training.session import MonitoredTrainingSession
+    ev = EmbeddingVariable("lrsched/ev", 4)
+    dense = torch.nn.Linear(4, 1)
+    opt = AdagradOptimizer(params=dense.parameters(),
+                           embedding_variables=[ev], learning_rate=0.1)
+    hook = LearningRateScheduleHook(opt, piecewise_constant([2], [0.1,
+                                                                  0.001]))
+    seen = []
+
+    def step_fn():
+        ids = RaggedIds(torch.randint(0, 20, (8,)),
+                        torch.arange(0, 9, 2))
+        out = dense(embedding_lookup_sparse(ev, ids, combiner="sum"))
+        loss = out.sum()
+        opt.zero_grad()
+        loss.backward()
+        seen.append(opt.lr)
+        opt.step()
+        return {"loss": loss.detach()}
+
+    with MonitoredTrainingSession(hooks=[hook], max_steps=4) as sess:
+        while not sess.should_stop():
+            sess.run(step_fn)
+    assert seen[0] == 0.1 and seen[-1] == 0.001  # schedule applied
