@@ -28,10 +28,14 @@ class Optimizer:
 
     def __init__(self, params: Iterable = None,
                  embedding_variables: Optional[List[EmbeddingVariable]] = None,
-                 learning_rate: float = 0.01):
+                 learning_rate: float = 0.01, clip_norm: float = None):
         self.lr = learning_rate
+        # per-gradient norm clip (reference: the DIN/DIEN train.py
+        # tf.clip_by_norm(grad, 5) wrap around compute_gradients)
+        self.clip_norm = clip_norm
         self.evs = list(embedding_variables or [])
         params = [p for p in (params or []) if p.requires_grad]
+        self._params = params
         self._dense = self._make_dense(params) if params else None
         # distributed seam: set to DenseGradAllreducer.wait to overlap an
         # async dense-grad all-reduce with the sparse applies
@@ -78,8 +82,26 @@ class Optimizer:
             getattr(cpu_apply, f"apply_{self.sparse_name}")(
                 ev.storage, slots, grad, **hyper)
 
+    def _clip_grads(self):
+        """tf.clip_by_norm semantics: each gradient tensor (and each
+        sparse grad-rows tensor) independently rescaled to norm <=
+        clip_norm."""
+        cn = self.clip_norm
+        for p in self._params:
+            if p.grad is not None:
+                n = p.grad.norm()
+                if n > cn:
+                    p.grad.mul_(cn / (n + 1e-12))
+        for ev in self.evs:
+            for _, _, grad in ev._pending_grads:
+                n = grad.norm()
+                if n > cn:
+                    grad.mul_(cn / (n + 1e-12))
+
     def step(self, increment_global_step: bool = True):
         self._step_count += 1
+        if self.clip_norm is not None:
+            self._clip_grads()
         # sparse applies first: they are independent of the dense grads,
         # so an in-flight async all-reduce overlaps with them
         for ev in self.evs:

@@ -147,3 +147,36 @@ def test_ftrl_v2_l2_shrinkage():
     v1 = run(0.5)
     assert not torch.allclose(v0, v1), \
         "l2_shrinkage must change the trajectory"
+
+
+def test_gradient_clipping_by_norm():
+    """clip_norm: per-gradient tf.clip_by_norm semantics on both the
+    dense params and the sparse EV grad rows (reference: DIN/DIEN
+    train.py wraps compute_gradients with clip_by_norm(grad, 5))."""
+    import torch
+    from deeprec_amd.embedding import (EmbeddingVariable, RaggedIds,
+                                       embedding_lookup_sparse)
+    from deeprec_amd.optimizers import GradientDescentOptimizer
+
+    ev = EmbeddingVariable("clip/ev", 4)
+    dense = torch.nn.Linear(4, 1, bias=False)
+    opt = GradientDescentOptimizer(params=dense.parameters(),
+                                   embedding_variables=[ev],
+                                   learning_rate=1.0, clip_norm=0.5)
+    ids = RaggedIds(torch.tensor([1, 2, 3, 4]), torch.arange(0, 5))
+    before = ev.gather(torch.tensor([1, 2, 3, 4])).clone()
+    w_before = dense.weight.detach().clone()
+    out = dense(embedding_lookup_sparse(ev, ids, combiner="sum"))
+    (out.sum() * 100).backward()  # huge grads
+    opt.step()
+    # dense update bounded by lr * clip_norm
+    assert float((dense.weight.detach() - w_before).norm()) <= 0.5 + 1e-5
+    # each sparse row moved, but the whole grad tensor norm was clipped
+    after = ev.gather(torch.tensor([1, 2, 3, 4]))
+    moved = (after - before).norm()
+    assert 0 < float(moved) <= 0.5 + 1e-5
+    # attribute form works for subclasses with custom __init__
+    from deeprec_amd.optimizers import AdagradOptimizer
+    o2 = AdagradOptimizer(embedding_variables=[ev], learning_rate=0.1)
+    o2.clip_norm = 5.0
+    assert o2.clip_norm == 5.0
