@@ -74,3 +74,32 @@ class PrefetchIterator:
                 self.q.get_nowait()
         except queue.Empty:
             pass
+
+
+class ShuffleBuffer:
+    """Streaming shuffle with a bounded buffer (reference:
+    ``dataset.shuffle(buffer_size=20000)`` in the zoo input pipelines,
+    modelzoo/dlrm/train.py:324): fills ``buffer_size`` items, then each
+    pull swaps a uniformly random buffered item with the next source
+    item. Every source item is yielded exactly once; composes under
+    PrefetchIterator (shuffle first, then prefetch)."""
+
+    def __init__(self, source, buffer_size: int, seed: int = 0):
+        self.source = source
+        self.buffer_size = int(buffer_size)
+        self.seed = seed
+
+    def __iter__(self):
+        import random
+        rng = random.Random(self.seed)
+        buf = []
+        it = iter(self.source)
+        for item in it:
+            if len(buf) < self.buffer_size:
+                buf.append(item)
+                continue
+            j = rng.randrange(len(buf))
+            out, buf[j] = buf[j], item
+            yield out
+        rng.shuffle(buf)
+        yield from buf

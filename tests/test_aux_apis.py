@@ -684,3 +684,18 @@ def test_lr_schedules():
         while not sess.should_stop():
             sess.run(step_fn)
     assert seen[0] == 0.1 and seen[-1] == 0.001  # schedule applied
+
+
+def test_shuffle_buffer():
+    """Streaming shuffle: exactly-once delivery, seed-deterministic,
+    actually permutes, composes with PrefetchIterator."""
+    from deeprec_amd.data.prefetch import PrefetchIterator, ShuffleBuffer
+    items = list(range(100))
+    out1 = list(ShuffleBuffer(items, buffer_size=10, seed=3))
+    out2 = list(ShuffleBuffer(items, buffer_size=10, seed=3))
+    out3 = list(ShuffleBuffer(items, buffer_size=10, seed=4))
+    assert sorted(out1) == items and out1 == out2
+    assert out1 != items and out1 != out3
+    piped = list(PrefetchIterator(
+        ShuffleBuffer(items, buffer_size=10, seed=3), depth=2))
+    assert piped == out1
