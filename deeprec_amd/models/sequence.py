@@ -114,9 +114,20 @@ class DIEN(_SeqBase):
 
     def __init__(self, embedding_dim=16, item_dim=32, gru_hidden=32,
                  att_hidden=64, mlp_sizes=(256, 128, 64), device="cpu",
-                 bf16=True, **kw):
+                 bf16=True, use_aux_loss=True, aux_alpha=1.0, **kw):
         super().__init__(embedding_dim, item_dim, device, bf16, name="dien",
                          **kw)
+        # auxiliary next-item supervision on the GRU hidden states
+        # (reference: _auxiliary_loss, modelzoo/dien/train.py:231-251:
+        # click pair [h_t; e_{t+1}] vs negative pair, masked log-loss
+        # added to the main objective). Negatives = within-batch
+        # permutation of the real next items (no fabricated ids).
+        self.use_aux_loss = use_aux_loss
+        self.aux_alpha = aux_alpha
+        self._aux_loss = None
+        self.aux_net = nn.Sequential(
+            nn.Linear(gru_hidden + item_dim, 64), nn.Sigmoid(),
+            nn.Linear(64, 1))
         from deeprec_amd.ops.fused_gru import FusedGRU
         # fused single-kernel recurrences (MIOpen's RNN path ran this at
         # 44 ms/step; the python AUGRU loop added ~600 launches)
@@ -146,6 +157,8 @@ class DIEN(_SeqBase):
                                   train=train).float()   # [B,Di]
         mask = (seq_ids > 0).float()
         h_seq = self.gru(seq)                            # [B,T,H]
+        if train and self.use_aux_loss and seq.shape[1] > 1:
+            self._aux_loss = self._auxiliary_loss(h_seq, seq, mask)
         tgt_h = self.target_proj(target)                 # [B,H]
         cd = self.compute_dtype if self.bf16 else h_seq.dtype
         att_in = torch.cat(
@@ -160,6 +173,34 @@ class DIEN(_SeqBase):
         with self.amp():
             out = self.mlp(x.to(self.compute_dtype))
         return out.float().squeeze(1)
+
+    def _auxiliary_loss(self, h_seq, seq, mask):
+        """Masked BCE over click pairs [h_t; e_{t+1}] vs no-click pairs
+        [h_t; e_shuffled_{t+1}] (reference semantics; sigmoid+BCE is the
+        numerically-stable form of its softmax+log)."""
+        h = h_seq[:, :-1].float()                        # [B,T-1,H]
+        click = seq[:, 1:].float()                       # real next items
+        perm = torch.randperm(seq.shape[0], device=seq.device)
+        noclick = click[perm]                            # in-batch negatives
+        m = (mask[:, :-1] * mask[:, 1:])                 # both steps valid
+        m = m * (perm != torch.arange(seq.shape[0],
+                                      device=seq.device)).float() \
+            .unsqueeze(1)                                # drop self-pairs
+        if m.sum() == 0:
+            return None
+        logit_c = self.aux_net(torch.cat([h, click], -1)).squeeze(2)
+        logit_n = self.aux_net(torch.cat([h, noclick], -1)).squeeze(2)
+        bce = nn.functional.binary_cross_entropy_with_logits
+        lc = bce(logit_c, torch.ones_like(logit_c), reduction="none")
+        ln = bce(logit_n, torch.zeros_like(logit_n), reduction="none")
+        return ((lc + ln) * m).sum() / m.sum()
+
+    def loss_fn(self, logits, labels):
+        loss = super().loss_fn(logits, labels)
+        if self._aux_loss is not None:
+            loss = loss + self.aux_alpha * self._aux_loss
+            self._aux_loss = None
+        return loss
 
 
 class BST(_SeqBase):
