@@ -60,8 +60,11 @@ class Optimizer:
         captured with — re-capture (or run eager) to change it."""
         self.lr = lr
         if self._dense is not None:
-            for g in self._dense.param_groups:
-                g["lr"] = lr
+            if hasattr(self._dense, "param_groups"):
+                for g in self._dense.param_groups:
+                    g["lr"] = lr
+            else:  # FlatDenseAdam delegate reads .lr per step
+                self._dense.lr = lr
 
     def zero_grad(self, set_to_none: bool = True):
         if self._dense is not None:
