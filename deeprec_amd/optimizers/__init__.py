@@ -90,11 +90,22 @@ class Optimizer:
         sparse grad-rows tensor) independently rescaled to norm <=
         clip_norm."""
         cn = self.clip_norm
-        for p in self._params:
-            if p.grad is not None:
-                n = p.grad.norm()
+        if self._dense is not None and hasattr(self._dense, "_slices"):
+            # FlatDenseAdam: dense grads live in the flat buffer, one
+            # [dW|db] slice per layer — clip per slice (same
+            # per-gradient semantics)
+            g = self._dense.g
+            for off, sz in self._dense._slices:
+                sl = g[off:off + sz]
+                n = sl.norm()
                 if n > cn:
-                    p.grad.mul_(cn / (n + 1e-12))
+                    sl.mul_(cn / (n + 1e-12))
+        else:
+            for p in self._params:
+                if p.grad is not None:
+                    n = p.grad.norm()
+                    if n > cn:
+                        p.grad.mul_(cn / (n + 1e-12))
         for ev in self.evs:
             for _, _, grad in ev._pending_grads:
                 n = grad.norm()
