@@ -94,6 +94,11 @@ def build_argparser():
     p.add_argument("--workqueue", default=None, metavar="GLOB",
                    help="shard parquet files across workers through the "
                         "checkpointable WorkQueue (reference: --workqueue)")
+    p.add_argument("--clip_norm", type=float, default=None,
+                   help="per-gradient norm clip (the reference DIN/DIEN "
+                        "train.py uses tf.clip_by_norm(grad, 5))")
+    p.add_argument("--lr_decay", default=None, metavar="STEPS:RATE",
+                   help="exponential learning-rate decay, e.g. 1000:0.9")
     p.add_argument("--eval_steps", type=int, default=0,
                    help="after training, evaluate N batches (no inserts, "
                         "no training) and print loss/accuracy/AUC — the "
@@ -230,8 +235,17 @@ def main(argv=None):
     saver = Saver(module=model,
                   embedding_variables=model.embedding_variables(),
                   optimizer=opt, rank=rank, world_size=world)
+    if args.clip_norm is not None:
+        opt.clip_norm = args.clip_norm
     hooks = [LoggingTensorHook(args.log_steps),
              StepCounterHook(args.log_steps, args.batch_size * world)]
+    if args.lr_decay:
+        from deeprec_amd.training.schedules import (
+            LearningRateScheduleHook, exponential_decay)
+        dsteps, drate = args.lr_decay.split(":")
+        hooks.append(LearningRateScheduleHook(
+            opt, exponential_decay(args.learning_rate, int(dsteps),
+                                   float(drate))))
     if args.timeline:
         hooks.append(ProfilerHook(args.timeline))
 
