@@ -147,12 +147,22 @@ class Saver:
         save_file(self._ev_payload(base), fn)
 
     def _cleanup(self, directory: str):
-        cks = sorted(glob.glob(os.path.join(directory, "ckpt-*")),
-                     key=lambda p: int(p.rsplit("-", 1)[1])
-                     if p.rsplit("-", 1)[1].isdigit() else -1)
-        cks = [c for c in cks if not c.endswith(".incr")]
+        def step_of(p):
+            tail = p.rsplit("-", 1)[1].replace(".incr", "")
+            return int(tail) if tail.isdigit() else -1
+
+        cks = sorted((c for c in glob.glob(os.path.join(directory,
+                                                        "ckpt-*"))
+                      if not c.endswith(".incr")), key=step_of)
         while len(cks) > self.keep_checkpoint_max:
             shutil.rmtree(cks.pop(0), ignore_errors=True)
+        if cks:
+            # incremental deltas older than the oldest kept full ckpt
+            # can never be replayed again (restore = full + NEWER incr)
+            oldest = step_of(cks[0])
+            for inc in glob.glob(os.path.join(directory, "ckpt-*.incr")):
+                if step_of(inc) < oldest:
+                    shutil.rmtree(inc, ignore_errors=True)
 
     # ------------- restore -------------
     def restore(self, ckpt_path: str) -> int:

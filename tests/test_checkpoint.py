@@ -230,3 +230,31 @@ def test_repartitioned_restore(tmp_path):
     o0, o1 = torch.argsort(keys), torch.argsort(k1)
     torch.testing.assert_close(keys[o0], k1[o1])
     torch.testing.assert_close(values[o0], v1[o1])
+
+
+def test_cleanup_prunes_fulls_and_stale_incrementals(tmp_path):
+    """keep_checkpoint_max prunes old fulls AND incremental deltas older
+    than the oldest kept full (they can never be replayed: restore =
+    latest full + NEWER deltas only)."""
+    import glob
+    import os
+
+    import torch
+    from deeprec_amd.checkpoint.saver import Saver
+    from deeprec_amd.embedding import EmbeddingVariable
+
+    ev = EmbeddingVariable("prune/ev", 4)
+    saver = Saver(embedding_variables=[ev], keep_checkpoint_max=2)
+    for step in range(1, 6):
+        ev.lookup_or_create(torch.tensor([step, step + 100]))
+        saver.save(str(tmp_path), step * 10)
+        ev.lookup_or_create(torch.tensor([step + 200]))
+        saver.incremental_save(str(tmp_path), step * 10 + 5)
+    fulls = sorted(os.path.basename(p) for p in
+                   glob.glob(str(tmp_path / "ckpt-*"))
+                   if not p.endswith(".incr"))
+    incrs = sorted(os.path.basename(p) for p in
+                   glob.glob(str(tmp_path / "ckpt-*.incr")))
+    assert fulls == ["ckpt-40", "ckpt-50"]
+    # deltas older than ckpt-40 are gone; newer ones kept
+    assert incrs == ["ckpt-45.incr", "ckpt-55.incr"]
