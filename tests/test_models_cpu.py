@@ -65,7 +65,8 @@ def test_dien_auxiliary_loss():
 
     m = DIEN(device="cpu", bf16=False)
     logits = m(dense, ids, seq, target)
-    assert m._aux_loss is not None and float(m._aux_loss) > 0
+    assert m._aux_loss is not None
+    assert float(m._aux_loss.detach()) > 0
     loss_aux = m.loss_fn(logits, labels)
     assert m._aux_loss is None  # consumed by loss_fn
 
