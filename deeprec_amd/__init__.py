@@ -9,7 +9,7 @@ HIP/CDNA4 kernels for the sparse hot path, RCCL over xGMI for collectives.
 Reference capability map: see SURVEY.md (reference: DeepRec-AI/DeepRec).
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from deeprec_amd.embedding.options import (  # noqa: F401
     EmbeddingVariableOption,
