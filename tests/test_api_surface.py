@@ -60,6 +60,11 @@ SYMBOLS = [
     "deeprec_amd.training.session:StepCounterHook",
     "deeprec_amd.training.session:LoggingTensorHook",
     "deeprec_amd.training.graph_step:GraphedTrainStep",
+    "deeprec_amd.training.schedules:exponential_decay",
+    "deeprec_amd.training.schedules:polynomial_decay",
+    "deeprec_amd.training.schedules:piecewise_constant",
+    "deeprec_amd.training.schedules:LearningRateScheduleHook",
+    "deeprec_amd.training.metrics:StreamingAUC",
     "deeprec_amd.training.cluster:parse_tf_config",
     "deeprec_amd.training.cluster:start_ps",
     "deeprec_amd.training.cluster:worker_embeddings",
@@ -95,6 +100,7 @@ SYMBOLS = [
     "deeprec_amd.ops.fp8:Fp8Linear",
     "deeprec_amd.ops.fp8:convert_mlp_to_fp8",
     "deeprec_amd.ops.fp8:quantize_fp8_rows",
+    "deeprec_amd.ops.fp8:Fp8MlpConverter",
     # models
     "deeprec_amd.models:MODEL_REGISTRY",
 ]
